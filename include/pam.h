@@ -1,0 +1,124 @@
+/* pam — pylops-mpi hot path, MI355X-native (gfx950) C-ABI.
+ *
+ * Stateless kernel entry points over raw device pointers + HIP streams.
+ * Device memory ownership, rank partitioning and RCCL collectives live in
+ * the host layer (pylops_mpi_amd, PyTorch-ROCm device tensors +
+ * torch.distributed over RCCL); this library is pure compute.
+ *
+ * Each entry point names the reference interface it replaces
+ * (paths relative to /root/reference/pylops_mpi/).
+ *
+ * Conventions:
+ *  - `stream` is a hipStream_t passed as void* (the caller's current
+ *    stream; the library never synchronizes).
+ *  - all functions return 0 on success, a hipError_t (>0) on launch
+ *    failure, or a PAM_E* code (<0) on argument errors.
+ *  - `dtype`: 0 = float64, 1 = float32.
+ *  - scalar outputs (`out`) are single-element float64 device buffers.
+ */
+#ifndef PAM_H
+#define PAM_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+#define PAM_F64 0
+#define PAM_F32 1
+
+#define PAM_EARG   (-1)  /* bad argument */
+#define PAM_EDTYPE (-2)  /* unsupported dtype */
+#define PAM_EOP    (-3)  /* unknown op code */
+
+/* ABI version, bumped on any signature change. */
+int64_t pam_version(void);
+
+/* Number of float64 workspace elements the reduction entry points need in
+ * `ws` (device buffer). */
+int64_t pam_reduce_ws_elems(void);
+
+/* ------------------------------------------------------------------ *
+ * Element-wise vector ops.
+ * Replaces DistributedArray.add/iadd/multiply/__neg__/__setitem__
+ * (ref DistributedArray.py:605-683,214-252) — per-rank local math on the
+ * HBM-resident block.
+ * ------------------------------------------------------------------ */
+int pam_fill(void* stream, void* y, int64_t n, double value, int dtype);
+int pam_neg(void* stream, void* y, const void* x, int64_t n, int dtype);
+/* y = a + b */
+int pam_add(void* stream, void* y, const void* a, const void* b, int64_t n,
+            int dtype);
+/* y = a - b (bitwise equal to the reference's add(-b), ref :624-628) */
+int pam_sub(void* stream, void* y, const void* a, const void* b, int64_t n,
+            int dtype);
+/* y = a * b */
+int pam_mul(void* stream, void* y, const void* a, const void* b, int64_t n,
+            int dtype);
+/* y = alpha * x */
+int pam_scale(void* stream, void* y, const void* x, double alpha, int64_t n,
+              int dtype);
+/* y += alpha * x — the fused CGLS update (x += a*c / s -= a*q,
+ * ref optimization/cls_basic.py:390-391: the reference materializes two
+ * temporaries per axpy; this is one pass). */
+int pam_axpy(void* stream, void* y, const void* x, double alpha, int64_t n,
+             int dtype);
+/* y = x + beta * y — the CGLS direction update c = r + b*c
+ * (ref optimization/cls_basic.py:396). */
+int pam_xpby(void* stream, void* y, const void* x, double beta, int64_t n,
+             int dtype);
+
+/* ------------------------------------------------------------------ *
+ * Reductions (wavefront-shuffle + LDS tree, deterministic tree shape for
+ * a given n — fixed partial count, fixed combine order).
+ * `ws` is a device float64 scratch of >= pam_reduce_ws_elems() elements;
+ * `out` is a 1-element float64 device buffer (the caller allreduces it
+ * across ranks, replacing ref Distributed.py:35-112).
+ * ------------------------------------------------------------------ */
+/* out = sum(x[i] * y[i]) in float64.
+ * Replaces the local part of DistributedArray.dot (ref :685-717). */
+int pam_dot(void* stream, const void* x, const void* y, int64_t n, void* ws,
+            void* out, int dtype);
+
+/* Local part of DistributedArray.norm / _compute_vector_norm
+ * (ref :719-838).  op: 0 = sum(|x^p|) (float_power semantics, ref :786),
+ * 1 = max(|x|), 2 = min(|x|), 3 = count_nonzero. */
+int pam_norm_local(void* stream, const void* x, int64_t n, int op, double p,
+                   void* ws, void* out, int dtype);
+
+/* ------------------------------------------------------------------ *
+ * Fused finite-difference stencils along axis 0 of a SCATTER-partitioned
+ * array.  One kernel per (operator, kind, direction): no materialized
+ * ghost concatenation (the reference copies the whole array per apply,
+ * ref DistributedArray.py:974,992-994,1028).
+ *
+ * Replaces MPIFirstDerivative._matvec_* and ._rmatvec_*
+ * (ref basicoperators/FirstDerivative.py:141-318) and
+ * MPISecondDerivative (ref basicoperators/SecondDerivative.py:124-256).
+ *
+ * Layout: x, y are [nloc, m] row-major (m = prod(dims[1:]) elements);
+ * gf/gb hold pam_fd_halo_width(op) neighbour planes ([w, m]; gf = last
+ * planes of rank-1, gb = first planes of rank+1; may be NULL at the
+ * global edges).  row0 = global row of local row 0, nglob = dims[0].
+ * coeff = 1/sampling (first derivative) or 1/sampling^2 (second).
+ *
+ * op codes:
+ *   0 fd1 forward  matvec   1 fd1 forward  rmatvec
+ *   2 fd1 backward matvec   3 fd1 backward rmatvec
+ *   4 fd1 cent3    matvec   5 fd1 cent3    rmatvec
+ *   6 fd1 cent5    matvec   7 fd1 cent5    rmatvec
+ *   8 fd2 forward  matvec   9 fd2 forward  rmatvec
+ *  10 fd2 backward matvec  11 fd2 backward rmatvec
+ *  12 fd2 centered matvec  13 fd2 centered rmatvec
+ * ------------------------------------------------------------------ */
+int64_t pam_fd_halo_width(int op);
+int pam_fd_apply(void* stream, int op, int edge, const void* x,
+                 const void* gf, const void* gb, void* y, int64_t nloc,
+                 int64_t m, int64_t row0, int64_t nglob, double coeff,
+                 int dtype);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* PAM_H */

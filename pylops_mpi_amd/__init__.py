@@ -1,0 +1,18 @@
+"""pylops_mpi_amd — MI355X-native rebuild of the pylops-mpi hot path.
+
+DistributedArray + MPILinearOperator matvec/rmatvec (stencils, elementwise
+math, dot/norm, CGLS) as hand-written HIP/CDNA4 kernels + RCCL over xGMI,
+drop-in under the reference's operator API
+(reference: PyLops/pylops-mpi @ /root/reference).
+"""
+from .distributedarray import (DistributedArray, Partition,  # noqa: F401
+                               local_split)
+from .linearoperator import MPILinearOperator  # noqa: F401
+from .derivative import (MPIFirstDerivative,  # noqa: F401
+                         MPISecondDerivative)
+from .solvers import CG, CGLS, cg, cgls  # noqa: F401
+from .dottest import dottest  # noqa: F401
+from .comm import (PamComm, get_default_comm,  # noqa: F401
+                   init_default_comm)
+
+__version__ = "0.1.0"

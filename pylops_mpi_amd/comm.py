@@ -1,0 +1,166 @@
+"""Communication layer: torch.distributed over RCCL (xGMI) / gloo.
+
+Replaces the reference's two-tier mpi4py + NCCL plane
+(ref pylops_mpi/Distributed.py:35-349, utils/_mpi.py, utils/_nccl.py):
+
+  * data plane  -> RCCL ("nccl" backend IS RCCL on ROCm) over xGMI,
+    one process per GPU, rendezvous via torchrun (no MPI bootstrap —
+    this image has no MPI).
+  * control plane (object allgathers of shapes/ints) -> a gloo side group,
+    mirroring the reference's rule that metadata always travels over MPI
+    even when NCCL carries the data (ref Distributed.py:143-153).
+
+World size 1 needs no process group at all (every op is a local no-op),
+so single-GPU runs work without torchrun.
+"""
+import datetime
+import os
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+_REDUCE_OPS = {"sum": "SUM", "max": "MAX", "min": "MIN", "prod": "PRODUCT"}
+
+
+class PamComm:
+    """A communicator: rank/size + the collectives the hot path uses."""
+
+    def __init__(self, rank: int = 0, size: int = 1,
+                 device: Optional[torch.device] = None,
+                 use_dist: bool = False):
+        self.rank = rank
+        self.size = size
+        self.device = device
+        self._use_dist = use_dist and size > 1
+        self._gloo_group = None
+        if self._use_dist and dist.get_backend() == "nccl":
+            # control-plane side group (object collectives off the GPU)
+            self._gloo_group = dist.new_group(backend="gloo")
+
+    # ------------------------------------------------------------ helpers
+    def Get_rank(self) -> int:  # reference-API spelling (MPI.Comm)
+        return self.rank
+
+    def Get_size(self) -> int:
+        return self.size
+
+    def barrier(self) -> None:
+        if self._use_dist:
+            dist.barrier()
+
+    # ------------------------------------------------------- collectives
+    def allreduce_(self, t: torch.Tensor, op: str = "sum") -> torch.Tensor:
+        """In-place allreduce of a tensor (scalar dots/norms,
+        ref Distributed.py:35-73)."""
+        if self._use_dist:
+            dist.all_reduce(t, op=getattr(dist.ReduceOp, _REDUCE_OPS[op]))
+        return t
+
+    def broadcast_(self, t: torch.Tensor, root: int = 0) -> torch.Tensor:
+        if self._use_dist:
+            dist.broadcast(t, src=root)
+        return t
+
+    def allgather_obj(self, obj) -> List:
+        """Object allgather on the control plane (ref Distributed.py:113-154
+        object-mode branch)."""
+        if not self._use_dist:
+            return [obj]
+        out = [None] * self.size
+        dist.all_gather_object(out, obj, group=self._gloo_group)
+        return out
+
+    def allgather_tensors(self, t: torch.Tensor,
+                          shapes: List[tuple]) -> List[torch.Tensor]:
+        """Allgather of same-rank, possibly unequal-size tensors via
+        pad-to-max (the reference's NCCL scheme, ref utils/_nccl.py:168,
+        363-403)."""
+        if not self._use_dist:
+            return [t]
+        counts = [int(torch.tensor(s).prod()) if len(s) else 1 for s in shapes]
+        mx = max(max(counts), 1)
+        send = torch.zeros(mx, dtype=t.dtype, device=t.device)
+        send[: t.numel()] = t.reshape(-1)
+        out = [torch.empty(mx, dtype=t.dtype, device=t.device)
+               for _ in range(self.size)]
+        dist.all_gather(out, send)
+        return [o[: counts[r]].reshape(shapes[r]) for r, o in enumerate(out)]
+
+    def sendrecv_neighbors(self, send_prev: Optional[torch.Tensor],
+                           send_next: Optional[torch.Tensor],
+                           recv_prev: Optional[torch.Tensor],
+                           recv_next: Optional[torch.Tensor]) -> None:
+        """Bidirectional nearest-neighbour exchange (halo / ghost cells,
+        ref DistributedArray.py:955-1032).  Any argument may be None
+        (global edges).  Posted as one batched isend/irecv group so RCCL
+        pairs them without deadlock."""
+        if not self._use_dist:
+            return
+        ops = []
+        if recv_prev is not None:
+            ops.append(dist.P2POp(dist.irecv, recv_prev, self.rank - 1))
+        if recv_next is not None:
+            ops.append(dist.P2POp(dist.irecv, recv_next, self.rank + 1))
+        if send_prev is not None:
+            ops.append(dist.P2POp(dist.isend, send_prev, self.rank - 1))
+        if send_next is not None:
+            ops.append(dist.P2POp(dist.isend, send_next, self.rank + 1))
+        if ops:
+            for w in dist.batch_isend_irecv(ops):
+                w.wait()
+
+    def sendrecv(self, sendbuf: torch.Tensor, dest: int,
+                 recvbuf: torch.Tensor, source: int) -> torch.Tensor:
+        """Pairwise exchange (ref Distributed.py:308-349)."""
+        if not self._use_dist:
+            recvbuf.copy_(sendbuf)
+            return recvbuf
+        ops = [dist.P2POp(dist.irecv, recvbuf, source),
+               dist.P2POp(dist.isend, sendbuf, dest)]
+        for w in dist.batch_isend_irecv(ops):
+            w.wait()
+        return recvbuf
+
+
+_default_comm: Optional[PamComm] = None
+
+
+def init_default_comm(device: Optional[torch.device] = None) -> PamComm:
+    """Initialize the process-wide communicator from torchrun env vars
+    (RANK/WORLD_SIZE/LOCAL_RANK/MASTER_*), RCCL backend on GPU, gloo on
+    CPU.  Idempotent."""
+    global _default_comm
+    if _default_comm is not None:
+        return _default_comm
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if device is None:
+        if torch.cuda.is_available():
+            device = torch.device(f"cuda:{local_rank}")
+        else:
+            device = torch.device("cpu")
+    if device.type == "cuda":
+        torch.cuda.set_device(device)
+    if world > 1:
+        if not dist.is_initialized():
+            backend = "nccl" if device.type == "cuda" else "gloo"
+            dist.init_process_group(
+                backend=backend, rank=rank, world_size=world,
+                timeout=datetime.timedelta(seconds=300))
+        _default_comm = PamComm(rank, world, device, use_dist=True)
+    else:
+        _default_comm = PamComm(0, 1, device, use_dist=False)
+    return _default_comm
+
+
+def get_default_comm() -> PamComm:
+    if _default_comm is None:
+        return init_default_comm()
+    return _default_comm
+
+
+def set_default_comm(comm: PamComm) -> None:
+    global _default_comm
+    _default_comm = comm

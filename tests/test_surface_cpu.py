@@ -1,0 +1,105 @@
+"""Drop-in surface behaviour that does not need a GPU (world size 1):
+constructor/validation semantics, scatter indexing, operator algebra, and
+the fail-loud rule for compute without CUDA."""
+import numpy as np
+import pytest
+import torch
+
+import pylops_mpi_amd as pm
+from pylops_mpi_amd.distributedarray import local_split
+import oracle
+
+
+def test_local_split_matches_oracle():
+    for n in (7, 8, 1023):
+        for P in (1, 2, 3, 8):
+            for r in range(P):
+                assert local_split((n, 5), P, r) == \
+                    oracle.local_split((n, 5), P, r)
+
+
+def test_ctor_validation():
+    # ref DistributedArray.py:175-180,207-211
+    with pytest.raises(IndexError):
+        pm.DistributedArray((4,), axis=1)
+    with pytest.raises(ValueError):
+        pm.DistributedArray((4,), partition="scatter")
+    with pytest.raises(ValueError):
+        pm.DistributedArray((4,), local_array=torch.zeros(3, dtype=torch.float64))
+    with pytest.raises(ValueError):
+        pm.DistributedArray((4,), local_shapes=[(3,), (1,)])  # wrong count
+
+
+def test_to_dist_and_asarray_world1():
+    x = torch.arange(24, dtype=torch.float64).reshape(6, 4)
+    d = pm.DistributedArray.to_dist(x)
+    assert torch.equal(d.local_array, x)
+    assert torch.equal(d.asarray(), x)
+    assert d.local_shapes == [(6, 4)]
+    r = d.ravel()
+    assert r.global_shape == (24,)
+    assert torch.equal(r.local_array, x.reshape(-1))
+
+
+def test_setitem_getitem():
+    d = pm.DistributedArray((5,), dtype=np.float64)
+    d[:] = 3.0
+    assert torch.all(d.local_array == 3.0)
+    d[1] = 7.0
+    assert float(d[1]) == 7.0
+
+
+def test_compute_requires_gpu_fails_loudly():
+    if torch.cuda.is_available():
+        pytest.skip("GPU present")
+    a = pm.DistributedArray((4,))
+    b = pm.DistributedArray((4,))
+    a[:] = 1.0
+    b[:] = 2.0
+    with pytest.raises(RuntimeError, match="no CPU compute path"):
+        a.add(b)
+    with pytest.raises(RuntimeError, match="no CPU compute path"):
+        a.dot(b)
+    with pytest.raises(RuntimeError, match="no CPU compute path"):
+        a.norm()
+
+
+def test_operator_algebra_shapes():
+    op = pm.MPIFirstDerivative((8, 4))
+    assert op.shape == (32, 32)
+    assert op.H.shape == (32, 32)
+    assert op.T.shape == (32, 32)
+    assert (op * op).shape == (32, 32)
+    assert (2.0 * op).shape == (32, 32)
+    assert (op + op).shape == (32, 32)
+    assert (op ** 2).shape == (32, 32)
+    assert (-op).shape == (32, 32)
+    with pytest.raises(ValueError):
+        op @ 2.0
+
+
+def test_matvec_dimension_mismatch():
+    # ref LinearOperator.py:190,228 — the shape check precedes any compute
+    op = pm.MPIFirstDerivative((8, 4))
+    x = pm.DistributedArray((31,))
+    with pytest.raises(ValueError, match="dimension mismatch"):
+        op.matvec(x)
+    with pytest.raises(ValueError, match="dimension mismatch"):
+        op.rmatvec(x)
+
+
+def test_kind_validation():
+    with pytest.raises(NotImplementedError):
+        pm.MPIFirstDerivative((8,), kind="sideways")
+    with pytest.raises(NotImplementedError):
+        pm.MPIFirstDerivative((8,), kind="centered", order=7)
+    with pytest.raises(NotImplementedError):
+        pm.MPISecondDerivative((8,), kind="sideways")
+
+
+def test_reshaped_partition_check():
+    # ref decorators.py:45-46
+    op = pm.MPIFirstDerivative((8,))
+    x = pm.DistributedArray((8,), partition=pm.Partition.UNSAFE_BROADCAST)
+    with pytest.raises(ValueError, match="should have partition"):
+        op.matvec(x)
