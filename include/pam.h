@@ -118,6 +118,21 @@ int pam_fd_apply(void* stream, int op, int edge, const void* x,
                  int64_t m, int64_t row0, int64_t nglob, double coeff,
                  int dtype);
 
+/* ------------------------------------------------------------------ *
+ * Dense GEMV: y = A @ x (trans=0) or y = A^T @ x (trans=1) for a
+ * row-major [nr, nc] matrix resident in HBM.
+ *
+ * The local apply of a dense operator inside MPIBlockDiag
+ * (ref basicoperators/BlockDiag.py:122-144 calling a serial dense
+ * MatrixMult per rank, as in examples/plot_cgls.py:30-33) — HBM-bound
+ * matrix read, wave-per-row (trans=0) or column-tile partials with a
+ * deterministic chunk combine (trans=1; `ws` is a float64 device scratch
+ * of pam_gemv_ws_elems(nr, nc) elements, unused for trans=0).
+ * ------------------------------------------------------------------ */
+int64_t pam_gemv_ws_elems(int64_t nr, int64_t nc);
+int pam_gemv(void* stream, int trans, const void* A, const void* x, void* y,
+             int64_t nr, int64_t nc, void* ws, int dtype);
+
 #ifdef __cplusplus
 }
 #endif

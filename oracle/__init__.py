@@ -45,3 +45,4 @@ from .stencils import (  # noqa: F401
     SimSecondDerivative,
 )
 from .cgls import sim_cgls, sim_cg  # noqa: F401
+from .blockdiag import SimBlockDiag  # noqa: F401

@@ -197,17 +197,23 @@ class SimArray:
         return add_ghost_cells(self.locals, cf, cb, axis=self.axis)
 
 
-def reshaped_apply(body, dims: Tuple[int, ...], x: SimArray) -> SimArray:
+def reshaped_apply(body, dims: Tuple[int, ...], x: SimArray,
+                   target_counts: Optional[Sequence[int]] = None) -> SimArray:
     """The ``@reshaped`` wrapper, ref utils/decorators.py:44-82.
 
     Rebalances the flat 1-D input ``x`` to the plane-aligned split of
-    ``dims`` (axis 0), reshapes, calls ``body(list_of_locals) ->
-    list_of_locals`` and ravels the result back to 1-D.
+    ``dims`` (axis 0) — or, in the stacking form (ref decorators.py:47-52),
+    to the explicit per-rank ``target_counts`` — reshapes, calls
+    ``body(list_of_locals) -> list_of_locals`` and ravels the result
+    back to 1-D.
     """
     if x.partition is not Partition.SCATTER:
         raise ValueError(f"x should have partition={Partition.SCATTER}")
     P = x.size
-    arr_shapes = [local_split(dims, P, r) for r in range(P)]
+    if target_counts is None:
+        arr_shapes = [local_split(dims, P, r) for r in range(P)]
+    else:
+        arr_shapes = [(int(c),) for c in target_counts]
     arr_counts = np.asarray([int(np.prod(s)) for s in arr_shapes])
     x_counts = np.asarray([int(a.size) for a in x.locals])
     # cumulative imbalance, ref decorators.py:69-73
@@ -221,5 +227,7 @@ def reshaped_apply(body, dims: Tuple[int, ...], x: SimArray) -> SimArray:
         arr_locals.append(
             ghosted[r][index: arr_counts[r] + index].reshape(arr_shapes[r]))
     y_locals = body(arr_locals)
+    gsize = int(np.prod(dims)) if dims is not None \
+        else int(sum(y.size for y in y_locals))
     return SimArray([y.ravel() for y in y_locals],
-                    (int(np.prod(dims)),), 0, Partition.SCATTER)
+                    (gsize,), 0, Partition.SCATTER)

@@ -8,6 +8,8 @@ drop-in under the reference's operator API
 from .distributedarray import (DistributedArray, Partition,  # noqa: F401
                                local_split)
 from .linearoperator import MPILinearOperator  # noqa: F401
+from .blockdiag import MPIBlockDiag  # noqa: F401
+from .localops import DenseLocal, CallableLocal, LocalOperator  # noqa: F401
 from .derivative import (MPIFirstDerivative,  # noqa: F401
                          MPISecondDerivative)
 from .solvers import CG, CGLS, cg, cgls  # noqa: F401

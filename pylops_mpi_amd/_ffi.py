@@ -42,6 +42,11 @@ _SIGS = {
     "pam_norm_local": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
                         ctypes.c_int, ctypes.c_double, ctypes.c_void_p,
                         ctypes.c_void_p, ctypes.c_int], ctypes.c_int),
+    "pam_gemv_ws_elems": ([ctypes.c_int64, ctypes.c_int64], ctypes.c_int64),
+    "pam_gemv": ([ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+                  ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
+                  ctypes.c_int64, ctypes.c_void_p, ctypes.c_int],
+                 ctypes.c_int),
     "pam_fd_halo_width": ([ctypes.c_int], ctypes.c_int64),
     "pam_fd_apply": ([ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
                       ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,

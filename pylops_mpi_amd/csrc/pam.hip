@@ -318,6 +318,115 @@ extern "C" int pam_norm_local(void* stream, const void* x, int64_t n, int op,
 }
 
 // ---------------------------------------------------------------------------
+// dense GEMV (local apply of MPIBlockDiag dense operators,
+// ref basicoperators/BlockDiag.py:122-144).  HBM-bound: the matrix read
+// dominates (nr*nc elements, read once).
+// ---------------------------------------------------------------------------
+#define GEMV_CHUNKS 64  // deterministic row-chunk count for the trans path
+
+extern "C" int64_t pam_gemv_ws_elems(int64_t nr, int64_t nc) {
+  (void)nr;
+  return GEMV_CHUNKS * nc;
+}
+
+// y = A @ x: one 64-lane wave per output row, lanes stride the row with
+// 16-B vector loads, wavefront shuffle reduction.
+template <typename T, int V>
+__global__ void __launch_bounds__(BLK) gemv_n_kernel(
+    const T* __restrict__ A, const T* __restrict__ x, T* __restrict__ y,
+    int64_t nr, int64_t nc) {
+  const int64_t wave = ((int64_t)blockIdx.x * (BLK / 64))
+                       + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  const int64_t nwaves = (int64_t)gridDim.x * (BLK / 64);
+  const int64_t ncv = nc / V;
+  for (int64_t r = wave; r < nr; r += nwaves) {
+    const T* __restrict__ row = A + r * nc;
+    double acc = 0.0;
+    for (int64_t cv = lane; cv < ncv; cv += 64) {
+      T a[V], b[V];
+      loadv<T, V>(row + cv * V, a);
+      loadv<T, V>(x + cv * V, b);
+#pragma unroll
+      for (int k = 0; k < V; ++k) acc += (double)a[k] * (double)b[k];
+    }
+    for (int64_t c = ncv * V + lane; c < nc; c += 64)
+      acc += (double)row[c] * (double)x[c];
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
+    if (lane == 0) y[r] = (T)acc;
+  }
+}
+
+// y = A^T @ x, stage 1: block (cb, chunk) accumulates its row chunk into
+// partials[chunk*nc + c] — fixed chunk count => deterministic combine.
+template <typename T>
+__global__ void __launch_bounds__(BLK) gemv_t_stage1(
+    const T* __restrict__ A, const T* __restrict__ x,
+    double* __restrict__ partials, int64_t nr, int64_t nc) {
+  const int64_t c = (int64_t)blockIdx.x * BLK + threadIdx.x;
+  const int chunk = blockIdx.y;
+  const int64_t r0 = (nr * chunk) / GEMV_CHUNKS;
+  const int64_t r1 = (nr * (chunk + 1)) / GEMV_CHUNKS;
+  if (c >= nc) return;
+  double acc = 0.0;
+  for (int64_t r = r0; r < r1; ++r)
+    acc += (double)A[r * nc + c] * (double)x[r];
+  partials[(int64_t)chunk * nc + c] = acc;
+}
+
+template <typename T>
+__global__ void __launch_bounds__(BLK) gemv_t_stage2(
+    const double* __restrict__ partials, T* __restrict__ y, int64_t nc) {
+  const int64_t c = (int64_t)blockIdx.x * BLK + threadIdx.x;
+  if (c >= nc) return;
+  double acc = 0.0;
+  for (int ch = 0; ch < GEMV_CHUNKS; ++ch)
+    acc += partials[(int64_t)ch * nc + c];
+  y[c] = (T)acc;
+}
+
+template <typename T>
+static int gemv_launch(void* stream, int trans, const void* A, const void* x,
+                       void* y, int64_t nr, int64_t nc, void* ws) {
+  if (nr <= 0 || nc <= 0 || !A || !x || !y) return PAM_EARG;
+  hipStream_t s = (hipStream_t)stream;
+  if (!trans) {
+    constexpr int V = VecW<T>::value;
+    const bool vec_ok = (nc % V == 0) && ((uintptr_t)A % 16 == 0) &&
+                        ((uintptr_t)x % 16 == 0);
+    int64_t nb = (nr + (BLK / 64) - 1) / (BLK / 64);
+    if (nb > 4096) nb = 4096;
+    if (vec_ok)
+      hipLaunchKernelGGL((gemv_n_kernel<T, V>), dim3((uint32_t)nb), dim3(BLK),
+                         0, s, (const T*)A, (const T*)x, (T*)y, nr, nc);
+    else
+      hipLaunchKernelGGL((gemv_n_kernel<T, 1>), dim3((uint32_t)nb), dim3(BLK),
+                         0, s, (const T*)A, (const T*)x, (T*)y, nr, nc);
+    return check(hipGetLastError());
+  }
+  if (!ws) return PAM_EARG;
+  dim3 g1((uint32_t)((nc + BLK - 1) / BLK), GEMV_CHUNKS);
+  hipLaunchKernelGGL((gemv_t_stage1<T>), g1, dim3(BLK), 0, s, (const T*)A,
+                     (const T*)x, (double*)ws, nr, nc);
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) return (int)e;
+  hipLaunchKernelGGL((gemv_t_stage2<T>),
+                     dim3((uint32_t)((nc + BLK - 1) / BLK)), dim3(BLK), 0, s,
+                     (const double*)ws, (T*)y, nc);
+  return check(hipGetLastError());
+}
+
+extern "C" int pam_gemv(void* stream, int trans, const void* A, const void* x,
+                        void* y, int64_t nr, int64_t nc, void* ws, int dtype) {
+  if (dtype == PAM_F64)
+    return gemv_launch<double>(stream, trans, A, x, y, nr, nc, ws);
+  if (dtype == PAM_F32)
+    return gemv_launch<float>(stream, trans, A, x, y, nr, nc, ws);
+  return PAM_EDTYPE;
+}
+
+// ---------------------------------------------------------------------------
 // fused finite-difference stencils
 // (ref FirstDerivative.py:141-318, SecondDerivative.py:124-256 — closed
 //  forms of the slice algebra; masks are global-row intervals)

@@ -23,6 +23,7 @@ from . import _ffi
 from .distributedarray import (DistributedArray, Partition, as_torch_dtype,
                                local_split)
 from .linearoperator import MPILinearOperator
+from .rebalance import rebalance_1d
 
 _FD1_OPS = {("forward", 0): (0, 1), ("backward", 0): (2, 3),
             ("centered", 3): (4, 5), ("centered", 5): (6, 7)}
@@ -66,38 +67,7 @@ class _FDBase(MPILinearOperator):
     def _rebalance(self, x: DistributedArray, counts):
         """ref decorators.py:66-77 — move the 1-D input to the plane-aligned
         split.  Balanced inputs are a zero-copy view."""
-        P, r = x.size, x.rank
-        x_counts = [int(np.prod(s)) for s in x.local_shapes]
-        if x_counts == counts:
-            return x.local_array.reshape(-1)
-        dif = np.cumsum(np.asarray(counts) - np.asarray(x_counts))
-        cf = [int(abs(min(0, dif[q - 1]))) for q in range(P)]
-        cb = [int(max(0, dif[q])) for q in range(P)]
-        t = x.local_array.reshape(-1)
-        # rank r sends its last cf[r+1] elements to r+1 and its first
-        # cb[r-1] elements to r-1 (ref DistributedArray.py:976-1031)
-        if r < P - 1 and cf[r + 1] > t.numel():
-            raise ValueError(
-                f"Local Shape at rank={r} along axis=0 should be > "
-                f"{cf[r + 1]}")
-        if r > 0 and cb[r - 1] > t.numel():
-            raise ValueError(
-                f"Local Shape at rank={r} along axis=0 should be > "
-                f"{cb[r - 1]}")
-        send_next = t[-cf[r + 1]:].contiguous() \
-            if r < P - 1 and cf[r + 1] > 0 else None
-        send_prev = t[: cb[r - 1]].contiguous() \
-            if r > 0 and cb[r - 1] > 0 else None
-        recv_front = torch.empty(cf[r], dtype=t.dtype, device=t.device) \
-            if r > 0 and cf[r] > 0 else None
-        recv_back = torch.empty(cb[r], dtype=t.dtype, device=t.device) \
-            if r < P - 1 and cb[r] > 0 else None
-        x.base_comm.sendrecv_neighbors(send_prev, send_next,
-                                       recv_front, recv_back)
-        parts = [p for p in (recv_front, t, recv_back) if p is not None]
-        ghosted = torch.cat(parts) if len(parts) > 1 else t
-        index = int(max(0, dif[r - 1]))
-        return ghosted[index: index + counts[r]]
+        return rebalance_1d(x, counts)
 
     def _halo(self, planes: torch.Tensor, w: int, comm):
         """Exchange w boundary planes of the [nloc, m] block with the

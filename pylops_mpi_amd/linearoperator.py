@@ -22,8 +22,18 @@ class MPILinearOperator:
     def __init__(self, Op=None, shape=None, dims=None, dimsd=None,
                  dtype=None, base_comm: Optional[PamComm] = None):
         if Op is not None:
-            raise NotImplementedError(
-                "wrapping a serial operator is deferred (MPIBlockDiag round)")
+            # wrap a serial (local) operator, ref :60-66: it is applied
+            # identically on every rank's local array (meant for
+            # BROADCAST-partitioned arrays, ref :22-27)
+            self.Op = Op
+            dtype = Op.dtype if dtype is None else dtype
+            shape = Op.shape if shape is None else shape
+            dims = getattr(Op, "dims", (Op.shape[1],)) if dims is None \
+                else dims
+            dimsd = getattr(Op, "dimsd", (Op.shape[0],)) if dimsd is None \
+                else dimsd
+        else:
+            self.Op = None
         if shape is not None:
             self.shape = shape
         if dims is not None:
@@ -100,9 +110,21 @@ class MPILinearOperator:
         return self._rmatvec(x)
 
     def _matvec(self, x):
+        # ref :194-204 — serial-op wrap applies locally on every rank
+        if getattr(self, "Op", None) is not None:
+            y = DistributedArray(self.shape[0], x.base_comm, x.partition,
+                                 x.axis, dtype=self.dtype)
+            y[:] = self.Op.matvec(x.local_array)
+            return y
         raise NotImplementedError
 
     def _rmatvec(self, x):
+        # ref :232-242
+        if getattr(self, "Op", None) is not None:
+            y = DistributedArray(self.shape[1], x.base_comm, x.partition,
+                                 x.axis, dtype=self.dtype)
+            y[:] = self.Op.rmatvec(x.local_array)
+            return y
         raise NotImplementedError
 
     # ------------------------------------------------------------ algebra

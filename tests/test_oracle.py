@@ -212,3 +212,39 @@ def test_cg_solves_spd(P):
     x, cost = sim_cg(NormalOp(), to_dist(yg, P), to_dist(np.zeros(n), P),
                      niter=200, tol=1e-30)
     assert_allclose(x.asarray(), np.linalg.solve(A, yg), rtol=1e-7, atol=1e-8)
+
+
+# ----------------------------------------------------------------- blockdiag
+@pytest.mark.parametrize("P", RANKS)
+def test_blockdiag_vs_dense(P):
+    """SimBlockDiag vs its explicit global block-diagonal matrix, with
+    uneven per-rank blocks (ref BlockDiag.py:100-144 semantics)."""
+    from oracle import SimBlockDiag
+    rng = np.random.default_rng(21)
+    mats = []
+    for r in range(P):
+        ms = [rng.standard_normal((3 + r, 4)), rng.standard_normal((2, 2 + r))]
+        mats.append(ms)
+    op = SimBlockDiag(mats)
+    A = op.dense()
+    n, m = op.shape
+    x = to_dist(rng.standard_normal(m), P)
+    y = to_dist(rng.standard_normal(n), P)
+    assert_allclose(op.matvec(x).asarray(), A @ x.asarray(), rtol=1e-13)
+    assert_allclose(op.rmatvec(y).asarray(), A.T @ y.asarray(), rtol=1e-13)
+
+
+@pytest.mark.parametrize("P", [1, 2, 4])
+def test_cgls_blockdiag(P):
+    """CGLS over a block-diagonal dense system (the reference's
+    examples/plot_cgls.py:30-43 recipe) vs a direct solve."""
+    from oracle import SimBlockDiag
+    rng = np.random.default_rng(30)
+    mats = [[rng.standard_normal((6, 6)) + 6 * np.eye(6)] for _ in range(P)]
+    op = SimBlockDiag(mats)
+    A = op.dense()
+    n = op.shape[0]
+    yg = rng.standard_normal(n)
+    x, cost = sim_cgls(op, to_dist(yg, P), to_dist(np.zeros(n), P),
+                       niter=80, damp=0.0, tol=1e-30)
+    assert_allclose(x.asarray(), np.linalg.solve(A, yg), rtol=1e-8, atol=1e-9)
