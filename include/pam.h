@@ -133,6 +133,25 @@ int64_t pam_gemv_ws_elems(int64_t nr, int64_t nc);
 int pam_gemv(void* stream, int trans, const void* A, const void* x, void* y,
              int64_t nr, int64_t nc, void* ws, int dtype);
 
+/* ------------------------------------------------------------------ *
+ * Panel GEMM: C = A @ B (accumulate != 0: C += A @ B) for row-major
+ * A [M,K] (lda), B [K,N] (ldb), C [M,N] (ldc).
+ *
+ * The local panel product of MPIMatrixMult — block kind
+ * (ref basicoperators/MatrixMult.py:369-372 `ncp.matmul(A_local,
+ * X_local)`) and the SUMMA accumulation step (ref :661-668
+ * `Y_local += ncp.dot(Atemp, Xtemp)`) — on MFMA matrix cores:
+ * v_mfma_f32_32x32x2_f32 (exact f32) / v_mfma_f64_16x16x4_f64.
+ *
+ * pam_transpose materializes A^T for the adjoint panels
+ * (ref MatrixMult.py:416,737 `A.T.conj()`; real dtypes).
+ * ------------------------------------------------------------------ */
+int pam_gemm(void* stream, const void* A, const void* B, void* C, int64_t M,
+             int64_t N, int64_t K, int64_t lda, int64_t ldb, int64_t ldc,
+             int accumulate, int dtype);
+int pam_transpose(void* stream, const void* A, void* At, int64_t nr,
+                  int64_t nc, int dtype);
+
 #ifdef __cplusplus
 }
 #endif

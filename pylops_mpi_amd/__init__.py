@@ -9,6 +9,9 @@ from .distributedarray import (DistributedArray, Partition,  # noqa: F401
                                local_split)
 from .linearoperator import MPILinearOperator  # noqa: F401
 from .blockdiag import MPIBlockDiag  # noqa: F401
+from . import matmult  # noqa: F401
+from .matmult import (MPIMatrixMult, active_grid_comm,  # noqa: F401
+                      block_gather, local_block_split)
 from .localops import DenseLocal, CallableLocal, LocalOperator  # noqa: F401
 from .derivative import (MPIFirstDerivative,  # noqa: F401
                          MPISecondDerivative)

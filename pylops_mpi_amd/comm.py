@@ -9,13 +9,16 @@ Replaces the reference's two-tier mpi4py + NCCL plane
   * control plane (object allgathers of shapes/ints) -> a gloo side group,
     mirroring the reference's rule that metadata always travels over MPI
     even when NCCL carries the data (ref Distributed.py:143-153).
+  * sub-communicators (the reference's comm.Split for MatrixMult process
+    grids, ref MatrixMult.py:305-306, DistributedArray.py:74-100) ->
+    torch process groups created collectively via split_by().
 
 World size 1 needs no process group at all (every op is a local no-op),
 so single-GPU runs work without torchrun.
 """
 import datetime
 import os
-from typing import List, Optional
+from typing import List, Optional, Sequence
 
 import torch
 import torch.distributed as dist
@@ -24,17 +27,26 @@ _REDUCE_OPS = {"sum": "SUM", "max": "MAX", "min": "MIN", "prod": "PRODUCT"}
 
 
 class PamComm:
-    """A communicator: rank/size + the collectives the hot path uses."""
+    """A communicator: rank/size + the collectives the hot path uses.
+
+    ``group`` is a torch.distributed ProcessGroup (None = default/world);
+    ``ranks`` maps group rank -> global rank.
+    """
 
     def __init__(self, rank: int = 0, size: int = 1,
                  device: Optional[torch.device] = None,
-                 use_dist: bool = False):
+                 use_dist: bool = False, group=None,
+                 ranks: Optional[List[int]] = None,
+                 gloo_group=None):
         self.rank = rank
         self.size = size
         self.device = device
         self._use_dist = use_dist and size > 1
-        self._gloo_group = None
-        if self._use_dist and dist.get_backend() == "nccl":
+        self._group = group
+        self.ranks = ranks if ranks is not None else list(range(size))
+        self._gloo_group = gloo_group
+        if (self._use_dist and group is None and gloo_group is None
+                and dist.get_backend() == "nccl"):
             # control-plane side group (object collectives off the GPU)
             self._gloo_group = dist.new_group(backend="gloo")
 
@@ -45,21 +57,52 @@ class PamComm:
     def Get_size(self) -> int:
         return self.size
 
+    def global_rank(self, r: int) -> int:
+        return self.ranks[r]
+
     def barrier(self) -> None:
         if self._use_dist:
-            dist.barrier()
+            dist.barrier(group=self._group)
+
+    def split_by(self, colors: Sequence[int], keys: Optional[Sequence[int]]
+                 = None) -> "PamComm":
+        """The reference's ``comm.Split(color, key)``
+        (ref MatrixMult.py:305-306): ``colors``/``keys`` are indexed by
+        GLOBAL rank and must be identical on every rank (they are
+        computed from the deterministic grid layout).  Every process
+        creates every group (torch.distributed requirement) and returns
+        its own."""
+        if not self._use_dist:
+            return PamComm(0, 1, self.device, use_dist=False)
+        if keys is None:
+            keys = list(range(len(colors)))
+        mine = None
+        my_ranks = None
+        for color in sorted(set(colors)):
+            members = sorted((r for r in range(len(colors))
+                              if colors[r] == color),
+                             key=lambda r: (keys[r], r))
+            g = dist.new_group(ranks=members)
+            if colors[dist.get_rank()] == color:
+                mine, my_ranks = g, members
+        gr = my_ranks.index(dist.get_rank())
+        return PamComm(gr, len(my_ranks), self.device,
+                       use_dist=len(my_ranks) > 1, group=mine,
+                       ranks=my_ranks, gloo_group=self._gloo_group)
 
     # ------------------------------------------------------- collectives
     def allreduce_(self, t: torch.Tensor, op: str = "sum") -> torch.Tensor:
         """In-place allreduce of a tensor (scalar dots/norms,
         ref Distributed.py:35-73)."""
         if self._use_dist:
-            dist.all_reduce(t, op=getattr(dist.ReduceOp, _REDUCE_OPS[op]))
+            dist.all_reduce(t, op=getattr(dist.ReduceOp, _REDUCE_OPS[op]),
+                            group=self._group)
         return t
 
     def broadcast_(self, t: torch.Tensor, root: int = 0) -> torch.Tensor:
+        """Broadcast from group rank ``root`` (ref Distributed.py:195-225)."""
         if self._use_dist:
-            dist.broadcast(t, src=root)
+            dist.broadcast(t, src=self.ranks[root], group=self._group)
         return t
 
     def allgather_obj(self, obj) -> List:
@@ -68,7 +111,9 @@ class PamComm:
         if not self._use_dist:
             return [obj]
         out = [None] * self.size
-        dist.all_gather_object(out, obj, group=self._gloo_group)
+        dist.all_gather_object(out, obj,
+                               group=self._gloo_group if self._group is None
+                               else self._group)
         return out
 
     def allgather_tensors(self, t: torch.Tensor,
@@ -78,13 +123,13 @@ class PamComm:
         363-403)."""
         if not self._use_dist:
             return [t]
-        counts = [int(torch.tensor(s).prod()) if len(s) else 1 for s in shapes]
+        counts = [int(np_prod(s)) for s in shapes]
         mx = max(max(counts), 1)
         send = torch.zeros(mx, dtype=t.dtype, device=t.device)
         send[: t.numel()] = t.reshape(-1)
         out = [torch.empty(mx, dtype=t.dtype, device=t.device)
                for _ in range(self.size)]
-        dist.all_gather(out, send)
+        dist.all_gather(out, send, group=self._group)
         return [o[: counts[r]].reshape(shapes[r]) for r, o in enumerate(out)]
 
     def sendrecv_neighbors(self, send_prev: Optional[torch.Tensor],
@@ -97,30 +142,56 @@ class PamComm:
         pairs them without deadlock."""
         if not self._use_dist:
             return
+        prev = self.ranks[self.rank - 1] if self.rank > 0 else None
+        nxt = self.ranks[self.rank + 1] if self.rank < self.size - 1 else None
         ops = []
         if recv_prev is not None:
-            ops.append(dist.P2POp(dist.irecv, recv_prev, self.rank - 1))
+            ops.append(dist.P2POp(dist.irecv, recv_prev, prev))
         if recv_next is not None:
-            ops.append(dist.P2POp(dist.irecv, recv_next, self.rank + 1))
+            ops.append(dist.P2POp(dist.irecv, recv_next, nxt))
         if send_prev is not None:
-            ops.append(dist.P2POp(dist.isend, send_prev, self.rank - 1))
+            ops.append(dist.P2POp(dist.isend, send_prev, prev))
         if send_next is not None:
-            ops.append(dist.P2POp(dist.isend, send_next, self.rank + 1))
+            ops.append(dist.P2POp(dist.isend, send_next, nxt))
         if ops:
             for w in dist.batch_isend_irecv(ops):
                 w.wait()
 
     def sendrecv(self, sendbuf: torch.Tensor, dest: int,
                  recvbuf: torch.Tensor, source: int) -> torch.Tensor:
-        """Pairwise exchange (ref Distributed.py:308-349)."""
+        """Pairwise exchange by GROUP rank (ref Distributed.py:308-349)."""
         if not self._use_dist:
             recvbuf.copy_(sendbuf)
             return recvbuf
-        ops = [dist.P2POp(dist.irecv, recvbuf, source),
-               dist.P2POp(dist.isend, sendbuf, dest)]
+        ops = [dist.P2POp(dist.irecv, recvbuf, self.ranks[source]),
+               dist.P2POp(dist.isend, sendbuf, self.ranks[dest])]
         for w in dist.batch_isend_irecv(ops):
             w.wait()
         return recvbuf
+
+    def exchange(self, sends: List, recvs: List) -> None:
+        """Batched point-to-point round: ``sends``/``recvs`` are lists of
+        (tensor, group_rank) pairs, posted together (the SUMMA adjoint's
+        tag-routed A^T exchange, ref MatrixMult.py:745-760 — NCCL has no
+        tags; each (src,dst) pair carries exactly one message per round)."""
+        if not self._use_dist:
+            for (st, _), (rt, _) in zip(sends, recvs):
+                rt.copy_(st)
+            return
+        ops = [dist.P2POp(dist.irecv, t, self.ranks[src])
+               for t, src in recvs]
+        ops += [dist.P2POp(dist.isend, t, self.ranks[dst])
+                for t, dst in sends]
+        if ops:
+            for w in dist.batch_isend_irecv(ops):
+                w.wait()
+
+
+def np_prod(s) -> int:
+    p = 1
+    for v in s:
+        p *= int(v)
+    return p
 
 
 _default_comm: Optional[PamComm] = None

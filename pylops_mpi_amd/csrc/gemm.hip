@@ -1,0 +1,255 @@
+// pam — MFMA GEMM + transpose for the MatrixMult path (gfx950/CDNA4).
+//
+// Y = A @ B (optionally += onto C) for row-major matrices, used by
+// MPIMatrixMult block/SUMMA local products
+// (ref basicoperators/MatrixMult.py:341-427, :610-765: the reference calls
+// ncp.matmul on CuPy; here the panels run on hand-written MFMA kernels).
+//
+// Shapes (cdna_hip_programming.md §3/§5 canonical anatomy):
+//   f32: v_mfma_f32_32x32x2_f32  — exact f32 at the 157 TF vector-rate peak
+//        (no xf32/TF32 on gfx950); 128x128 block tile, BK=32, 4 waves as
+//        2x2, each wave 2x2 MFMA tiles of 32x32 (64 acc VGPRs/lane).
+//   f64: v_mfma_f64_16x16x4_f64  — 64x64 block tile, BK=16, 4 waves as
+//        2x2, each wave 2x2 MFMA tiles of 16x16.
+// A is staged to LDS TRANSPOSED ([k][row], +1 element row pad so the
+// k-major writes and the row-major fragment reads are conflict-free);
+// B is staged linear [k][col] (coalesced both ways).
+// Arbitrary M/N/K via zero-padded loads + guarded stores.
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#include "../../include/pam.h"
+
+#define GBLK 256
+
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+typedef double f64x4 __attribute__((ext_vector_type(4)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef double f64x2 __attribute__((ext_vector_type(2)));
+
+static inline int gcheck(hipError_t e) { return (int)e; }
+
+template <typename T> struct GemmCfg;
+template <> struct GemmCfg<float> {
+  static constexpr int TM = 32;   // MFMA tile edge (square)
+  static constexpr int TK = 2;    // K per MFMA
+  static constexpr int BM = 128, BN = 128, BK = 32;
+  using acc_t = f32x16;
+  using vec_t = f32x4;            // 16-B staging vector
+  static constexpr int VW = 4;
+  __device__ static acc_t mfma(float a, float b, acc_t c) {
+    return __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, c, 0, 0, 0);
+  }
+  // C/D lane mapping for 32x32 shapes (dtype-independent on gfx950):
+  // col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+  __device__ static int crow(int lane, int reg) {
+    return (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+  }
+  static constexpr int NREG = 16;
+};
+template <> struct GemmCfg<double> {
+  static constexpr int TM = 16;
+  static constexpr int TK = 4;
+  static constexpr int BM = 64, BN = 64, BK = 16;
+  using acc_t = f64x4;
+  using vec_t = f64x2;
+  static constexpr int VW = 2;
+  __device__ static acc_t mfma(double a, double b, acc_t c) {
+    return __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, c, 0, 0, 0);
+  }
+  // 16x16 shapes: col = lane&15, row = (lane>>4)*4 + reg
+  __device__ static int crow(int lane, int reg) {
+    return (lane >> 4) * 4 + reg;
+  }
+  static constexpr int NREG = 4;
+};
+
+// C = A@B (+C when ACC), A [M,K] lda, B [K,N] ldb, C [M,N] ldc, row-major.
+template <typename T, bool ACC>
+__global__ void __launch_bounds__(GBLK) gemm_kernel(
+    const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
+    int64_t M, int64_t N, int64_t K, int64_t lda, int64_t ldb, int64_t ldc) {
+  using CFG = GemmCfg<T>;
+  constexpr int BM = CFG::BM, BN = CFG::BN, BK = CFG::BK;
+  constexpr int TM = CFG::TM, TK = CFG::TK, VW = CFG::VW;
+  using acc_t = typename CFG::acc_t;
+  using vec_t = typename CFG::vec_t;
+
+  __shared__ T As[BK][BM + 1];  // transposed, padded
+  __shared__ T Bs[BK][BN];      // linear
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;          // 4 waves as 2x2
+  const int wr = wave >> 1, wc = wave & 1;
+  const int li = lane & (TM - 1);     // fragment row/col within MFMA tile
+  const int lk = lane / TM;           // fragment k (0..TK-1)
+
+  const int64_t brow = (int64_t)blockIdx.y * BM;
+  const int64_t bcol = (int64_t)blockIdx.x * BN;
+
+  acc_t acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {};
+
+  for (int64_t k0 = 0; k0 < K; k0 += BK) {
+    // ---- stage A[brow:brow+BM, k0:k0+BK] -> As[k][row] (transposed)
+    constexpr int AV = (BM * BK) / VW / GBLK;  // vector loads per thread
+#pragma unroll
+    for (int e = 0; e < AV; ++e) {
+      const int vi = tid + GBLK * e;
+      const int row = vi / (BK / VW);
+      const int kv = vi % (BK / VW);
+      T vals[VW];
+      const int64_t gr = brow + row;
+      if (gr < M && k0 + (int64_t)kv * VW + VW <= K) {
+        vec_t v = *reinterpret_cast<const vec_t*>(A + gr * lda + k0
+                                                  + (int64_t)kv * VW);
+#pragma unroll
+        for (int j = 0; j < VW; ++j) vals[j] = v[j];
+      } else {
+#pragma unroll
+        for (int j = 0; j < VW; ++j) {
+          const int64_t gk = k0 + kv * VW + j;
+          vals[j] = (gr < M && gk < K) ? A[gr * lda + gk] : (T)0;
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < VW; ++j) As[kv * VW + j][row] = vals[j];
+    }
+    // ---- stage B[k0:k0+BK, bcol:bcol+BN] -> Bs[k][col] (linear)
+    constexpr int BV = (BK * BN) / VW / GBLK;
+#pragma unroll
+    for (int e = 0; e < BV; ++e) {
+      const int vi = tid + GBLK * e;
+      const int kb = vi / (BN / VW);
+      const int nv = vi % (BN / VW);
+      const int64_t gk = k0 + kb;
+      const int64_t gn = bcol + (int64_t)nv * VW;
+      if (gk < K && gn + VW <= N) {
+        *reinterpret_cast<vec_t*>(&Bs[kb][nv * VW]) =
+            *reinterpret_cast<const vec_t*>(B + gk * ldb + gn);
+      } else {
+#pragma unroll
+        for (int j = 0; j < VW; ++j)
+          Bs[kb][nv * VW + j] =
+              (gk < K && gn + j < N) ? B[gk * ldb + gn + j] : (T)0;
+      }
+    }
+    __syncthreads();
+    // ---- MFMA inner loop
+#pragma unroll
+    for (int kk = 0; kk < BK / TK; ++kk) {
+      const int krow = kk * TK + lk;
+      T a0 = As[krow][wr * 2 * TM + li];
+      T a1 = As[krow][wr * 2 * TM + TM + li];
+      T b0 = Bs[krow][wc * 2 * TM + li];
+      T b1 = Bs[krow][wc * 2 * TM + TM + li];
+      acc[0][0] = CFG::mfma(a0, b0, acc[0][0]);
+      acc[0][1] = CFG::mfma(a0, b1, acc[0][1]);
+      acc[1][0] = CFG::mfma(a1, b0, acc[1][0]);
+      acc[1][1] = CFG::mfma(a1, b1, acc[1][1]);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: C/D fragment layout -> global (guarded)
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int nj = 0; nj < 2; ++nj) {
+      const int64_t r0 = brow + wr * 2 * TM + mi * TM;
+      const int64_t c0 = bcol + wc * 2 * TM + nj * TM;
+      const int64_t cc = c0 + li;
+      if (cc >= N) continue;
+#pragma unroll
+      for (int reg = 0; reg < CFG::NREG; ++reg) {
+        const int64_t rr = r0 + CFG::crow(lane, reg);
+        if (rr < M) {
+          if constexpr (ACC)
+            C[rr * ldc + cc] += acc[mi][nj][reg];
+          else
+            C[rr * ldc + cc] = acc[mi][nj][reg];
+        }
+      }
+    }
+}
+
+template <typename T>
+static int gemm_launch(void* stream, const void* A, const void* B, void* C,
+                       int64_t M, int64_t N, int64_t K, int64_t lda,
+                       int64_t ldb, int64_t ldc, int accumulate) {
+  using CFG = GemmCfg<T>;
+  if (M <= 0 || N <= 0 || K < 0 || !A || !B || !C) return PAM_EARG;
+  dim3 grid((uint32_t)((N + CFG::BN - 1) / CFG::BN),
+            (uint32_t)((M + CFG::BM - 1) / CFG::BM));
+  hipStream_t s = (hipStream_t)stream;
+  if (accumulate)
+    hipLaunchKernelGGL((gemm_kernel<T, true>), grid, dim3(GBLK), 0, s,
+                       (const T*)A, (const T*)B, (T*)C, M, N, K, lda, ldb,
+                       ldc);
+  else
+    hipLaunchKernelGGL((gemm_kernel<T, false>), grid, dim3(GBLK), 0, s,
+                       (const T*)A, (const T*)B, (T*)C, M, N, K, lda, ldb,
+                       ldc);
+  return gcheck(hipGetLastError());
+}
+
+extern "C" int pam_gemm(void* stream, const void* A, const void* B, void* C,
+                        int64_t M, int64_t N, int64_t K, int64_t lda,
+                        int64_t ldb, int64_t ldc, int accumulate, int dtype) {
+  if (dtype == PAM_F64)
+    return gemm_launch<double>(stream, A, B, C, M, N, K, lda, ldb, ldc,
+                               accumulate);
+  if (dtype == PAM_F32)
+    return gemm_launch<float>(stream, A, B, C, M, N, K, lda, ldb, ldc,
+                              accumulate);
+  return PAM_EDTYPE;
+}
+
+// ---------------------------------------------------------------------------
+// LDS-tiled transpose: At[c][r] = A[r][c] (for the adjoint's A^T panels,
+// ref MatrixMult.py:416,737 "A.T.conj()"; real dtypes -> plain transpose).
+// 32x32 tiles, +1 pad against bank conflicts.
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void __launch_bounds__(GBLK) transpose_kernel(
+    const T* __restrict__ A, T* __restrict__ At, int64_t nr, int64_t nc) {
+  __shared__ T tile[32][33];
+  const int64_t r0 = (int64_t)blockIdx.y * 32;
+  const int64_t c0 = (int64_t)blockIdx.x * 32;
+  const int tr = threadIdx.x / 32;   // 8 rows of 32 threads
+  const int tc = threadIdx.x % 32;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int64_t r = r0 + tr + 8 * i;
+    if (r < nr && c0 + tc < nc) tile[tr + 8 * i][tc] = A[r * nc + c0 + tc];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int64_t c = c0 + tr + 8 * i;
+    if (c < nc && r0 + tc < nr) At[c * nr + r0 + tc] = tile[tc][tr + 8 * i];
+  }
+}
+
+extern "C" int pam_transpose(void* stream, const void* A, void* At,
+                             int64_t nr, int64_t nc, int dtype) {
+  if (nr <= 0 || nc <= 0 || !A || !At) return PAM_EARG;
+  dim3 grid((uint32_t)((nc + 31) / 32), (uint32_t)((nr + 31) / 32));
+  hipStream_t s = (hipStream_t)stream;
+  if (dtype == PAM_F64) {
+    hipLaunchKernelGGL((transpose_kernel<double>), grid, dim3(GBLK), 0, s,
+                       (const double*)A, (double*)At, nr, nc);
+    return gcheck(hipGetLastError());
+  }
+  if (dtype == PAM_F32) {
+    hipLaunchKernelGGL((transpose_kernel<float>), grid, dim3(GBLK), 0, s,
+                       (const float*)A, (float*)At, nr, nc);
+    return gcheck(hipGetLastError());
+  }
+  return PAM_EDTYPE;
+}
