@@ -58,9 +58,11 @@ template <> struct GemmCfg<double> {
   __device__ static acc_t mfma(double a, double b, acc_t c) {
     return __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, c, 0, 0, 0);
   }
-  // 16x16 shapes: col = lane&15, row = (lane>>4)*4 + reg
+  // v_mfma_f64_16x16x4f64 C/D: col = lane&15, row = (lane>>4) + 4*reg
+  // (probed empirically on gfx950 — scripts/probe_f64_mfma.hip; NOTE this
+  // differs from the bf16 16x16 mapping (lane>>4)*4 + reg)
   __device__ static int crow(int lane, int reg) {
-    return (lane >> 4) * 4 + reg;
+    return (lane >> 4) + 4 * reg;
   }
   static constexpr int NREG = 4;
 };
