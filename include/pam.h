@@ -27,6 +27,8 @@ extern "C" {
 
 #define PAM_F64 0
 #define PAM_F32 1
+#define PAM_C128 2 /* complex: interleaved (re, im) pairs of the base type */
+#define PAM_C64 3
 
 #define PAM_EARG   (-1)  /* bad argument */
 #define PAM_EDTYPE (-2)  /* unsupported dtype */
@@ -69,6 +71,17 @@ int pam_axpy(void* stream, void* y, const void* x, double alpha, int64_t n,
 int pam_xpby(void* stream, void* y, const void* x, double beta, int64_t n,
              int dtype);
 
+/* Complex element-wise ops (interleaved storage).  add/sub/neg/fill and
+ * real-alpha axpy/xpby on complex arrays are the REAL kernels above on the
+ * 2n-float view (the solvers' a,b scalars are real, ref cls_basic.py:389).
+ * These cover the truly complex cases (ref DistributedArray.py:661-683
+ * multiply, :840-854 conj). */
+int pam_cmul(void* stream, void* y, const void* a, const void* b, int64_t n,
+             int dtype);
+int pam_cscale(void* stream, void* y, const void* x, double alpha_re,
+               double alpha_im, int64_t n, int dtype);
+int pam_conj(void* stream, void* y, const void* x, int64_t n, int dtype);
+
 /* ------------------------------------------------------------------ *
  * Reductions (wavefront-shuffle + LDS tree, deterministic tree shape for
  * a given n — fixed partial count, fixed combine order).
@@ -81,9 +94,17 @@ int pam_xpby(void* stream, void* y, const void* x, double beta, int64_t n,
 int pam_dot(void* stream, const void* x, const void* y, int64_t n, void* ws,
             void* out, int dtype);
 
+/* out = sum(x[i] * y[i]) for complex arrays (conjx != 0: vdot,
+ * conj(x)*y); `out` is a 2-element float64 device buffer (re, im) and
+ * `ws` needs 2*pam_reduce_ws_elems() float64 elements.
+ * Replaces the complex branch of DistributedArray.dot (ref :685-717). */
+int pam_cdot(void* stream, const void* x, const void* y, int64_t n,
+             int conjx, void* ws, void* out, int dtype);
+
 /* Local part of DistributedArray.norm / _compute_vector_norm
  * (ref :719-838).  op: 0 = sum(|x^p|) (float_power semantics, ref :786),
- * 1 = max(|x|), 2 = min(|x|), 3 = count_nonzero. */
+ * 1 = max(|x|), 2 = min(|x|), 3 = count_nonzero.  Complex dtypes reduce
+ * over |z| (|z^p| == |z|^p for real p). */
 int pam_norm_local(void* stream, const void* x, int64_t n, int op, double p,
                    void* ws, void* out, int dtype);
 
@@ -151,6 +172,17 @@ int pam_gemm(void* stream, const void* A, const void* B, void* C, int64_t M,
              int accumulate, int dtype);
 int pam_transpose(void* stream, const void* A, void* At, int64_t nr,
                   int64_t nc, int dtype);
+
+/* Batched complex GEMM: for b in [0, batch):
+ *   C_b = op(A_b) @ B_b   with op = N (opa=0) or conjugate-transpose
+ *   (opa=1; then A_b is [K, M]).
+ * A_b = A + b*strideA etc. (strides in ELEMENTS = complex pairs).
+ * The Fredholm1 batched integral kernel (ref signalprocessing/
+ * Fredholm1.py:123 `ncp.matmul(G, x)` and :149-156 adjoint). */
+int pam_cgemm_batched(void* stream, const void* A, const void* B, void* C,
+                      int64_t batch, int64_t M, int64_t N, int64_t K,
+                      int64_t strideA, int64_t strideB, int64_t strideC,
+                      int opa, int dtype);
 
 #ifdef __cplusplus
 }

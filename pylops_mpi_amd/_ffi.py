@@ -12,7 +12,7 @@ _LIB_PATH = os.path.join(os.path.dirname(os.path.abspath(__file__)),
 _lib = None
 _load_error = None
 
-F64, F32 = 0, 1  # PAM_F64 / PAM_F32
+F64, F32, C128, C64 = 0, 1, 2, 3  # PAM_* dtype codes
 
 _SIGS = {
     "pam_version": ([], ctypes.c_int64),
@@ -39,6 +39,22 @@ _SIGS = {
     "pam_dot": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                  ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
                  ctypes.c_int], ctypes.c_int),
+    "pam_cmul": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                  ctypes.c_void_p, ctypes.c_int64, ctypes.c_int],
+                 ctypes.c_int),
+    "pam_cscale": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                    ctypes.c_double, ctypes.c_double, ctypes.c_int64,
+                    ctypes.c_int], ctypes.c_int),
+    "pam_conj": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                  ctypes.c_int64, ctypes.c_int], ctypes.c_int),
+    "pam_cdot": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                  ctypes.c_int64, ctypes.c_int, ctypes.c_void_p,
+                  ctypes.c_void_p, ctypes.c_int], ctypes.c_int),
+    "pam_cgemm_batched": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                           ctypes.c_void_p, ctypes.c_int64, ctypes.c_int64,
+                           ctypes.c_int64, ctypes.c_int64, ctypes.c_int64,
+                           ctypes.c_int64, ctypes.c_int64, ctypes.c_int,
+                           ctypes.c_int], ctypes.c_int),
     "pam_norm_local": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
                         ctypes.c_int, ctypes.c_double, ctypes.c_void_p,
                         ctypes.c_void_p, ctypes.c_int], ctypes.c_int),
@@ -111,4 +127,8 @@ def dtype_code(torch_dtype) -> int:
         return F64
     if torch_dtype == torch.float32:
         return F32
+    if torch_dtype == torch.complex128:
+        return C128
+    if torch_dtype == torch.complex64:
+        return C64
     raise TypeError(f"pam: unsupported dtype {torch_dtype}")

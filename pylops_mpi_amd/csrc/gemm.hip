@@ -213,6 +213,126 @@ extern "C" int pam_gemm(void* stream, const void* A, const void* B, void* C,
 }
 
 // ---------------------------------------------------------------------------
+// batched complex GEMM: C_b = op(A_b) @ B_b (op = N or conj-transpose).
+// The Fredholm1 batched integral kernel (ref signalprocessing/
+// Fredholm1.py:123,149-156).  Interleaved (re,im) storage; 32x32 output
+// tiles, 256 threads (4 outputs/thread), BKC=8 LDS-staged K-steps.
+// Per-slice panels are small (cfg5: 64x256x256 c64), so this is a
+// latency/HBM-bound VALU kernel; the batch dimension fills the chip.
+// ---------------------------------------------------------------------------
+#define BKC 8
+
+template <typename T, bool CT>
+__global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
+    const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
+    int64_t M, int64_t N, int64_t K, int64_t strideA, int64_t strideB,
+    int64_t strideC) {
+  __shared__ T Asr[32][BKC + 1], Asi[32][BKC + 1];
+  __shared__ T Bsr[BKC][33], Bsi[BKC][33];
+  const int64_t b = blockIdx.z;
+  const T* __restrict__ Ab = A + 2 * b * strideA;
+  const T* __restrict__ Bb = B + 2 * b * strideB;
+  T* __restrict__ Cb = C + 2 * b * strideC;
+  const int64_t m0 = (int64_t)blockIdx.y * 32;
+  const int64_t n0 = (int64_t)blockIdx.x * 32;
+  const int tn = threadIdx.x & 31;
+  const int tm = threadIdx.x >> 5;  // 0..7
+  T accr[4] = {}, acci[4] = {};
+  for (int64_t k0 = 0; k0 < K; k0 += BKC) {
+    // stage op(A)[m0:m0+32, k0:k0+BKC]
+    {
+      const int kk = threadIdx.x & (BKC - 1);
+      const int mm = threadIdx.x / BKC;  // 0..31 (256/8)
+      const int64_t gm = m0 + mm;
+      const int64_t gk = k0 + kk;
+      T vr = 0, vi = 0;
+      if (gm < M && gk < K) {
+        if constexpr (CT) {  // op(A)[m][k] = conj(A[k][m]), A is [K, M]
+          const int64_t off = 2 * (gk * M + gm);
+          vr = Ab[off];
+          vi = -Ab[off + 1];
+        } else {             // A is [M, K]
+          const int64_t off = 2 * (gm * K + gk);
+          vr = Ab[off];
+          vi = Ab[off + 1];
+        }
+      }
+      Asr[mm][kk] = vr;
+      Asi[mm][kk] = vi;
+    }
+    // stage B[k0:k0+BKC, n0:n0+32]
+    {
+      const int nn = threadIdx.x & 31;
+      const int kk = threadIdx.x >> 5;  // 0..7
+      const int64_t gk = k0 + kk;
+      const int64_t gn = n0 + nn;
+      T vr = 0, vi = 0;
+      if (gk < K && gn < N) {
+        const int64_t off = 2 * (gk * N + gn);
+        vr = Bb[off];
+        vi = Bb[off + 1];
+      }
+      Bsr[kk][nn] = vr;
+      Bsi[kk][nn] = vi;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < BKC; ++kk) {
+      const T br = Bsr[kk][tn], bi = Bsi[kk][tn];
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        const T ar = Asr[tm + 8 * e][kk], ai = Asi[tm + 8 * e][kk];
+        accr[e] += ar * br - ai * bi;
+        acci[e] += ar * bi + ai * br;
+      }
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int e = 0; e < 4; ++e) {
+    const int64_t gm = m0 + tm + 8 * e;
+    const int64_t gn = n0 + tn;
+    if (gm < M && gn < N) {
+      Cb[2 * (gm * N + gn)] = accr[e];
+      Cb[2 * (gm * N + gn) + 1] = acci[e];
+    }
+  }
+}
+
+template <typename T>
+static int cgemm_launch(void* stream, const void* A, const void* B, void* C,
+                        int64_t batch, int64_t M, int64_t N, int64_t K,
+                        int64_t sA, int64_t sB, int64_t sC, int opa) {
+  if (batch <= 0 || M <= 0 || N <= 0 || K < 0 || !A || !B || !C)
+    return PAM_EARG;
+  dim3 grid((uint32_t)((N + 31) / 32), (uint32_t)((M + 31) / 32),
+            (uint32_t)batch);
+  hipStream_t s = (hipStream_t)stream;
+  if (opa)
+    hipLaunchKernelGGL((cgemm_batched_kernel<T, true>), grid, dim3(GBLK), 0,
+                       s, (const T*)A, (const T*)B, (T*)C, M, N, K, sA, sB,
+                       sC);
+  else
+    hipLaunchKernelGGL((cgemm_batched_kernel<T, false>), grid, dim3(GBLK), 0,
+                       s, (const T*)A, (const T*)B, (T*)C, M, N, K, sA, sB,
+                       sC);
+  return gcheck(hipGetLastError());
+}
+
+extern "C" int pam_cgemm_batched(void* stream, const void* A, const void* B,
+                                 void* C, int64_t batch, int64_t M, int64_t N,
+                                 int64_t K, int64_t strideA, int64_t strideB,
+                                 int64_t strideC, int opa, int dtype) {
+  if (dtype == PAM_C128)
+    return cgemm_launch<double>(stream, A, B, C, batch, M, N, K, strideA,
+                                strideB, strideC, opa);
+  if (dtype == PAM_C64)
+    return cgemm_launch<float>(stream, A, B, C, batch, M, N, K, strideA,
+                               strideB, strideC, opa);
+  return PAM_EDTYPE;
+}
+
+// ---------------------------------------------------------------------------
 // LDS-tiled transpose: At[c][r] = A[r][c] (for the adjoint's A^T panels,
 // ref MatrixMult.py:416,737 "A.T.conj()"; real dtypes -> plain transpose).
 // 32x32 tiles, +1 pad against bank conflicts.

@@ -200,6 +200,67 @@ EW_ENTRY(pam_xpby(void* stream, void* y, const void* x, double beta,
          7, x, nullptr, beta)
 
 // ---------------------------------------------------------------------------
+// complex element-wise (interleaved re,im).  EWC op: 0 cmul, 1 cscale, 2 conj
+// (ref DistributedArray.py:661-683, :840-854).  add/neg/real-axpy on complex
+// arrays reuse the real kernels on the 2n view (solver scalars are real).
+// ---------------------------------------------------------------------------
+template <typename T, int OP>
+__global__ void __launch_bounds__(BLK) ewc_kernel(T* __restrict__ y,
+                                                  const T* __restrict__ a,
+                                                  const T* __restrict__ b,
+                                                  T ar, T ai, int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    const T xr = a[2 * i], xi = a[2 * i + 1];
+    T zr, zi;
+    if constexpr (OP == 0) {
+      const T br = b[2 * i], bi = b[2 * i + 1];
+      zr = xr * br - xi * bi;
+      zi = xr * bi + xi * br;
+    } else if constexpr (OP == 1) {
+      zr = xr * ar - xi * ai;
+      zi = xr * ai + xi * ar;
+    } else {
+      zr = xr;
+      zi = -xi;
+    }
+    y[2 * i] = zr;
+    y[2 * i + 1] = zi;
+  }
+}
+
+template <typename T, int OP>
+static int ewc_launch(void* stream, void* y, const void* a, const void* b,
+                      double ar, double ai, int64_t n) {
+  if (n < 0) return PAM_EARG;
+  if (n == 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL((ewc_kernel<T, OP>), dim3(grid_1d(n)), dim3(BLK), 0, s,
+                     (T*)y, (const T*)a, (const T*)b, (T)ar, (T)ai, n);
+  return check(hipGetLastError());
+}
+
+#define EWC_ENTRY(name, OP, B, AR, AI)                                        \
+  extern "C" int name {                                                       \
+    if (dtype == PAM_C128)                                                    \
+      return ewc_launch<double, OP>(stream, y, a, B, AR, AI, n);              \
+    if (dtype == PAM_C64)                                                     \
+      return ewc_launch<float, OP>(stream, y, a, B, AR, AI, n);               \
+    return PAM_EDTYPE;                                                        \
+  }
+
+EWC_ENTRY(pam_cmul(void* stream, void* y, const void* a, const void* b,
+                   int64_t n, int dtype),
+          0, b, 0.0, 0.0)
+EWC_ENTRY(pam_cscale(void* stream, void* y, const void* a, double alpha_re,
+                     double alpha_im, int64_t n, int dtype),
+          1, nullptr, alpha_re, alpha_im)
+EWC_ENTRY(pam_conj(void* stream, void* y, const void* a, int64_t n,
+                   int dtype),
+          2, nullptr, 0.0, 0.0)
+
+// ---------------------------------------------------------------------------
 // reductions (ref DistributedArray.py:685-717 dot, :719-838 norms)
 // RED codes: 0 dot, 1 powsum |x^p|, 2 max|x|, 3 min|x|, 4 count_nonzero
 // ---------------------------------------------------------------------------
@@ -216,7 +277,7 @@ template <int RED> __device__ __forceinline__ double red_init() {
   else return 0.0;
 }
 
-template <typename T, int RED>
+template <typename T, int RED, bool CPLX = false>
 __global__ void __launch_bounds__(BLK) reduce_stage1(
     const T* __restrict__ x, const T* __restrict__ y, int64_t n, double p,
     double* __restrict__ partials) {
@@ -224,12 +285,20 @@ __global__ void __launch_bounds__(BLK) reduce_stage1(
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
-    const double xv = (double)x[i];
     double v;
-    if constexpr (RED == 0) v = xv * (double)y[i];
-    else if constexpr (RED == 1) v = fabs(pow(xv, p));  // float_power, ref :786
-    else if constexpr (RED == 2 || RED == 3) v = fabs(xv);
-    else v = (xv != 0.0) ? 1.0 : 0.0;
+    if constexpr (CPLX) {
+      // reduce over |z| (norms only; |z^p| == |z|^p for real p)
+      const double zr = (double)x[2 * i], zi = (double)x[2 * i + 1];
+      if constexpr (RED == 1) v = pow(hypot(zr, zi), p);
+      else if constexpr (RED == 2 || RED == 3) v = hypot(zr, zi);
+      else v = (zr != 0.0 || zi != 0.0) ? 1.0 : 0.0;
+    } else {
+      const double xv = (double)x[i];
+      if constexpr (RED == 0) v = xv * (double)y[i];
+      else if constexpr (RED == 1) v = fabs(pow(xv, p));  // ref :786
+      else if constexpr (RED == 2 || RED == 3) v = fabs(xv);
+      else v = (xv != 0.0) ? 1.0 : 0.0;
+    }
     acc = red_combine<RED>(acc, v);
   }
   // 64-lane wavefront shuffle reduction
@@ -269,7 +338,7 @@ __global__ void __launch_bounds__(BLK) reduce_stage2(
   }
 }
 
-template <typename T, int RED>
+template <typename T, int RED, bool CPLX = false>
 static int reduce_launch(void* stream, const void* x, const void* y, int64_t n,
                          double p, void* ws, void* out) {
   if (n < 0 || !x || !ws || !out) return PAM_EARG;
@@ -282,13 +351,119 @@ static int reduce_launch(void* stream, const void* x, const void* y, int64_t n,
   (void)s;
   int64_t nb = (n + BLK - 1) / BLK;
   if (nb > NPARTIAL) nb = NPARTIAL;
-  hipLaunchKernelGGL((reduce_stage1<T, RED>), dim3(nb), dim3(BLK), 0, s,
+  hipLaunchKernelGGL((reduce_stage1<T, RED, CPLX>), dim3(nb), dim3(BLK), 0, s,
                      (const T*)x, (const T*)y, n, p, (double*)ws);
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) return (int)e;
   hipLaunchKernelGGL((reduce_stage2<RED>), dim3(1), dim3(BLK), 0, s,
                      (const double*)ws, nb, (double*)out);
   return check(hipGetLastError());
+}
+
+// ---- complex dot (out = sum x*y or sum conj(x)*y; 2-double result)
+template <typename T, bool CONJX>
+__global__ void __launch_bounds__(BLK) cdot_stage1(
+    const T* __restrict__ x, const T* __restrict__ y, int64_t n,
+    double* __restrict__ pre, double* __restrict__ pim) {
+  double ar = 0.0, ai = 0.0;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    const double xr = (double)x[2 * i];
+    const double xi = CONJX ? -(double)x[2 * i + 1] : (double)x[2 * i + 1];
+    const double yr = (double)y[2 * i], yi = (double)y[2 * i + 1];
+    ar += xr * yr - xi * yi;
+    ai += xr * yi + xi * yr;
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    ar += __shfl_down(ar, off, 64);
+    ai += __shfl_down(ai, off, 64);
+  }
+  __shared__ double lr[BLK / 64], li[BLK / 64];
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) {
+    lr[wave] = ar;
+    li[wave] = ai;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double sr = lr[0], si = li[0];
+#pragma unroll
+    for (int w = 1; w < BLK / 64; ++w) {
+      sr += lr[w];
+      si += li[w];
+    }
+    pre[blockIdx.x] = sr;
+    pim[blockIdx.x] = si;
+  }
+}
+
+__global__ void __launch_bounds__(BLK) cdot_stage2(
+    const double* __restrict__ pre, const double* __restrict__ pim,
+    int64_t np, double* __restrict__ out) {
+  double ar = 0.0, ai = 0.0;
+  for (int64_t i = threadIdx.x; i < np; i += BLK) {
+    ar += pre[i];
+    ai += pim[i];
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    ar += __shfl_down(ar, off, 64);
+    ai += __shfl_down(ai, off, 64);
+  }
+  __shared__ double lr[BLK / 64], li[BLK / 64];
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) {
+    lr[wave] = ar;
+    li[wave] = ai;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double sr = lr[0], si = li[0];
+#pragma unroll
+    for (int w = 1; w < BLK / 64; ++w) {
+      sr += lr[w];
+      si += li[w];
+    }
+    out[0] = sr;
+    out[1] = si;
+  }
+}
+
+template <typename T>
+static int cdot_launch(void* stream, const void* x, const void* y, int64_t n,
+                       int conjx, void* ws, void* out) {
+  if (n < 0 || !x || !y || !ws || !out) return PAM_EARG;
+  hipStream_t s = (hipStream_t)stream;
+  if (n == 0) {
+    double z[2] = {0.0, 0.0};
+    return check(hipMemcpy(out, z, sizeof(z), hipMemcpyHostToDevice));
+  }
+  int64_t nb = (n + BLK - 1) / BLK;
+  if (nb > NPARTIAL) nb = NPARTIAL;
+  double* pre = (double*)ws;
+  double* pim = pre + NPARTIAL;
+  if (conjx)
+    hipLaunchKernelGGL((cdot_stage1<T, true>), dim3(nb), dim3(BLK), 0, s,
+                       (const T*)x, (const T*)y, n, pre, pim);
+  else
+    hipLaunchKernelGGL((cdot_stage1<T, false>), dim3(nb), dim3(BLK), 0, s,
+                       (const T*)x, (const T*)y, n, pre, pim);
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) return (int)e;
+  hipLaunchKernelGGL(cdot_stage2, dim3(1), dim3(BLK), 0, s, pre, pim, nb,
+                     (double*)out);
+  return check(hipGetLastError());
+}
+
+extern "C" int pam_cdot(void* stream, const void* x, const void* y, int64_t n,
+                        int conjx, void* ws, void* out, int dtype) {
+  if (dtype == PAM_C128)
+    return cdot_launch<double>(stream, x, y, n, conjx, ws, out);
+  if (dtype == PAM_C64)
+    return cdot_launch<float>(stream, x, y, n, conjx, ws, out);
+  return PAM_EDTYPE;
 }
 
 extern "C" int pam_dot(void* stream, const void* x, const void* y, int64_t n,
@@ -314,6 +489,16 @@ extern "C" int pam_norm_local(void* stream, const void* x, int64_t n, int op,
   if (dtype == PAM_F64) { NORM_CASE(double) }
   if (dtype == PAM_F32) { NORM_CASE(float) }
 #undef NORM_CASE
+#define CNORM_CASE(T)                                                         \
+  switch (op) {                                                               \
+    case 0: return reduce_launch<T, 1, true>(stream, x, nullptr, n, p, ws, out); \
+    case 1: return reduce_launch<T, 2, true>(stream, x, nullptr, n, p, ws, out); \
+    case 2: return reduce_launch<T, 3, true>(stream, x, nullptr, n, p, ws, out); \
+    default: return reduce_launch<T, 4, true>(stream, x, nullptr, n, p, ws, out); \
+  }
+  if (dtype == PAM_C128) { CNORM_CASE(double) }
+  if (dtype == PAM_C64) { CNORM_CASE(float) }
+#undef CNORM_CASE
   return PAM_EDTYPE;
 }
 

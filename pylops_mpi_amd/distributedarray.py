@@ -29,6 +29,8 @@ from .comm import PamComm, get_default_comm
 _TORCH_DTYPES = {
     np.dtype(np.float64): torch.float64,
     np.dtype(np.float32): torch.float32,
+    np.dtype(np.complex128): torch.complex128,
+    np.dtype(np.complex64): torch.complex64,
 }
 _NP_DTYPES = {v: k for k, v in _TORCH_DTYPES.items()}
 
@@ -69,8 +71,8 @@ def _reduce_buffers(device):
     if key not in _red_scratch:
         n = int(_ffi.lib().pam_reduce_ws_elems())
         _red_scratch[key] = (
-            torch.empty(n, dtype=torch.float64, device=device),
-            torch.empty(1, dtype=torch.float64, device=device),
+            torch.empty(2 * n, dtype=torch.float64, device=device),
+            torch.empty(2, dtype=torch.float64, device=device),
         )
     return _red_scratch[key]
 
@@ -218,6 +220,20 @@ class DistributedArray:
     def _flat(self) -> torch.Tensor:
         t = self._local_array
         return t.reshape(-1) if t.is_contiguous() else t.contiguous().view(-1)
+
+    def _is_cplx(self) -> bool:
+        return self._local_array.is_complex()
+
+    def _ew(self, t: Optional[torch.Tensor] = None):
+        """(flat, count, dtype_code) for element-wise kernels; complex
+        arrays are handled as their 2n-float real view (valid for
+        add/sub/neg/fill and REAL-alpha axpy/xpby/scale)."""
+        t = self._local_array if t is None else t
+        flat = t.reshape(-1) if t.is_contiguous() else t.contiguous().view(-1)
+        if t.is_complex():
+            rv = torch.view_as_real(flat).reshape(-1)
+            return rv, rv.numel(), _ffi.dtype_code(rv.dtype)
+        return flat, flat.numel(), _ffi.dtype_code(t.dtype)
 
     def _like(self, local: Optional[torch.Tensor] = None) -> "DistributedArray":
         return DistributedArray(self._global_shape, self._base_comm,
@@ -393,9 +409,10 @@ class DistributedArray:
     def __neg__(self):
         self._require_compute()
         out = self._like()
+        a, n, dt = self._ew()
+        o, _, _ = self._ew(out._local_array)
         _ffi.checked(_ffi.lib().pam_neg(
-            self._stream(), out._flat().data_ptr(), self._flat().data_ptr(),
-            self._local_array.numel(), self._dt()), "neg")
+            self._stream(), o.data_ptr(), a.data_ptr(), n, dt), "neg")
         return out
 
     def add(self, other: "DistributedArray") -> "DistributedArray":
@@ -403,9 +420,11 @@ class DistributedArray:
         self._check_partition_shape(other)
         self._require_compute()
         out = self._like()
+        a, n, dt = self._ew()
+        b, _, _ = self._ew(other._local_array)
+        o, _, _ = self._ew(out._local_array)
         _ffi.checked(_ffi.lib().pam_add(
-            self._stream(), out._flat().data_ptr(), self._flat().data_ptr(),
-            other._flat().data_ptr(), self._local_array.numel(), self._dt()),
+            self._stream(), o.data_ptr(), a.data_ptr(), b.data_ptr(), n, dt),
             "add")
         return out
 
@@ -413,9 +432,10 @@ class DistributedArray:
         # ref :653-659
         self._check_partition_shape(other)
         self._require_compute()
+        a, n, dt = self._ew()
+        b, _, _ = self._ew(other._local_array)
         _ffi.checked(_ffi.lib().pam_axpy(
-            self._stream(), self._flat().data_ptr(), other._flat().data_ptr(),
-            1.0, self._local_array.numel(), self._dt()), "iadd")
+            self._stream(), a.data_ptr(), b.data_ptr(), 1.0, n, dt), "iadd")
         return self
 
     def sub(self, other: "DistributedArray") -> "DistributedArray":
@@ -423,9 +443,11 @@ class DistributedArray:
         self._check_partition_shape(other)
         self._require_compute()
         out = self._like()
+        a, n, dt = self._ew()
+        b, _, _ = self._ew(other._local_array)
+        o, _, _ = self._ew(out._local_array)
         _ffi.checked(_ffi.lib().pam_sub(
-            self._stream(), out._flat().data_ptr(), self._flat().data_ptr(),
-            other._flat().data_ptr(), self._local_array.numel(), self._dt()),
+            self._stream(), o.data_ptr(), a.data_ptr(), b.data_ptr(), n, dt),
             "sub")
         return out
 
@@ -435,36 +457,55 @@ class DistributedArray:
         out = self._like()
         if isinstance(x, DistributedArray):
             self._check_partition_shape(x)
-            _ffi.checked(_ffi.lib().pam_mul(
+            if self._is_cplx():
+                _ffi.checked(_ffi.lib().pam_cmul(
+                    self._stream(), out._flat().data_ptr(),
+                    self._flat().data_ptr(), x._flat().data_ptr(),
+                    self._local_array.numel(), self._dt()), "cmul")
+            else:
+                _ffi.checked(_ffi.lib().pam_mul(
+                    self._stream(), out._flat().data_ptr(),
+                    self._flat().data_ptr(), x._flat().data_ptr(),
+                    self._local_array.numel(), self._dt()), "mul")
+        elif self._is_cplx() and isinstance(x, complex) and x.imag != 0.0:
+            _ffi.checked(_ffi.lib().pam_cscale(
                 self._stream(), out._flat().data_ptr(),
-                self._flat().data_ptr(), x._flat().data_ptr(),
-                self._local_array.numel(), self._dt()), "mul")
+                self._flat().data_ptr(), float(x.real), float(x.imag),
+                self._local_array.numel(), self._dt()), "cscale")
         else:
+            # real scalar: componentwise on the (possibly complex) view
+            a, n, dt = self._ew()
+            o, _, _ = self._ew(out._local_array)
             _ffi.checked(_ffi.lib().pam_scale(
-                self._stream(), out._flat().data_ptr(),
-                self._flat().data_ptr(), float(x),
-                self._local_array.numel(), self._dt()), "scale")
+                self._stream(), o.data_ptr(), a.data_ptr(),
+                float(np.real(x)), n, dt), "scale")
         return out
 
     # fused solver updates (not in the reference surface — the reference
     # allocates 2 temporaries per axpy, ref cls_basic.py:390-391 via
     # :618-683; these are the one-pass HIP equivalents)
     def iaxpy_(self, alpha: float, x: "DistributedArray") -> "DistributedArray":
-        """self += alpha * x, fused."""
+        """self += alpha * x (REAL alpha), fused — componentwise on the
+        real view for complex arrays (CG/CGLS scalars are real,
+        ref cls_basic.py:389-395)."""
         self._check_partition_shape(x)
         self._require_compute()
+        a, n, dt = self._ew()
+        b, _, _ = self._ew(x._local_array)
         _ffi.checked(_ffi.lib().pam_axpy(
-            self._stream(), self._flat().data_ptr(), x._flat().data_ptr(),
-            float(alpha), self._local_array.numel(), self._dt()), "axpy")
+            self._stream(), a.data_ptr(), b.data_ptr(), float(alpha), n, dt),
+            "axpy")
         return self
 
     def xpby_(self, x: "DistributedArray", beta: float) -> "DistributedArray":
-        """self = x + beta * self, fused (CGLS c = r + b*c)."""
+        """self = x + beta * self (REAL beta), fused (CGLS c = r + b*c)."""
         self._check_partition_shape(x)
         self._require_compute()
+        a, n, dt = self._ew()
+        b, _, _ = self._ew(x._local_array)
         _ffi.checked(_ffi.lib().pam_xpby(
-            self._stream(), self._flat().data_ptr(), x._flat().data_ptr(),
-            float(beta), self._local_array.numel(), self._dt()), "xpby")
+            self._stream(), a.data_ptr(), b.data_ptr(), float(beta), n, dt),
+            "xpby")
         return self
 
     def __add__(self, x):
@@ -496,12 +537,21 @@ class DistributedArray:
             y = DistributedArray.to_dist(other._local_array, self._base_comm)
             return x.dot(y, vdot=vdot)
         ws, out = _reduce_buffers(self._local_array.device)
+        if self._is_cplx():
+            _ffi.checked(_ffi.lib().pam_cdot(
+                self._stream(), self._flat().data_ptr(),
+                other._flat().data_ptr(), self._local_array.numel(),
+                1 if vdot else 0, ws.data_ptr(), out.data_ptr(),
+                self._dt()), "cdot")
+            self._base_comm.allreduce_(out, "sum")
+            v = out.cpu()
+            return np.complex128(complex(float(v[0]), float(v[1])))
         _ffi.checked(_ffi.lib().pam_dot(
             self._stream(), self._flat().data_ptr(), other._flat().data_ptr(),
             self._local_array.numel(), ws.data_ptr(), out.data_ptr(),
             self._dt()), "dot")
-        self._base_comm.allreduce_(out, "sum")
-        return np.float64(out.item())
+        self._base_comm.allreduce_(out[:1], "sum")
+        return np.float64(out[0].item())
 
     def _norm_local(self, op: int, p: float) -> torch.Tensor:
         ws, out = _reduce_buffers(self._local_array.device)
@@ -509,7 +559,7 @@ class DistributedArray:
             self._stream(), self._flat().data_ptr(),
             self._local_array.numel(), op, p, ws.data_ptr(), out.data_ptr(),
             self._dt()), "norm")
-        return out
+        return out[:1]
 
     def norm(self, ord: Optional[float] = None,
              axis: Optional[int] = None):
@@ -542,8 +592,15 @@ class DistributedArray:
 
     # ------------------------------------------------------------ structure
     def conj(self):
-        # real dtypes: conj == copy (complex deferred with Fredholm round)
-        return self.copy()
+        # ref :840-854
+        if not self._is_cplx():
+            return self.copy()
+        self._require_compute()
+        out = self._like()
+        _ffi.checked(_ffi.lib().pam_conj(
+            self._stream(), out._flat().data_ptr(), self._flat().data_ptr(),
+            self._local_array.numel(), self._dt()), "conj")
+        return out
 
     def copy(self):
         return self._like(self._local_array.clone())

@@ -168,7 +168,39 @@ def body_blockdiag(c):
 @pytest.mark.parametrize("body", [
     "body_allreduce", "body_allgather_obj", "body_allgather_tensors",
     "body_to_dist_asarray", "body_ghost_cells", "body_rebalance",
-    "body_sendrecv", "body_blockdiag",
+    "body_sendrecv", "body_blockdiag", "body_fredholm",
 ])
 def test_gloo_world2(body):
     _spawn(body)
+
+
+def body_fredholm(c):
+    """MPIFredholm1 slice/allgather logic across 2 ranks vs the oracle
+    (torch-matmul test adapter for the GPU-only batched kernel)."""
+    import numpy as np
+    import pylops_mpi_amd as pm
+    from oracle.fredholm import SimFredholm1
+    from oracle.ranksim import Partition as SP, SimArray
+    rng = np.random.default_rng(60)
+    nsl, nx, ny, nz = 6, 3, 4, 2
+    G = (rng.standard_normal((nsl, nx, ny))
+         + 1j * rng.standard_normal((nsl, nx, ny)))
+    blocks = [G[:3].copy(), G[3:].copy()]
+    op = pm.MPIFredholm1(torch.as_tensor(blocks[c.rank]), nz=nz,
+                         saveGt=False, base_comm=c, dtype="complex128")
+    op._batched = lambda A, X, opa: (
+        A.conj().transpose(1, 2) @ X if opa else A @ X)
+    sop = SimFredholm1(blocks, nz=nz, saveGt=False)
+    assert op.shape == sop.shape
+    x = rng.standard_normal(op.shape[1]) + 0j
+    y = rng.standard_normal(op.shape[0]) + 0j
+    xd = pm.DistributedArray.to_dist(torch.as_tensor(x), base_comm=c,
+                                     partition=pm.Partition.BROADCAST)
+    yd = pm.DistributedArray.to_dist(torch.as_tensor(y), base_comm=c,
+                                     partition=pm.Partition.BROADCAST)
+    bx = SimArray([x.copy(), x.copy()], x.shape, partition=SP.BROADCAST)
+    by = SimArray([y.copy(), y.copy()], y.shape, partition=SP.BROADCAST)
+    np.testing.assert_allclose(op.matvec(xd).local_array.numpy(),
+                               sop.matvec(bx).locals[c.rank], rtol=1e-12)
+    np.testing.assert_allclose(op.rmatvec(yd).local_array.numpy(),
+                               sop.rmatvec(by).locals[c.rank], rtol=1e-12)
