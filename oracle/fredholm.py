@@ -80,7 +80,10 @@ class SimFredholm1:
 
 # --------------------------------------------------------------- serial FFT
 def serial_rfft_op(x: np.ndarray, nt: int, ifftshift_before=False):
-    """Forward of the re-derived pylops real-FFT convention (axis 0)."""
+    """Forward of the re-derived pylops real-FFT convention (axis 0).
+    Complex inputs take the real part (real FFT of a real model that may
+    be carried in complex storage by the solver)."""
+    x = np.asarray(x).real
     if ifftshift_before:
         x = np.fft.ifftshift(x, axes=0)
     y = np.fft.rfft(x, n=nt, axis=0, norm="ortho")

@@ -19,8 +19,8 @@ def dottest(Op, u: DistributedArray, v: DistributedArray,
         raise AssertionError("Provided nr and nc do not match operator shape")
     y = Op.matvec(u)
     x = Op.rmatvec(v)
-    yy = float(y.dot(v, vdot=True))
-    xx = float(u.dot(x, vdot=True))
+    yy = y.dot(v, vdot=True)
+    xx = u.dot(x, vdot=True)
     passed = bool(np.isclose(xx, yy, rtol, atol))
     if (not passed and raiseerror) or verb:
         status = "passed" if passed else "failed"
