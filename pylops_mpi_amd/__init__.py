@@ -18,7 +18,12 @@ from .fftlocal import FFTLocal, IdentityLocal  # noqa: F401
 from .localops import DenseLocal, CallableLocal, LocalOperator  # noqa: F401
 from .derivative import (MPIFirstDerivative,  # noqa: F401
                          MPISecondDerivative)
-from .solvers import CG, CGLS, cg, cgls  # noqa: F401
+from .solvers import CG, CGLS, cg, cgls, power_iteration  # noqa: F401
+from .stacked import (StackedDistributedArray,  # noqa: F401
+                      MPIStackedLinearOperator)
+from .vstack import (MPIVStack, MPIHStack,  # noqa: F401
+                     MPIStackedVStack, MPIStackedBlockDiag)
+from .localops import AdjointLocal  # noqa: F401
 from .dottest import dottest  # noqa: F401
 from .comm import (PamComm, get_default_comm,  # noqa: F401
                    init_default_comm)

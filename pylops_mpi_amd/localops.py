@@ -78,6 +78,21 @@ class DenseLocal(LocalOperator):
         return self._gemv(x.reshape(-1), 1)
 
 
+class AdjointLocal(LocalOperator):
+    """Adjoint view of a local operator (swap matvec/rmatvec)."""
+
+    def __init__(self, op: LocalOperator):
+        self.op = op
+        self.shape = (op.shape[1], op.shape[0])
+        self.dtype = op.dtype
+
+    def matvec(self, x):
+        return self.op.rmatvec(x)
+
+    def rmatvec(self, x):
+        return self.op.matvec(x)
+
+
 class CallableLocal(LocalOperator):
     """Adapter for tests/composition: wrap a (matvec, rmatvec) pair."""
 

@@ -169,3 +169,42 @@ def cgls(Op, y, x0, niter: int = 10, damp: float = 0.0, tol: float = 1e-4,
         y=y, x0=x0, niter=niter, damp=damp, tol=tol, show=show,
         itershow=itershow)
     return x, istop, iiter, r1norm, r2norm, cost
+
+
+def power_iteration(Op, b_k, niter: int = 10, tol: float = 1e-5,
+                    dtype="float64"):
+    """Largest-eigenpair power iteration,
+    ref optimization/eigs.py:10-100 (random re-init of b_k, vdot Rayleigh
+    quotient, renormalization via the HIP scale kernel)."""
+    import torch
+
+    from .stacked import StackedDistributedArray
+
+    cmpx = np.issubdtype(np.dtype(dtype), np.complexfloating)
+
+    def _randomize(d):
+        t = torch.rand(d.local_shape, dtype=torch.float64,
+                       device=d.local_array.device)
+        if cmpx:
+            t = t + 1j * torch.rand(d.local_shape, dtype=torch.float64,
+                                    device=d.local_array.device)
+        d[:] = t.to(d.local_array.dtype)
+
+    if isinstance(b_k, StackedDistributedArray):
+        for d in b_k.distarrays:
+            _randomize(d)
+    else:
+        _randomize(b_k)
+    b_k = b_k * (1.0 / float(b_k.norm()))
+    maxeig_old = 0.0
+    iiter = 0
+    for iiter in range(niter):
+        b1_k = Op.matvec(b_k)
+        maxeig = b_k.dot(b1_k, vdot=True)
+        maxeig = complex(maxeig) if cmpx else float(maxeig)
+        b1_k_norm = float(b1_k.norm())
+        b_k = b1_k * (1.0 / b1_k_norm)
+        if abs(maxeig - maxeig_old) < tol * abs(maxeig):
+            break
+        maxeig_old = maxeig
+    return maxeig, b_k, iiter + 1

@@ -1,0 +1,134 @@
+"""GPU parity: StackedDistributedArray math, MPIVStack/MPIHStack,
+stacked operators driving CG/CGLS, and power_iteration."""
+import numpy as np
+import pytest
+import torch
+from numpy.testing import assert_allclose
+
+import pylops_mpi_amd as pm
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _init():
+    from pylops_mpi_amd.comm import init_default_comm
+    init_default_comm(torch.device("cuda:0"))
+
+
+def dev(a):
+    return torch.as_tensor(a, device="cuda:0")
+
+
+def host(t):
+    return t.cpu().numpy()
+
+
+def test_stacked_array_math():
+    rng = np.random.default_rng(0)
+    a1, a2 = rng.standard_normal(40), rng.standard_normal(25)
+    b1, b2 = rng.standard_normal(40), rng.standard_normal(25)
+    A = pm.StackedDistributedArray([pm.DistributedArray.to_dist(dev(a1)),
+                                    pm.DistributedArray.to_dist(dev(a2))])
+    B = pm.StackedDistributedArray([pm.DistributedArray.to_dist(dev(b1)),
+                                    pm.DistributedArray.to_dist(dev(b2))])
+    ca, cb = np.concatenate([a1, a2]), np.concatenate([b1, b2])
+    assert_allclose(host((A + B).asarray()), ca + cb, rtol=1e-14)
+    assert_allclose(host((A - B).asarray()), ca - cb, rtol=1e-14)
+    assert_allclose(host((2.0 * A).asarray()), 2 * ca, rtol=1e-14)
+    assert_allclose(host((-A).asarray()), -ca, rtol=1e-14)
+    assert_allclose(A.dot(B), np.dot(ca, cb), rtol=1e-13)
+    assert_allclose(float(A.norm()), np.linalg.norm(ca), rtol=1e-13)
+    assert_allclose(float(A.norm(np.inf)), np.abs(ca).max(), rtol=1e-14)
+    C = A.copy()
+    C.iaxpy_(0.5, B)
+    assert_allclose(host(C.asarray()), ca + 0.5 * cb, rtol=1e-14)
+    C = A.copy()
+    C.xpby_(B, 0.25)
+    assert_allclose(host(C.asarray()), cb + 0.25 * ca, rtol=1e-14)
+
+
+def test_vstack_hstack_vs_dense():
+    rng = np.random.default_rng(1)
+    A1 = rng.standard_normal((7, 10))
+    A2 = rng.standard_normal((5, 10))
+    op = pm.MPIVStack([pm.DenseLocal(dev(A1)), pm.DenseLocal(dev(A2))])
+    D = np.vstack([A1, A2])
+    assert op.shape == D.shape
+    x = rng.standard_normal(10)
+    y = rng.standard_normal(12)
+    xd = pm.DistributedArray.to_dist(dev(x),
+                                     partition=pm.Partition.BROADCAST)
+    yd = pm.DistributedArray.to_dist(dev(y))
+    assert_allclose(host(op.matvec(xd).asarray()), D @ x, rtol=1e-12)
+    assert_allclose(host(op.rmatvec(yd).asarray()), D.T @ y, rtol=1e-12)
+    u = pm.DistributedArray.to_dist(dev(rng.standard_normal(10)),
+                                    partition=pm.Partition.BROADCAST)
+    v = pm.DistributedArray.to_dist(dev(rng.standard_normal(12)))
+    assert pm.dottest(op, u, v, rtol=1e-10)
+    # HStack = adjoint flip
+    H1, H2 = rng.standard_normal((6, 4)), rng.standard_normal((6, 3))
+    hop = pm.MPIHStack([pm.DenseLocal(dev(H1)), pm.DenseLocal(dev(H2))])
+    Dh = np.hstack([H1, H2])
+    xs = rng.standard_normal(7)
+    ys = rng.standard_normal(6)
+    xsd = pm.DistributedArray.to_dist(dev(xs))
+    ysd = pm.DistributedArray.to_dist(dev(ys),
+                                      partition=pm.Partition.BROADCAST)
+    assert_allclose(host(hop.matvec(xsd).asarray()), Dh @ xs, rtol=1e-12)
+    assert_allclose(host(hop.rmatvec(ysd).asarray()), Dh.T @ ys, rtol=1e-12)
+
+
+def test_stacked_vstack_cgls():
+    """CGLS on a vertical stack of MPI operators (ref test_solver-style
+    overdetermined system), vs the dense normal-equation solve."""
+    rng = np.random.default_rng(2)
+    n = 16
+    A1 = rng.standard_normal((n, n)) + 4 * np.eye(n)
+    A2 = rng.standard_normal((n, n))
+    op1 = pm.MPIBlockDiag([pm.DenseLocal(dev(A1))])
+    op2 = pm.MPIBlockDiag([pm.DenseLocal(dev(A2))])
+    vop = pm.MPIStackedVStack([op1, op2])
+    D = np.vstack([A1, A2])
+    yg = rng.standard_normal(2 * n)
+    y = pm.StackedDistributedArray(
+        [pm.DistributedArray.to_dist(dev(yg[:n])),
+         pm.DistributedArray.to_dist(dev(yg[n:]))])
+    x0 = pm.DistributedArray((n,))
+    x0[:] = 0.0
+    xs, _, _, _, _, _ = pm.cgls(vop, y, x0, niter=120, tol=1e-30)
+    want = np.linalg.lstsq(D, yg, rcond=None)[0]
+    assert_allclose(host(xs.asarray()), want, rtol=1e-8, atol=1e-9)
+
+
+def test_stacked_blockdiag_cg():
+    rng = np.random.default_rng(3)
+    n1, n2 = 12, 9
+    S1 = rng.standard_normal((n1, n1))
+    S1 = S1 @ S1.T + n1 * np.eye(n1)
+    S2 = rng.standard_normal((n2, n2))
+    S2 = S2 @ S2.T + n2 * np.eye(n2)
+    op = pm.MPIStackedBlockDiag(
+        [pm.MPIBlockDiag([pm.DenseLocal(dev(S1))]),
+         pm.MPIBlockDiag([pm.DenseLocal(dev(S2))])])
+    yg = rng.standard_normal(n1 + n2)
+    y = pm.StackedDistributedArray(
+        [pm.DistributedArray.to_dist(dev(yg[:n1])),
+         pm.DistributedArray.to_dist(dev(yg[n1:]))])
+    x0 = y.zeros_like()
+    xs, _, _ = pm.cg(op, y, x0, niter=150, tol=1e-30)
+    want = np.concatenate([np.linalg.solve(S1, yg[:n1]),
+                           np.linalg.solve(S2, yg[n1:])])
+    assert_allclose(host(xs.asarray()), want, rtol=1e-8, atol=1e-9)
+
+
+def test_power_iteration():
+    rng = np.random.default_rng(4)
+    n = 24
+    S = rng.standard_normal((n, n))
+    S = S @ S.T  # SPD: dominant eigenvalue = largest
+    op = pm.MPIBlockDiag([pm.DenseLocal(dev(S))])
+    b0 = pm.DistributedArray((n,))
+    maxeig, b, it = pm.power_iteration(op, b0, niter=400, tol=1e-12)
+    want = np.linalg.eigvalsh(S).max()
+    assert_allclose(maxeig, want, rtol=1e-6)
