@@ -30,6 +30,7 @@
 #include <hip/hip_runtime.h>
 #include <math.h>
 #include <stdint.h>
+#include <stdlib.h>
 
 #include "../../include/pam.h"
 
@@ -49,12 +50,22 @@ template <> struct VecW<float> { static constexpr int value = 4; };
 
 template <typename T, int V> struct VecT;
 template <> struct VecT<double, 2> { using type = double2; };
+template <> struct VecT<double, 4> { using type = double4; };
+template <> struct VecT<float, 2> { using type = float2; };
 template <> struct VecT<float, 4> { using type = float4; };
 
 template <typename T, int V>
 __device__ __forceinline__ void loadv(const T* __restrict__ p, T* v) {
   if constexpr (V == 1) {
     v[0] = *p;
+  } else if constexpr (sizeof(T) == 8 && V == 4) {
+    // two 16-B loads (32 B/lane; no native 32-B vector load)
+    double2 a = *reinterpret_cast<const double2*>(p);
+    double2 b = *reinterpret_cast<const double2*>(p + 2);
+    v[0] = a.x;
+    v[1] = a.y;
+    v[2] = b.x;
+    v[3] = b.y;
   } else {
     using VT = typename VecT<T, V>::type;
     VT t = *reinterpret_cast<const VT*>(p);
@@ -71,6 +82,14 @@ template <typename T, int V>
 __device__ __forceinline__ void storev(T* __restrict__ p, const T* v) {
   if constexpr (V == 1) {
     *p = v[0];
+  } else if constexpr (sizeof(T) == 8 && V == 4) {
+    double2 a, b;
+    a.x = v[0];
+    a.y = v[1];
+    b.x = v[2];
+    b.y = v[3];
+    *reinterpret_cast<double2*>(p) = a;
+    *reinterpret_cast<double2*>(p + 2) = b;
   } else {
     using VT = typename VecT<T, V>::type;
     VT t;
@@ -887,6 +906,14 @@ __global__ void __launch_bounds__(BLK) fd_kernel(Rows<T> R, T* __restrict__ y,
   }
 }
 
+static int fd_vec_override() {
+  static int v = [] {
+    const char* e = getenv("PAM_FD_VEC");
+    return e ? atoi(e) : 0;
+  }();
+  return v;
+}
+
 template <typename T, int OP>
 static int fd_launch(void* stream, int edge, const void* x, const void* gf,
                      const void* gb, void* y, int64_t nloc, int64_t m,
@@ -894,12 +921,18 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
   if (nloc < 0 || m <= 0) return PAM_EARG;
   if (nloc == 0) return 0;
   Rows<T> R{(const T*)x, (const T*)gf, (const T*)gb, nloc, m, FDDef<OP>::W};
-  constexpr int V = VecW<T>::value;
-  const bool vec_ok = (m % V == 0) && ((uintptr_t)x % 16 == 0) &&
-                      ((uintptr_t)y % 16 == 0) &&
-                      (gf == nullptr || (uintptr_t)gf % 16 == 0) &&
-                      (gb == nullptr || (uintptr_t)gb % 16 == 0);
-  const int64_t mv = vec_ok ? m / V : m;
+  const bool align16 = ((uintptr_t)x % 16 == 0) && ((uintptr_t)y % 16 == 0) &&
+                       (gf == nullptr || (uintptr_t)gf % 16 == 0) &&
+                       (gb == nullptr || (uintptr_t)gb % 16 == 0);
+  int V = 1;
+  if (align16) {
+    V = (m % 4 == 0) ? 4 : ((m % 2 == 0 && sizeof(T) == 8) ? 2 : 1);
+    const int ov = fd_vec_override();
+    if (ov == 1 || ov == 2 || ov == 4) {
+      if (m % ov == 0) V = ov;
+    }
+  }
+  const int64_t mv = m / V;
   int gy = (int)(nloc < 512 ? nloc : 512);
   int64_t gx64 = (mv + BLK - 1) / BLK;
   int64_t cap = 4096 / gy;
@@ -907,8 +940,11 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
   if (gx64 > cap) gx64 = cap;
   dim3 grid((uint32_t)gx64, (uint32_t)gy);
   hipStream_t s = (hipStream_t)stream;
-  if (vec_ok)
-    hipLaunchKernelGGL((fd_kernel<T, OP, V>), grid, dim3(BLK), 0, s, R, (T*)y,
+  if (V == 4)
+    hipLaunchKernelGGL((fd_kernel<T, OP, 4>), grid, dim3(BLK), 0, s, R, (T*)y,
+                       row0, nglob, (T)coeff, edge);
+  else if (V == 2 && sizeof(T) == 8)
+    hipLaunchKernelGGL((fd_kernel<T, OP, 2>), grid, dim3(BLK), 0, s, R, (T*)y,
                        row0, nglob, (T)coeff, edge);
   else
     hipLaunchKernelGGL((fd_kernel<T, OP, 1>), grid, dim3(BLK), 0, s, R, (T*)y,
