@@ -88,18 +88,12 @@ __global__ void __launch_bounds__(GBLK) gemm_kernel(
   const int li = lane & (TM - 1);     // fragment row/col within MFMA tile
   const int lk = lane / TM;           // fragment k (0..TK-1)
 
-  // XCD-aware bijective remap (T1, cdna_hip_programming.md §5.5): the
-  // dispatcher places block b on XCD b%8; giving each XCD a contiguous
-  // chunk of logical tiles makes neighbouring tiles' panel re-reads
-  // L2-local (+~10% when HBM-bound at large N).
-  const int64_t nwg = (int64_t)gridDim.x * gridDim.y;
-  const int64_t orig = (int64_t)blockIdx.y * gridDim.x + blockIdx.x;
-  const int64_t q8 = nwg / 8, r8 = nwg % 8, xcd = orig % 8;
-  const int64_t wgid = (xcd < r8 ? xcd * (q8 + 1)
-                                 : r8 * (q8 + 1) + (xcd - r8) * q8)
-                       + orig / 8;
-  const int64_t brow = (wgid / gridDim.x) * BM;
-  const int64_t bcol = (wgid % gridDim.x) * BN;
+  // NOTE: an XCD-aware bijective tile remap (T1) was measured and
+  // REGRESSED this kernel (f32 4096^3: 97->83 TF; 8192^3: 110->101 TF;
+  // f64: 49.7->46.7) — the 128^2-tile structure is MFMA-issue-bound here,
+  // not HBM-bound, so the remap only disturbed dispatch locality.
+  const int64_t brow = (int64_t)blockIdx.y * BM;
+  const int64_t bcol = (int64_t)blockIdx.x * BN;
 
   acc_t acc[2][2];
 #pragma unroll

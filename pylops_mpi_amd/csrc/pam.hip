@@ -924,13 +924,16 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
   const bool align16 = ((uintptr_t)x % 16 == 0) && ((uintptr_t)y % 16 == 0) &&
                        (gf == nullptr || (uintptr_t)gf % 16 == 0) &&
                        (gb == nullptr || (uintptr_t)gb % 16 == 0);
+  // default: 16 B/lane (double2 / float4) — the measured optimum; 32 B
+  // (V=4 double) was -14% in an A/B (PAM_FD_VEC keeps the knob)
   int V = 1;
   if (align16) {
-    V = (m % 4 == 0) ? 4 : ((m % 2 == 0 && sizeof(T) == 8) ? 2 : 1);
+    if (sizeof(T) == 8)
+      V = (m % 2 == 0) ? 2 : 1;
+    else
+      V = (m % 4 == 0) ? 4 : 1;
     const int ov = fd_vec_override();
-    if (ov == 1 || ov == 2 || ov == 4) {
-      if (m % ov == 0) V = ov;
-    }
+    if ((ov == 1 || ov == 2 || ov == 4) && m % ov == 0) V = ov;
   }
   const int64_t mv = m / V;
   int gy = (int)(nloc < 512 ? nloc : 512);
