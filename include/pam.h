@@ -82,6 +82,12 @@ int pam_cscale(void* stream, void* y, const void* x, double alpha_re,
                double alpha_im, int64_t n, int dtype);
 int pam_conj(void* stream, void* y, const void* x, int64_t n, int dtype);
 
+/* Thresholding for ISTA/FISTA (pylops _softthreshold/_hardthreshold
+ * formulas, imported by ref optimization/cls_sparsity.py:10).
+ * kind: 0 = soft, 1 = hard.  In-place safe (y may equal x). */
+int pam_thresh(void* stream, void* y, const void* x, int64_t n, int kind,
+               double thresh, int dtype);
+
 /* ------------------------------------------------------------------ *
  * Reductions (wavefront-shuffle + LDS tree, deterministic tree shape for
  * a given n — fixed partial count, fixed combine order).

@@ -19,6 +19,7 @@ from .localops import DenseLocal, CallableLocal, LocalOperator  # noqa: F401
 from .derivative import (MPIFirstDerivative,  # noqa: F401
                          MPISecondDerivative)
 from .solvers import CG, CGLS, cg, cgls, power_iteration  # noqa: F401
+from .sparsity import ISTA, FISTA, ista, fista  # noqa: F401
 from .stacked import (StackedDistributedArray,  # noqa: F401
                       MPIStackedLinearOperator)
 from .vstack import (MPIVStack, MPIHStack,  # noqa: F401

@@ -47,6 +47,9 @@ _SIGS = {
                     ctypes.c_int], ctypes.c_int),
     "pam_conj": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                   ctypes.c_int64, ctypes.c_int], ctypes.c_int),
+    "pam_thresh": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                    ctypes.c_int64, ctypes.c_int, ctypes.c_double,
+                    ctypes.c_int], ctypes.c_int),
     "pam_cdot": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                   ctypes.c_int64, ctypes.c_int, ctypes.c_void_p,
                   ctypes.c_void_p, ctypes.c_int], ctypes.c_int),

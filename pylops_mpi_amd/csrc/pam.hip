@@ -280,6 +280,73 @@ EWC_ENTRY(pam_conj(void* stream, void* y, const void* a, int64_t n,
           2, nullptr, 0.0, 0.0)
 
 // ---------------------------------------------------------------------------
+// thresholding for ISTA/FISTA (the pylops _softthreshold/_hardthreshold
+// formulas the reference imports, ref optimization/cls_sparsity.py:10):
+//   soft real   : sign(x) * max(|x|-t, 0)
+//   soft complex: z * max(|z|-t, 0)/|z|
+//   hard        : x * (|x| >= sqrt(2 t))
+// kind: 0 soft, 1 hard.  In-place safe (y may alias x).
+// ---------------------------------------------------------------------------
+template <typename T, bool CPLX, int KIND>
+__global__ void __launch_bounds__(BLK) thresh_kernel(T* __restrict__ y,
+                                                     const T* __restrict__ x,
+                                                     T thresh, int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    if constexpr (!CPLX) {
+      const T v = x[i];
+      if constexpr (KIND == 0) {
+        const T m = fabs(v) - thresh;
+        y[i] = m > (T)0 ? copysign(m, v) : (T)0;
+      } else {
+        y[i] = fabs(v) >= sqrt((T)2 * thresh) ? v : (T)0;
+      }
+    } else {
+      const T zr = x[2 * i], zi = x[2 * i + 1];
+      const T a = hypot(zr, zi);
+      T s;
+      if constexpr (KIND == 0) {
+        const T m = a - thresh;
+        s = (m > (T)0 && a > (T)0) ? m / a : (T)0;
+      } else {
+        s = a >= sqrt((T)2 * thresh) ? (T)1 : (T)0;
+      }
+      y[2 * i] = zr * s;
+      y[2 * i + 1] = zi * s;
+    }
+  }
+}
+
+template <typename T, bool CPLX>
+static int thresh_launch(void* stream, void* y, const void* x, int kind,
+                         double t, int64_t n) {
+  if (n < 0) return PAM_EARG;
+  if (n == 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  if (kind == 0)
+    hipLaunchKernelGGL((thresh_kernel<T, CPLX, 0>), dim3(grid_1d(n)),
+                       dim3(BLK), 0, s, (T*)y, (const T*)x, (T)t, n);
+  else if (kind == 1)
+    hipLaunchKernelGGL((thresh_kernel<T, CPLX, 1>), dim3(grid_1d(n)),
+                       dim3(BLK), 0, s, (T*)y, (const T*)x, (T)t, n);
+  else
+    return PAM_EOP;
+  return check(hipGetLastError());
+}
+
+extern "C" int pam_thresh(void* stream, void* y, const void* x, int64_t n,
+                          int kind, double thresh, int dtype) {
+  switch (dtype) {
+    case PAM_F64: return thresh_launch<double, false>(stream, y, x, kind, thresh, n);
+    case PAM_F32: return thresh_launch<float, false>(stream, y, x, kind, thresh, n);
+    case PAM_C128: return thresh_launch<double, true>(stream, y, x, kind, thresh, n);
+    case PAM_C64: return thresh_launch<float, true>(stream, y, x, kind, thresh, n);
+    default: return PAM_EDTYPE;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // reductions (ref DistributedArray.py:685-717 dot, :719-838 norms)
 // RED codes: 0 dot, 1 powsum |x^p|, 2 max|x|, 3 min|x|, 4 count_nonzero
 // ---------------------------------------------------------------------------

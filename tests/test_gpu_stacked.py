@@ -132,3 +132,52 @@ def test_power_iteration():
     maxeig, b, it = pm.power_iteration(op, b0, niter=400, tol=1e-12)
     want = np.linalg.eigvalsh(S).max()
     assert_allclose(maxeig, want, rtol=1e-6)
+
+
+# ------------------------------------------------------- sparsity solvers
+def test_threshold_kernel_vs_numpy():
+    from pylops_mpi_amd import _ffi
+    rng = np.random.default_rng(5)
+    for dt, tol in ((np.float64, 1e-14), (np.complex128, 1e-14)):
+        x = rng.standard_normal(257)
+        if np.dtype(dt).kind == "c":
+            x = x + 1j * rng.standard_normal(257)
+        x = x.astype(dt)
+        for kind, want in ((0, None), (1, None)):
+            t = 0.7
+            xd = dev(x.copy())
+            out = torch.empty_like(xd)
+            s = torch.cuda.current_stream().cuda_stream
+            _ffi.checked(_ffi.lib().pam_thresh(
+                s, out.data_ptr(), xd.data_ptr(), x.size, kind, t,
+                _ffi.dtype_code(xd.dtype)), "thresh")
+            a = np.abs(x)
+            if kind == 0:  # pylops _softthreshold
+                with np.errstate(invalid="ignore", divide="ignore"):
+                    w = np.where(a > t, (a - t) / np.where(a > 0, a, 1), 0.0)
+                want = x * w
+            else:          # pylops _hardthreshold
+                want = x * (a >= np.sqrt(2 * t))
+            assert_allclose(host(out), want, rtol=1e-12, atol=1e-14)
+
+
+@pytest.mark.parametrize("solver", ["ista", "fista"])
+def test_sparse_recovery(solver):
+    """Soft-thresholded proximal gradient recovers a sparse model
+    (ref tests/test_sparsity-style recipe)."""
+    rng = np.random.default_rng(6)
+    n, m, k = 60, 40, 4
+    A = rng.standard_normal((n, m)) / np.sqrt(n)
+    xt = np.zeros(m)
+    xt[rng.choice(m, k, replace=False)] = rng.standard_normal(k) + 3.0
+    yg = A @ xt
+    op = pm.MPIBlockDiag([pm.DenseLocal(dev(A))])
+    y = pm.DistributedArray.to_dist(dev(yg))
+    x0 = pm.DistributedArray((m,))
+    x0[:] = 0.0
+    fn = pm.ista if solver == "ista" else pm.fista
+    xs, niters, cost = fn(op, y, x0, niter=400, eps=1e-2, tol=1e-12)
+    got = host(xs.asarray())
+    # support recovered and large entries close
+    assert np.allclose(got, xt, atol=0.15)
+    assert cost[-1] <= cost[0]
