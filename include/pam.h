@@ -140,10 +140,13 @@ int pam_norm_local(void* stream, const void* x, int64_t n, int op, double p,
  *  12 fd2 centered matvec  13 fd2 centered rmatvec
  * ------------------------------------------------------------------ */
 int64_t pam_fd_halo_width(int op);
+/* rbegin/rend restrict the computed LOCAL row range [rbegin, rend) so the
+ * interior (halo-independent) rows can be launched concurrently with the
+ * RCCL halo exchange and the boundary rows after it completes. */
 int pam_fd_apply(void* stream, int op, int edge, const void* x,
                  const void* gf, const void* gb, void* y, int64_t nloc,
-                 int64_t m, int64_t row0, int64_t nglob, double coeff,
-                 int dtype);
+                 int64_t m, int64_t row0, int64_t nglob, int64_t rbegin,
+                 int64_t rend, double coeff, int dtype);
 
 /* ------------------------------------------------------------------ *
  * Dense GEMV: y = A @ x (trans=0) or y = A^T @ x (trans=1) for a

@@ -77,8 +77,9 @@ _SIGS = {
     "pam_fd_apply": ([ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
                       ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                       ctypes.c_void_p, ctypes.c_int64, ctypes.c_int64,
-                      ctypes.c_int64, ctypes.c_int64, ctypes.c_double,
-                      ctypes.c_int], ctypes.c_int),
+                      ctypes.c_int64, ctypes.c_int64, ctypes.c_int64,
+                      ctypes.c_int64, ctypes.c_double, ctypes.c_int],
+                     ctypes.c_int),
 }
 
 

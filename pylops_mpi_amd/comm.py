@@ -132,16 +132,16 @@ class PamComm:
         dist.all_gather(out, send, group=self._group)
         return [o[: counts[r]].reshape(shapes[r]) for r, o in enumerate(out)]
 
-    def sendrecv_neighbors(self, send_prev: Optional[torch.Tensor],
-                           send_next: Optional[torch.Tensor],
-                           recv_prev: Optional[torch.Tensor],
-                           recv_next: Optional[torch.Tensor]) -> None:
-        """Bidirectional nearest-neighbour exchange (halo / ghost cells,
-        ref DistributedArray.py:955-1032).  Any argument may be None
-        (global edges).  Posted as one batched isend/irecv group so RCCL
-        pairs them without deadlock."""
+    def post_neighbors(self, send_prev: Optional[torch.Tensor],
+                       send_next: Optional[torch.Tensor],
+                       recv_prev: Optional[torch.Tensor],
+                       recv_next: Optional[torch.Tensor]):
+        """Post the nearest-neighbour exchange WITHOUT waiting; returns the
+        list of Work handles.  The comm runs on RCCL's own stream, so
+        compute launched on the current stream between post and wait
+        overlaps the transfer (interior-rows/halo overlap)."""
         if not self._use_dist:
-            return
+            return []
         prev = self.ranks[self.rank - 1] if self.rank > 0 else None
         nxt = self.ranks[self.rank + 1] if self.rank < self.size - 1 else None
         ops = []
@@ -153,9 +153,19 @@ class PamComm:
             ops.append(dist.P2POp(dist.isend, send_prev, prev))
         if send_next is not None:
             ops.append(dist.P2POp(dist.isend, send_next, nxt))
-        if ops:
-            for w in dist.batch_isend_irecv(ops):
-                w.wait()
+        return dist.batch_isend_irecv(ops) if ops else []
+
+    def sendrecv_neighbors(self, send_prev: Optional[torch.Tensor],
+                           send_next: Optional[torch.Tensor],
+                           recv_prev: Optional[torch.Tensor],
+                           recv_next: Optional[torch.Tensor]) -> None:
+        """Bidirectional nearest-neighbour exchange (halo / ghost cells,
+        ref DistributedArray.py:955-1032).  Any argument may be None
+        (global edges).  Posted as one batched isend/irecv group so RCCL
+        pairs them without deadlock."""
+        for w in self.post_neighbors(send_prev, send_next, recv_prev,
+                                     recv_next):
+            w.wait()
 
     def sendrecv(self, sendbuf: torch.Tensor, dest: int,
                  recvbuf: torch.Tensor, source: int) -> torch.Tensor:

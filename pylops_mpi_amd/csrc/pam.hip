@@ -928,10 +928,11 @@ __device__ __forceinline__ void fd_edge(const Rows<T>& R, int64_t i, int64_t j,
 template <typename T, int OP, int V>
 __global__ void __launch_bounds__(BLK) fd_kernel(Rows<T> R, T* __restrict__ y,
                                                  int64_t row0, int64_t N, T c,
-                                                 int edge) {
+                                                 int edge, int64_t rbegin,
+                                                 int64_t rend) {
   const int64_t m = R.m, mv = m / V;
   const int64_t cstride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = blockIdx.y; i < R.nloc; i += gridDim.y) {
+  for (int64_t i = rbegin + blockIdx.y; i < rend; i += gridDim.y) {
     const int64_t g = row0 + i;
     T* __restrict__ yrow = y + i * m;
     for (int64_t jv = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; jv < mv;
@@ -984,9 +985,11 @@ static int fd_vec_override() {
 template <typename T, int OP>
 static int fd_launch(void* stream, int edge, const void* x, const void* gf,
                      const void* gb, void* y, int64_t nloc, int64_t m,
-                     int64_t row0, int64_t nglob, double coeff) {
-  if (nloc < 0 || m <= 0) return PAM_EARG;
-  if (nloc == 0) return 0;
+                     int64_t row0, int64_t nglob, int64_t rbegin,
+                     int64_t rend, double coeff) {
+  if (nloc < 0 || m <= 0 || rbegin < 0 || rend > nloc) return PAM_EARG;
+  const int64_t nrows = rend - rbegin;
+  if (nloc == 0 || nrows <= 0) return 0;
   Rows<T> R{(const T*)x, (const T*)gf, (const T*)gb, nloc, m, FDDef<OP>::W};
   const bool align16 = ((uintptr_t)x % 16 == 0) && ((uintptr_t)y % 16 == 0) &&
                        (gf == nullptr || (uintptr_t)gf % 16 == 0) &&
@@ -1003,7 +1006,7 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
     if ((ov == 1 || ov == 2 || ov == 4) && m % ov == 0) V = ov;
   }
   const int64_t mv = m / V;
-  int gy = (int)(nloc < 512 ? nloc : 512);
+  int gy = (int)(nrows < 512 ? nrows : 512);
   int64_t gx64 = (mv + BLK - 1) / BLK;
   int64_t cap = 4096 / gy;
   if (cap < 1) cap = 1;
@@ -1012,36 +1015,37 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
   hipStream_t s = (hipStream_t)stream;
   if (V == 4)
     hipLaunchKernelGGL((fd_kernel<T, OP, 4>), grid, dim3(BLK), 0, s, R, (T*)y,
-                       row0, nglob, (T)coeff, edge);
+                       row0, nglob, (T)coeff, edge, rbegin, rend);
   else if (V == 2 && sizeof(T) == 8)
     hipLaunchKernelGGL((fd_kernel<T, OP, 2>), grid, dim3(BLK), 0, s, R, (T*)y,
-                       row0, nglob, (T)coeff, edge);
+                       row0, nglob, (T)coeff, edge, rbegin, rend);
   else
     hipLaunchKernelGGL((fd_kernel<T, OP, 1>), grid, dim3(BLK), 0, s, R, (T*)y,
-                       row0, nglob, (T)coeff, edge);
+                       row0, nglob, (T)coeff, edge, rbegin, rend);
   return check(hipGetLastError());
 }
 
 extern "C" int pam_fd_apply(void* stream, int op, int edge, const void* x,
                             const void* gf, const void* gb, void* y,
                             int64_t nloc, int64_t m, int64_t row0,
-                            int64_t nglob, double coeff, int dtype) {
+                            int64_t nglob, int64_t rbegin, int64_t rend,
+                            double coeff, int dtype) {
 #define FD_CASE(T)                                                            \
   switch (op) {                                                               \
-    case 0: return fd_launch<T, 0>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, coeff); \
-    case 1: return fd_launch<T, 1>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, coeff); \
-    case 2: return fd_launch<T, 2>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, coeff); \
-    case 3: return fd_launch<T, 3>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, coeff); \
-    case 4: return fd_launch<T, 4>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, coeff); \
-    case 5: return fd_launch<T, 5>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, coeff); \
-    case 6: return fd_launch<T, 6>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, coeff); \
-    case 7: return fd_launch<T, 7>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, coeff); \
-    case 8: return fd_launch<T, 8>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, coeff); \
-    case 9: return fd_launch<T, 9>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, coeff); \
-    case 10: return fd_launch<T, 10>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, coeff); \
-    case 11: return fd_launch<T, 11>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, coeff); \
-    case 12: return fd_launch<T, 12>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, coeff); \
-    case 13: return fd_launch<T, 13>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, coeff); \
+    case 0: return fd_launch<T, 0>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, rbegin, rend, coeff); \
+    case 1: return fd_launch<T, 1>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, rbegin, rend, coeff); \
+    case 2: return fd_launch<T, 2>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, rbegin, rend, coeff); \
+    case 3: return fd_launch<T, 3>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, rbegin, rend, coeff); \
+    case 4: return fd_launch<T, 4>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, rbegin, rend, coeff); \
+    case 5: return fd_launch<T, 5>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, rbegin, rend, coeff); \
+    case 6: return fd_launch<T, 6>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, rbegin, rend, coeff); \
+    case 7: return fd_launch<T, 7>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, rbegin, rend, coeff); \
+    case 8: return fd_launch<T, 8>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, rbegin, rend, coeff); \
+    case 9: return fd_launch<T, 9>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, rbegin, rend, coeff); \
+    case 10: return fd_launch<T, 10>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, rbegin, rend, coeff); \
+    case 11: return fd_launch<T, 11>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, rbegin, rend, coeff); \
+    case 12: return fd_launch<T, 12>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, rbegin, rend, coeff); \
+    case 13: return fd_launch<T, 13>(stream, edge, x, gf, gb, y, nloc, m, row0, nglob, rbegin, rend, coeff); \
     default: return PAM_EOP;                                                  \
   }
   if (dtype == PAM_F64) { FD_CASE(double) }

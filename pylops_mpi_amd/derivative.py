@@ -106,19 +106,48 @@ class _FDBase(MPILinearOperator):
         planes = flat.view(nloc, m) if m > 1 else flat.view(nloc, 1)
         x._require_compute()
         w = int(_ffi.lib().pam_fd_halo_width(op))
-        gf, gb = self._halo(planes, w, comm)
         y = torch.empty_like(planes)
         row0 = int(np.sum([s[0] for s in shapes[: comm.rank]], initial=0))
         stream = torch.cuda.current_stream(planes.device).cuda_stream
+        m = planes.shape[1]
+        dt = _ffi.dtype_code(planes.dtype)
+        edge = 1 if self.edge else 0
+
+        def launch(gfp, gbp, r0, r1):
+            _ffi.checked(_ffi.lib().pam_fd_apply(
+                stream, op, edge, planes.data_ptr(),
+                gfp.data_ptr() if gfp is not None else None,
+                gbp.data_ptr() if gbp is not None else None,
+                y.data_ptr(), nloc, m, row0, self.dims[0], r0, r1,
+                self._coeff(), dt), "fd_apply")
+
         ev = _record_events(op) if KERNEL_TIMING else None
         if ev is not None:
             ev[0].record()
-        _ffi.checked(_ffi.lib().pam_fd_apply(
-            stream, op, 1 if self.edge else 0, planes.data_ptr(),
-            gf.data_ptr() if gf is not None else None,
-            gb.data_ptr() if gb is not None else None,
-            y.data_ptr(), nloc, planes.shape[1], row0, self.dims[0],
-            self._coeff(), _ffi.dtype_code(planes.dtype)), "fd_apply")
+        if comm.size > 1 and w > 0 and nloc > 2 * w:
+            # overlap: post the RCCL halo exchange (its own stream), run the
+            # halo-independent interior rows concurrently, then the
+            # boundary rows once the planes have arrived
+            r, P = comm.rank, comm.size
+            if nloc < w and 0 < r < P - 1:
+                raise ValueError(
+                    f"Local Shape at rank={r} along axis=0 should be > {w}")
+            gshape = (w,) + tuple(planes.shape[1:])
+            gf = torch.empty(gshape, dtype=planes.dtype,
+                             device=planes.device) if r > 0 else None
+            gb = torch.empty(gshape, dtype=planes.dtype,
+                             device=planes.device) if r < P - 1 else None
+            works = comm.post_neighbors(
+                planes[:w].contiguous() if r > 0 else None,
+                planes[-w:].contiguous() if r < P - 1 else None, gf, gb)
+            launch(None, None, w, nloc - w)      # interior
+            for wk in works:
+                wk.wait()
+            launch(gf, gb, 0, w)                 # boundary rows
+            launch(gf, gb, nloc - w, nloc)
+        else:
+            gf, gb = self._halo(planes, w, comm)
+            launch(gf, gb, 0, nloc)
         if ev is not None:
             ev[1].record()
         # ravel back to a 1-D plane-aligned DistributedArray

@@ -169,6 +169,7 @@ def body_blockdiag(c):
     "body_allreduce", "body_allgather_obj", "body_allgather_tensors",
     "body_to_dist_asarray", "body_ghost_cells", "body_rebalance",
     "body_sendrecv", "body_blockdiag", "body_fredholm", "body_vstack",
+    "body_post_neighbors_overlap",
 ])
 def test_gloo_world2(body):
     _spawn(body)
@@ -229,3 +230,23 @@ def body_vstack(c):
     gotr = op.rmatvec(yd)
     np.testing.assert_allclose(gotr.local_array.numpy(), D.T @ y,
                                rtol=1e-12)
+
+
+def body_post_neighbors_overlap(c):
+    """post_neighbors returns live Work handles; values land after wait
+    (the derivative overlap choreography on the comm layer)."""
+    import numpy as np
+    t = torch.full((3, 4), float(c.rank + 1), dtype=torch.float64)
+    gf = torch.empty((1, 4), dtype=torch.float64) if c.rank > 0 else None
+    gb = torch.empty((1, 4), dtype=torch.float64) if c.rank < 1 else None
+    works = c.post_neighbors(t[:1].contiguous() if c.rank > 0 else None,
+                             t[-1:].contiguous() if c.rank < 1 else None,
+                             gf, gb)
+    acc = t.sum()  # "interior compute" between post and wait
+    for w in works:
+        w.wait()
+    if c.rank == 0:
+        assert torch.all(gb == 2.0)
+    else:
+        assert torch.all(gf == 1.0)
+    assert float(acc) == 12.0 * (c.rank + 1)

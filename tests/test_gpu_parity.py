@@ -296,3 +296,44 @@ def test_wrapped_serial_operator():
                                     partition=pm.Partition.BROADCAST)
     y = op.matvec(x)
     assert_allclose(host(y.asarray()), A @ xg, rtol=1e-13)
+
+
+def test_fd_split_rank_emulation():
+    """Emulate the world-2 overlapped apply on one GPU: per-rank blocks,
+    manually filled halo buffers, interior+boundary range launches — must
+    reproduce the oracle's 2-rank distributed result exactly.  This is
+    the code path the multi-GPU bench takes (derivative._apply overlap
+    branch)."""
+    from pylops_mpi_amd import _ffi
+    for kind, order, opmv, oprmv in [("centered", 3, 4, 5),
+                                     ("centered", 5, 6, 7)]:
+        dims = (24, 6)
+        N, m = dims
+        rng = np.random.default_rng(77)
+        xg = rng.standard_normal(N * m)
+        sop = oracle.SimFirstDerivative(dims, 1.3, kind, True, order)
+        sx = oracle.to_dist(xg, 2)
+        want_mv = sop.matvec(sx).asarray()
+        want_rmv = sop.rmatvec(sx).asarray()
+        xt = dev(xg).reshape(N, m)
+        h = 12  # local_split(24, 2) -> 12/12
+        s = torch.cuda.current_stream().cuda_stream
+        for op, want in ((opmv, want_mv), (oprmv, want_rmv)):
+            w = int(_ffi.lib().pam_fd_halo_width(op))
+            outs = []
+            for r, (r0, r1) in enumerate(((0, h), (h, N))):
+                loc = xt[r0:r1].contiguous()
+                nloc = r1 - r0
+                gf = xt[r0 - w: r0].contiguous() if r > 0 else None
+                gb = xt[r1: r1 + w].contiguous() if r < 1 else None
+                y = torch.empty_like(loc)
+                for (a, b) in ((w, nloc - w), (0, w), (nloc - w, nloc)):
+                    _ffi.checked(_ffi.lib().pam_fd_apply(
+                        s, op, 1, loc.data_ptr(),
+                        gf.data_ptr() if gf is not None else None,
+                        gb.data_ptr() if gb is not None else None,
+                        y.data_ptr(), nloc, m, r0, N, a, b, 1.0 / 1.3,
+                        _ffi.dtype_code(loc.dtype)), "fd")
+                outs.append(y.reshape(-1))
+            got = host(torch.cat(outs))
+            assert_allclose(got, want, rtol=1e-13, atol=1e-14), (kind, op)
