@@ -68,7 +68,7 @@ template <> struct GemmCfg<double> {
 };
 
 // C = A@B (+C when ACC), A [M,K] lda, B [K,N] ldb, C [M,N] ldc, row-major.
-template <typename T, bool ACC>
+template <typename T, bool ACC, bool GUARD>
 __global__ void __launch_bounds__(GBLK) gemm_kernel(
     const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
     int64_t M, int64_t N, int64_t K, int64_t lda, int64_t ldb, int64_t ldc) {
@@ -111,7 +111,7 @@ __global__ void __launch_bounds__(GBLK) gemm_kernel(
       const int kv = vi % (BK / VW);
       T vals[VW];
       const int64_t gr = brow + row;
-      if (gr < M && k0 + (int64_t)kv * VW + VW <= K) {
+      if (!GUARD || (gr < M && k0 + (int64_t)kv * VW + VW <= K)) {
         vec_t v = *reinterpret_cast<const vec_t*>(A + gr * lda + k0
                                                   + (int64_t)kv * VW);
 #pragma unroll
@@ -135,7 +135,7 @@ __global__ void __launch_bounds__(GBLK) gemm_kernel(
       const int nv = vi % (BN / VW);
       const int64_t gk = k0 + kb;
       const int64_t gn = bcol + (int64_t)nv * VW;
-      if (gk < K && gn + VW <= N) {
+      if (!GUARD || (gk < K && gn + VW <= N)) {
         *reinterpret_cast<vec_t*>(&Bs[kb][nv * VW]) =
             *reinterpret_cast<const vec_t*>(B + gk * ldb + gn);
       } else {
@@ -170,11 +170,11 @@ __global__ void __launch_bounds__(GBLK) gemm_kernel(
       const int64_t r0 = brow + wr * 2 * TM + mi * TM;
       const int64_t c0 = bcol + wc * 2 * TM + nj * TM;
       const int64_t cc = c0 + li;
-      if (cc >= N) continue;
+      if (GUARD && cc >= N) continue;
 #pragma unroll
       for (int reg = 0; reg < CFG::NREG; ++reg) {
         const int64_t rr = r0 + CFG::crow(lane, reg);
-        if (rr < M) {
+        if (!GUARD || rr < M) {
           if constexpr (ACC)
             C[rr * ldc + cc] += acc[mi][nj][reg];
           else
@@ -193,14 +193,27 @@ static int gemm_launch(void* stream, const void* A, const void* B, void* C,
   dim3 grid((uint32_t)((N + CFG::BN - 1) / CFG::BN),
             (uint32_t)((M + CFG::BM - 1) / CFG::BM));
   hipStream_t s = (hipStream_t)stream;
-  if (accumulate)
-    hipLaunchKernelGGL((gemm_kernel<T, true>), grid, dim3(GBLK), 0, s,
-                       (const T*)A, (const T*)B, (T*)C, M, N, K, lda, ldb,
-                       ldc);
-  else
-    hipLaunchKernelGGL((gemm_kernel<T, false>), grid, dim3(GBLK), 0, s,
-                       (const T*)A, (const T*)B, (T*)C, M, N, K, lda, ldb,
-                       ldc);
+  const bool aligned = (M % CFG::BM == 0) && (N % CFG::BN == 0) &&
+                       (K % CFG::BK == 0);
+  if (accumulate) {
+    if (aligned)
+      hipLaunchKernelGGL((gemm_kernel<T, true, false>), grid, dim3(GBLK), 0,
+                         s, (const T*)A, (const T*)B, (T*)C, M, N, K, lda,
+                         ldb, ldc);
+    else
+      hipLaunchKernelGGL((gemm_kernel<T, true, true>), grid, dim3(GBLK), 0,
+                         s, (const T*)A, (const T*)B, (T*)C, M, N, K, lda,
+                         ldb, ldc);
+  } else {
+    if (aligned)
+      hipLaunchKernelGGL((gemm_kernel<T, false, false>), grid, dim3(GBLK), 0,
+                         s, (const T*)A, (const T*)B, (T*)C, M, N, K, lda,
+                         ldb, ldc);
+    else
+      hipLaunchKernelGGL((gemm_kernel<T, false, true>), grid, dim3(GBLK), 0,
+                         s, (const T*)A, (const T*)B, (T*)C, M, N, K, lda,
+                         ldb, ldc);
+  }
   return gcheck(hipGetLastError());
 }
 
