@@ -28,6 +28,8 @@ from .rebalance import rebalance_1d
 _FD1_OPS = {("forward", 0): (0, 1), ("backward", 0): (2, 3),
             ("centered", 3): (4, 5), ("centered", 5): (6, 7)}
 _FD2_OPS = {"forward": (8, 9), "backward": (10, 11), "centered": (12, 13)}
+_NP_OF = {torch.float64: np.float64, torch.float32: np.float32,
+          torch.complex128: np.complex128, torch.complex64: np.complex64}
 
 # bench instrumentation: when enabled, each stencil kernel launch records a
 # HIP event pair on the launch stream (torch's current stream, which is
@@ -106,7 +108,13 @@ class _FDBase(MPILinearOperator):
         planes = flat.view(nloc, m) if m > 1 else flat.view(nloc, 1)
         x._require_compute()
         w = int(_ffi.lib().pam_fd_halo_width(op))
-        y = torch.empty_like(planes)
+        y_out = torch.empty_like(planes)
+        y = y_out
+        if planes.is_complex():
+            # real-coefficient stencil acts componentwise: run the real
+            # kernels on the interleaved (re,im) view with doubled columns
+            planes = torch.view_as_real(planes).reshape(nloc, 2 * m)
+            y = torch.view_as_real(y_out).reshape(nloc, 2 * m)
         row0 = int(np.sum([s[0] for s in shapes[: comm.rank]], initial=0))
         stream = torch.cuda.current_stream(planes.device).cuda_stream
         m = planes.shape[1]
@@ -155,8 +163,8 @@ class _FDBase(MPILinearOperator):
         #  no-op, ref DistributedArray.py:516-517)
         return DistributedArray(
             int(np.prod(self.dims)), comm, Partition.SCATTER, 0,
-            local_array=y.view(-1), local_shapes=[(c,) for c in counts],
-            dtype=self.dtype)
+            local_array=y_out.view(-1), local_shapes=[(c,) for c in counts],
+            dtype=_NP_OF[y_out.dtype])
 
     def _matvec(self, x: DistributedArray) -> DistributedArray:
         return self._apply(x, self._op_mv)

@@ -337,3 +337,29 @@ def test_fd_split_rank_emulation():
                 outs.append(y.reshape(-1))
             got = host(torch.cat(outs))
             assert_allclose(got, want, rtol=1e-13, atol=1e-14), (kind, op)
+
+
+@pytest.mark.parametrize("kind,order", [("centered", 3), ("centered", 5),
+                                        ("forward", 3)])
+def test_fd1_complex_vs_oracle(kind, order):
+    """Complex-dtype stencils (real coefficients act componentwise on the
+    interleaved view; ref test_derivative.py complex parameter sets)."""
+    dims = (20, 6)
+    n = int(np.prod(dims))
+    rng = np.random.default_rng(42)
+    xg = rng.standard_normal(n) + 1j * rng.standard_normal(n)
+    op = pm.MPIFirstDerivative(dims, 1.5, kind, order=order,
+                               dtype=np.complex128)
+    sop = oracle.SimFirstDerivative(dims, 1.5, kind, False, order,
+                                    dtype=np.complex128)
+    x = pm.DistributedArray.to_dist(dev(xg))
+    sx = oracle.to_dist(xg, 1)
+    assert_allclose(host(op.matvec(x).asarray()), sop.matvec(sx).asarray(),
+                    rtol=1e-13, atol=1e-14)
+    assert_allclose(host(op.rmatvec(x).asarray()),
+                    sop.rmatvec(sx).asarray(), rtol=1e-13, atol=1e-14)
+    u = pm.DistributedArray.to_dist(
+        dev(rng.standard_normal(n) + 1j * rng.standard_normal(n)))
+    v = pm.DistributedArray.to_dist(
+        dev(rng.standard_normal(n) + 1j * rng.standard_normal(n)))
+    assert pm.dottest(op, u, v, rtol=1e-10)

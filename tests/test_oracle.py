@@ -248,3 +248,17 @@ def test_cgls_blockdiag(P):
     x, cost = sim_cgls(op, to_dist(yg, P), to_dist(np.zeros(n), P),
                        niter=80, damp=0.0, tol=1e-30)
     assert_allclose(x.asarray(), np.linalg.solve(A, yg), rtol=1e-8, atol=1e-9)
+
+
+@pytest.mark.parametrize("P", [1, 3])
+def test_fd1_complex_vs_serial(P):
+    # ref tests/test_derivative.py includes complex128 parameter sets
+    dims = (17, 5)
+    n = int(np.prod(dims))
+    rng = np.random.default_rng(42)
+    xg = rng.standard_normal(n) + 1j * rng.standard_normal(n)
+    op = SimFirstDerivative(dims, 1.5, "centered", True, 5,
+                            dtype=np.complex128)
+    y = op.matvec(to_dist(xg, P))
+    want = serial_fd1_matvec(xg.reshape(dims), 1.5, "centered", True, 5)
+    assert_allclose(y.asarray().reshape(dims), want, rtol=1e-14)
