@@ -148,6 +148,15 @@ int pam_fd_apply(void* stream, int op, int edge, const void* x,
                  int64_t m, int64_t row0, int64_t nglob, int64_t rbegin,
                  int64_t rend, double coeff, int dtype);
 
+/* Serial (single-rank) stencil along an arbitrary axis: the block is
+ * viewed as [batch, d, m] with the derivative along d.  The local
+ * operators inside MPIBlockDiag for Gradient/Laplacian axes >= 1
+ * (ref basicoperators/Gradient.py:101-118, Laplacian.py:98-126, which
+ * wrap serial pylops First/SecondDerivative). */
+int pam_fd_serial(void* stream, int op, int edge, const void* x, void* y,
+                  int64_t batch, int64_t d, int64_t m, double coeff,
+                  int dtype);
+
 /* ------------------------------------------------------------------ *
  * Dense GEMV: y = A @ x (trans=0) or y = A^T @ x (trans=1) for a
  * row-major [nr, nc] matrix resident in HBM.

@@ -14,6 +14,9 @@ from .matmult import (MPIMatrixMult, active_grid_comm,  # noqa: F401
                       block_gather, local_block_split)
 from .fredholm import MPIFredholm1  # noqa: F401
 from .halo import MPIHalo  # noqa: F401
+from .fdlocal import (FirstDerivativeLocal,  # noqa: F401
+                      SecondDerivativeLocal)
+from .gradient import MPIGradient, MPILaplacian  # noqa: F401
 from .mdc import MPIMDC  # noqa: F401
 from .fftlocal import FFTLocal, IdentityLocal  # noqa: F401
 from .localops import DenseLocal, CallableLocal, LocalOperator  # noqa: F401
