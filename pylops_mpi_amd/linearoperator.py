@@ -32,7 +32,9 @@ class MPILinearOperator:
                 else dims
             dimsd = getattr(Op, "dimsd", (Op.shape[0],)) if dimsd is None \
                 else dimsd
-        else:
+        elif not hasattr(self, "Op"):
+            # don't clobber an Op a subclass set before calling super()
+            # (e.g. MPILaplacian's composed operator)
             self.Op = None
         if shape is not None:
             self.shape = shape
