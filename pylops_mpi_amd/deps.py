@@ -1,0 +1,20 @@
+"""Environment feature gates — the analogue of the reference's
+pylops_mpi/utils/deps.py:58-66 (NCCL_PYLOPS_MPI / PYLOPS_MPI_CUDA_AWARE):
+
+  PAM_DISABLE_OVERLAP=1   serialize the halo exchange before the stencil
+                          kernel instead of overlapping it with the
+                          interior rows (debug aid)
+  PAM_FD_VEC={1,2,4}      force the stencil kernel's per-lane vector width
+                          (default: measured optimum, 16 B/lane)
+
+The RCCL data plane itself has no gate: it IS the backend (there is no
+MPI in this stack to fall back to)."""
+import os
+
+
+def env_flag(name: str, default: bool = False) -> bool:
+    v = os.environ.get(name)
+    return default if v is None else v not in ("0", "", "false", "False")
+
+
+overlap_enabled = not env_flag("PAM_DISABLE_OVERLAP")

@@ -132,7 +132,9 @@ class _FDBase(MPILinearOperator):
         ev = _record_events(op) if KERNEL_TIMING else None
         if ev is not None:
             ev[0].record()
-        if comm.size > 1 and w > 0 and nloc > 2 * w:
+        from . import deps as _deps
+        if (_deps.overlap_enabled and comm.size > 1 and w > 0
+                and nloc > 2 * w):
             # overlap: post the RCCL halo exchange (its own stream), run the
             # halo-independent interior rows concurrently, then the
             # boundary rows once the planes have arrived

@@ -381,3 +381,11 @@ class _ConjLinearOperator(MPILinearOperator):
 
     def _adjoint(self):
         return _ConjLinearOperator(self.A.H)
+
+
+def asmpilinearoperator(Op):
+    """Return Op as an MPILinearOperator (ref LinearOperator.py:583-602):
+    pass-through for MPI operators, serial-wrap for local operators."""
+    if isinstance(Op, MPILinearOperator):
+        return Op
+    return MPILinearOperator(Op=Op)
