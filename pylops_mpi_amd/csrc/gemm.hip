@@ -127,22 +127,27 @@ __global__ void __launch_bounds__(GBLK) gemm_kernel(
       for (int j = 0; j < VW; ++j) As[kv * VW + j][row] = vals[j];
     }
     // ---- stage B[k0:k0+BK, bcol:bcol+BN] -> Bs[k][col] (linear)
-    constexpr int BV = (BK * BN) / VW / GBLK;
+    // (a 16-B global_load_lds variant was A/B-measured: f32 within noise
+    //  (+1%), f64 -9% — the synchronous 2-barrier structure gets no
+    //  benefit from DMA staging without a counted-vmcnt pipeline)
+    {
+      constexpr int BV = (BK * BN) / VW / GBLK;
 #pragma unroll
-    for (int e = 0; e < BV; ++e) {
-      const int vi = tid + GBLK * e;
-      const int kb = vi / (BN / VW);
-      const int nv = vi % (BN / VW);
-      const int64_t gk = k0 + kb;
-      const int64_t gn = bcol + (int64_t)nv * VW;
-      if (!GUARD || (gk < K && gn + VW <= N)) {
-        *reinterpret_cast<vec_t*>(&Bs[kb][nv * VW]) =
-            *reinterpret_cast<const vec_t*>(B + gk * ldb + gn);
-      } else {
+      for (int e = 0; e < BV; ++e) {
+        const int vi = tid + GBLK * e;
+        const int kb = vi / (BN / VW);
+        const int nv = vi % (BN / VW);
+        const int64_t gk = k0 + kb;
+        const int64_t gn = bcol + (int64_t)nv * VW;
+        if (!GUARD || (gk < K && gn + VW <= N)) {
+          *reinterpret_cast<vec_t*>(&Bs[kb][nv * VW]) =
+              *reinterpret_cast<const vec_t*>(B + gk * ldb + gn);
+        } else {
 #pragma unroll
-        for (int j = 0; j < VW; ++j)
-          Bs[kb][nv * VW + j] =
-              (gk < K && gn + j < N) ? B[gk * ldb + gn + j] : (T)0;
+          for (int j = 0; j < VW; ++j)
+            Bs[kb][nv * VW + j] =
+                (gk < K && gn + j < N) ? B[gk * ldb + gn + j] : (T)0;
+        }
       }
     }
     __syncthreads();
