@@ -24,9 +24,8 @@ class MPIBlockDiag(MPILinearOperator):
     def __init__(self, ops: Sequence[LocalOperator],
                  base_comm: Optional[PamComm] = None,
                  mask=None, dtype=None):
-        if mask is not None:
-            raise NotImplementedError("mask/sub-communicators deferred")
         comm = base_comm if base_comm is not None else get_default_comm()
+        self.mask = mask  # ref :102-105: passed through to the outputs
         self.ops = list(ops)
         # per-rank row/col offsets, ref BlockDiag.py:106-116
         nops = np.array([op.shape[0] for op in self.ops], dtype=np.int64)
@@ -67,7 +66,7 @@ class MPIBlockDiag(MPILinearOperator):
         gshape = self.shape[0] if forward else self.shape[1]
         return DistributedArray(
             int(gshape), comm, Partition.SCATTER, 0, local_array=out,
-            local_shapes=out_shapes, dtype=self.dtype)
+            local_shapes=out_shapes, mask=self.mask, dtype=self.dtype)
 
     def _matvec(self, x: DistributedArray) -> DistributedArray:
         return self._apply(x, forward=True)

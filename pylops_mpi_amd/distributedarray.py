@@ -65,6 +65,18 @@ def local_split(global_shape: Tuple, size: int, rank: int,
 # per-device reduction scratch: (ws, out) float64 tensors
 _red_scratch = {}
 
+# sub-communicator cache: creating a torch process group is collective, so
+# identical masks must map to one group (ref subcomm_split, :74-100)
+_subcomm_cache = {}
+
+
+def _get_subcomm(comm, mask_key):
+    key = (id(comm), mask_key)
+    if key not in _subcomm_cache:
+        _subcomm_cache[key] = comm.split_by(list(mask_key),
+                                            keys=list(range(comm.size)))
+    return _subcomm_cache[key]
+
 
 def _reduce_buffers(device):
     key = (device.type, device.index)
@@ -88,7 +100,8 @@ class DistributedArray:
       * ``local_shapes``, when given, must list every rank's shape (as in
         the reference); it is cached so no control-plane allgather is
         needed per op;
-      * ``mask`` (sub-communicators) is deferred to a later round.
+      * ``mask`` reductions run over a cached RCCL sub-group
+        (ref subcomm_split :74-100).
     """
 
     def __init__(self, global_shape: Union[Tuple, Integral],
@@ -112,16 +125,22 @@ class DistributedArray:
             raise ValueError(f"Should be either {Partition.BROADCAST}, "
                              f"{Partition.UNSAFE_BROADCAST} or "
                              f"{Partition.SCATTER}")
-        if mask is not None:
-            raise NotImplementedError(
-                "mask/sub-communicators are not implemented yet")
         self._engine = engine
         self._global_shape = global_shape
         self._base_comm = base_comm if base_comm is not None \
             else get_default_comm()
         self._partition = partition
         self._axis = axis
-        self._mask = None
+        # masked sub-communicator (ref :74-100,193-195): ranks sharing a
+        # mask value form the group the 'global' reductions run over
+        self._mask = None if mask is None else [int(v) for v in mask]
+        if mask is None:
+            self._sub_comm = self._base_comm
+        else:
+            if len(self._mask) != self._base_comm.size:
+                raise ValueError("Mask length must equal communicator size")
+            self._sub_comm = _get_subcomm(self._base_comm,
+                                          tuple(self._mask))
         self.dtype = np.dtype(dtype) if local_array is None \
             else _NP_DTYPES[local_array.dtype]
         if local_shapes is not None:
@@ -199,6 +218,16 @@ class DistributedArray:
         return self._mask
 
     @property
+    def sub_comm(self):
+        # ref :392-399
+        return self._sub_comm
+
+    def _check_mask(self, other):
+        # ref :581-585
+        if not np.array_equal(self._mask, other._mask):
+            raise ValueError("Mask of both the arrays must be same")
+
+    @property
     def device(self):
         d = self._base_comm.device
         return d if d is not None else torch.device("cpu")
@@ -240,6 +269,7 @@ class DistributedArray:
                                 self._partition, self._axis,
                                 local_array=local,
                                 local_shapes=self._all_local_shapes,
+                                mask=self._mask,
                                 engine=self._engine, dtype=self.dtype)
 
     def _check_partition_shape(self, other):
@@ -330,8 +360,14 @@ class DistributedArray:
         if self._partition in (Partition.BROADCAST,
                                Partition.UNSAFE_BROADCAST):
             return self._local_array
-        pieces = self._base_comm.allgather_tensors(
-            self._local_array.contiguous(), self._all_local_shapes)
+        if masked:
+            comm = self._sub_comm
+            shapes = [self._all_local_shapes[r] for r in comm.ranks]
+        else:
+            comm = self._base_comm
+            shapes = self._all_local_shapes
+        pieces = comm.allgather_tensors(
+            self._local_array.contiguous(), shapes)
         return torch.cat(pieces, dim=self._axis)
 
     # ------------------------------------------------------------ halo moves
@@ -416,6 +452,7 @@ class DistributedArray:
         return out
 
     def add(self, other: "DistributedArray") -> "DistributedArray":
+        self._check_mask(other)
         # ref :636-651
         self._check_partition_shape(other)
         self._require_compute()
@@ -429,6 +466,7 @@ class DistributedArray:
         return out
 
     def iadd(self, other: "DistributedArray") -> "DistributedArray":
+        self._check_mask(other)
         # ref :653-659
         self._check_partition_shape(other)
         self._require_compute()
@@ -457,6 +495,7 @@ class DistributedArray:
         out = self._like()
         if isinstance(x, DistributedArray):
             self._check_partition_shape(x)
+            self._check_mask(x)
             if self._is_cplx():
                 _ffi.checked(_ffi.lib().pam_cmul(
                     self._stream(), out._flat().data_ptr(),
@@ -530,6 +569,7 @@ class DistributedArray:
     def dot(self, other: "DistributedArray", vdot: bool = False):
         """ref :685-717 — local dot + allreduce.  Real dtypes: vdot == dot."""
         self._check_partition_shape(other)
+        self._check_mask(other)
         self._require_compute()
         if self._partition in (Partition.BROADCAST,
                                Partition.UNSAFE_BROADCAST):
@@ -543,14 +583,14 @@ class DistributedArray:
                 other._flat().data_ptr(), self._local_array.numel(),
                 1 if vdot else 0, ws.data_ptr(), out.data_ptr(),
                 self._dt()), "cdot")
-            self._base_comm.allreduce_(out, "sum")
+            self._sub_comm.allreduce_(out, "sum")
             v = out.cpu()
             return np.complex128(complex(float(v[0]), float(v[1])))
         _ffi.checked(_ffi.lib().pam_dot(
             self._stream(), self._flat().data_ptr(), other._flat().data_ptr(),
             self._local_array.numel(), ws.data_ptr(), out.data_ptr(),
             self._dt()), "dot")
-        self._base_comm.allreduce_(out[:1], "sum")
+        self._sub_comm.allreduce_(out[:1], "sum")
         return np.float64(out[0].item())
 
     def _norm_local(self, op: int, p: float) -> torch.Tensor:
@@ -576,18 +616,18 @@ class DistributedArray:
             raise ValueError(f"norm-{ord} not possible for vectors")
         if ord == 0:
             out = self._norm_local(3, 0.0)
-            self._base_comm.allreduce_(out, "sum")
+            self._sub_comm.allreduce_(out, "sum")
             return np.float64(out.item())
         if ord == np.inf:
             out = self._norm_local(1, 0.0)
-            self._base_comm.allreduce_(out, "max")
+            self._sub_comm.allreduce_(out, "max")
             return np.float64(out.item())
         if ord == -np.inf:
             out = self._norm_local(2, 0.0)
-            self._base_comm.allreduce_(out, "min")
+            self._sub_comm.allreduce_(out, "min")
             return np.float64(out.item())
         out = self._norm_local(0, float(ord))
-        self._base_comm.allreduce_(out, "sum")
+        self._sub_comm.allreduce_(out, "sum")
         return np.float64(out.item() ** (1.0 / ord))
 
     # ------------------------------------------------------------ structure
@@ -606,7 +646,11 @@ class DistributedArray:
         return self._like(self._local_array.clone())
 
     def zeros_like(self):
-        out = self._like()
+        # ref :791-803 (NB: the reference drops the mask here)
+        out = DistributedArray(self._global_shape, self._base_comm,
+                               self._partition, self._axis,
+                               local_shapes=self._all_local_shapes,
+                               engine=self._engine, dtype=self.dtype)
         out._local_array.zero_()
         return out
 
@@ -621,6 +665,38 @@ class DistributedArray:
             int(np.prod(self._global_shape)), self._base_comm,
             self._partition, 0, local_array=flat,
             local_shapes=local_shapes, engine=self._engine, dtype=self.dtype)
+
+    def reshape(self, local_shape, axis: int = 0):
+        """ref :899-944 — re-view the local block; the new global shape is
+        stitched from the allgathered local shapes."""
+        local_shape = tuple(int(v) for v in (
+            (local_shape,) if isinstance(local_shape, Integral)
+            else local_shape))
+        local_shapes = [tuple(int(v) for v in s) for s in
+                        self._base_comm.allgather_obj(local_shape)]
+        ref_shape = local_shapes[0]
+        if self._partition is Partition.SCATTER:
+            if (local_shape[:axis] != ref_shape[:axis]
+                    or local_shape[axis + 1:] != ref_shape[axis + 1:]):
+                raise ValueError(
+                    f"All local shapes must match on every axis except "
+                    f"axis={axis}. Got {local_shape} in rank {self.rank} "
+                    f"and {ref_shape} in rank 0.")
+            global_shape = list(ref_shape)
+            global_shape[axis] = sum(ls[axis] for ls in local_shapes)
+        else:
+            if local_shape != ref_shape:
+                raise ValueError(
+                    "All local shapes must be identical for "
+                    "Partition.BROADCAST and Partition.UNSAFE_BROADCAST. "
+                    f"Got {local_shape} in rank {self.rank} and "
+                    f"{ref_shape} in rank 0.")
+            global_shape = list(ref_shape)
+        return DistributedArray(
+            tuple(global_shape), self._base_comm, self._partition, axis,
+            local_array=self._local_array.reshape(local_shapes[self.rank]),
+            local_shapes=local_shapes, mask=self._mask,
+            engine=self._engine, dtype=self.dtype)
 
     def redistribute(self, axis: int):
         # ref :493-552 — on the hot path the output of every stencil is

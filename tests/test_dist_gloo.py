@@ -169,7 +169,7 @@ def body_blockdiag(c):
     "body_allreduce", "body_allgather_obj", "body_allgather_tensors",
     "body_to_dist_asarray", "body_ghost_cells", "body_rebalance",
     "body_sendrecv", "body_blockdiag", "body_fredholm", "body_vstack",
-    "body_post_neighbors_overlap",
+    "body_post_neighbors_overlap", "body_mask_subcomm",
 ])
 def test_gloo_world2(body):
     _spawn(body)
@@ -250,3 +250,29 @@ def body_post_neighbors_overlap(c):
     else:
         assert torch.all(gf == 1.0)
     assert float(acc) == 12.0 * (c.rank + 1)
+
+
+def body_mask_subcomm(c):
+    """Masked sub-communicator semantics (ref DistributedArray.py:74-100):
+    world 2 with mask [0, 1] -> each rank reduces/gathers alone."""
+    import numpy as np
+    import pylops_mpi_amd as pm
+    d = pm.DistributedArray((4,), c, mask=[0, 1])
+    assert d.sub_comm.size == 1 and d.mask == [0, 1]
+    d[:] = float(c.rank + 1)
+    # masked asarray sees only this group's piece
+    ma = d.asarray(masked=True)
+    assert ma.numel() == 2 and torch.all(ma == float(c.rank + 1))
+    # same-mask group of 2: reductions span both ranks
+    d2 = pm.DistributedArray((4,), c, mask=[0, 0])
+    assert d2.sub_comm.size == 2
+    t = torch.tensor([float(c.rank + 1)])
+    d2.sub_comm.allreduce_(t, "sum")
+    assert t.item() == 3.0
+    # mask mismatch raises like ref :581-585
+    import pytest as _pt
+    with _pt.raises(ValueError, match="Mask of both the arrays"):
+        d._check_mask(d2)
+    # reshape round trip (ref :899-944)
+    r = d2.reshape((2, 1))
+    assert r.global_shape == (4, 1) and r.local_shape == (2, 1)
