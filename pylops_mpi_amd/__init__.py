@@ -10,6 +10,7 @@ from .distributedarray import (DistributedArray, Partition,  # noqa: F401
 from .linearoperator import (MPILinearOperator,  # noqa: F401
                              asmpilinearoperator)
 from . import deps  # noqa: F401
+from .benchmark import benchmark, mark  # noqa: F401
 from .blockdiag import MPIBlockDiag  # noqa: F401
 from . import matmult  # noqa: F401
 from .matmult import (MPIMatrixMult, active_grid_comm,  # noqa: F401

@@ -103,3 +103,29 @@ def test_reshaped_partition_check():
     x = pm.DistributedArray((8,), partition=pm.Partition.UNSAFE_BROADCAST)
     with pytest.raises(ValueError, match="should have partition"):
         op.matvec(x)
+
+
+def test_benchmark_decorator(capsys):
+    # ref utils/benchmark.py:76-173 semantics: nested trees + mark()
+    import pylops_mpi_amd as pm
+
+    @pm.benchmark(description="inner")
+    def inner():
+        pm.mark("i0")
+        pm.mark("i1")
+
+    @pm.benchmark(description="outer")
+    def outer():
+        pm.mark("a")
+        inner()
+        pm.mark("b")
+        return 7
+
+    assert outer() == 7
+    out = capsys.readouterr().out
+    assert "[decorator]outer: total runtime" in out
+    assert "[decorator]inner" in out
+    assert "a-->b" in out and "i0-->i1" in out
+    import pytest as _pt
+    with _pt.raises(RuntimeError, match="outside of a benchmarked region"):
+        pm.mark("stray")
