@@ -148,6 +148,16 @@ int pam_fd_apply(void* stream, int op, int edge, const void* x,
                  int64_t m, int64_t row0, int64_t nglob, int64_t rbegin,
                  int64_t rend, double coeff, int dtype);
 
+/* Serial non-stationary 1-D convolution along axis d of [batch, d, m]
+ * (the local operator of MPINonStationaryConvolve1D's Halo sandwich,
+ * ref signalprocessing/NonStatConvolve1d.py:129-168): filters hs[nf][hsize]
+ * anchored at oh + q*dh, linearly interpolated per position, clamped at
+ * the ends.  forward != 0: y[n] = sum_t x[n+hh-t] h_{n+hh-t}[t];
+ * adjoint: y[n] = sum_t x[n-hh+t] h_n[t]. */
+int pam_nsconv(void* stream, int forward, const void* x, void* y,
+               const void* hs, int64_t batch, int64_t d, int64_t m,
+               int64_t nf, int64_t hsize, double oh, double dh, int dtype);
+
 /* Serial (single-rank) stencil along an arbitrary axis: the block is
  * viewed as [batch, d, m] with the derivative along d.  The local
  * operators inside MPIBlockDiag for Gradient/Laplacian axes >= 1

@@ -48,3 +48,5 @@ from .cgls import sim_cgls, sim_cg  # noqa: F401
 from .blockdiag import SimBlockDiag  # noqa: F401
 from .fredholm import (SimFredholm1, SimMDC,  # noqa: F401
                        serial_rfft_adj, serial_rfft_op)
+from .nsconv import (serial_nsconv_mv,  # noqa: F401
+                     serial_nsconv_rmv)

@@ -17,6 +17,8 @@ from .matmult import (MPIMatrixMult, active_grid_comm,  # noqa: F401
                       block_gather, local_block_split)
 from .fredholm import MPIFredholm1  # noqa: F401
 from .halo import MPIHalo  # noqa: F401
+from .nonstatconv import (MPINonStationaryConvolve1D,  # noqa: F401
+                          NonStationaryConvolve1DLocal, halo_block_split)
 from .fdlocal import (FirstDerivativeLocal,  # noqa: F401
                       SecondDerivativeLocal)
 from .gradient import MPIGradient, MPILaplacian  # noqa: F401

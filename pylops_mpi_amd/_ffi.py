@@ -74,6 +74,11 @@ _SIGS = {
                        ctypes.c_int64, ctypes.c_int64, ctypes.c_int],
                       ctypes.c_int),
     "pam_fd_halo_width": ([ctypes.c_int], ctypes.c_int64),
+    "pam_nsconv": ([ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+                    ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
+                    ctypes.c_int64, ctypes.c_int64, ctypes.c_int64,
+                    ctypes.c_int64, ctypes.c_double, ctypes.c_double,
+                    ctypes.c_int], ctypes.c_int),
     "pam_fd_serial": ([ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
                        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
                        ctypes.c_int64, ctypes.c_int64, ctypes.c_double,

@@ -133,3 +133,81 @@ extern "C" int pam_fd_serial(void* stream, int op, int edge, const void* x,
   return PAM_EDTYPE;
 }
 
+
+// ---------------------------------------------------------------------------
+// serial non-stationary 1-D convolution along axis d of a [batch, d, m]
+// block (the local operator inside MPINonStationaryConvolve1D's
+// Halo.H @ BlockDiag @ Halo sandwich, ref signalprocessing/
+// NonStatConvolve1d.py:129-168).  pylops' published convention, re-derived
+// (pylops is not vendored; locked by dense-adjoint/dottest tests):
+//   filters hs[nf][hsize] (odd hsize, centred) anchored at ih = oh + q*dh;
+//   the filter at position ix is the linear interpolation of the two
+//   nearest anchors, clamped to hs[0] / hs[nf-1] outside;
+//   forward (scatter form): y[n] += x[ix] * h_ix[n - ix + hh]
+//   => gather: y[n] = sum_t x[n+hh-t] * h_{n+hh-t}[t]
+//   adjoint: z[n] = sum_t x[n-hh+t] * h_n[t]
+// ---------------------------------------------------------------------------
+template <typename T>
+__device__ __forceinline__ T nsc_h(const T* __restrict__ hs, int64_t nf,
+                                   int64_t hsize, double oh, double dh,
+                                   int64_t ix, int64_t t) {
+  const double q = ((double)ix - oh) / dh;
+  const int64_t ic = (int64_t)floor(q);
+  if (ic < 0) return hs[t];
+  if (ic >= nf - 1) return hs[(nf - 1) * hsize + t];
+  const T w = (T)(q - (double)ic);
+  return ((T)1 - w) * hs[ic * hsize + t] + w * hs[(ic + 1) * hsize + t];
+}
+
+template <typename T, bool FWD>
+__global__ void __launch_bounds__(BLK) nsconv_kernel(
+    const T* __restrict__ x, T* __restrict__ y, const T* __restrict__ hs,
+    int64_t batch, int64_t d, int64_t m, int64_t nf, int64_t hsize,
+    double oh, double dh) {
+  const int64_t hh = hsize / 2;
+  const int64_t total = batch * d * m;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += stride) {
+    const int64_t j = idx % m;
+    const int64_t rem = idx / m;
+    const int64_t n = rem % d;
+    const T* Xb = x + (rem - n) * m;
+    T acc = (T)0;
+    for (int64_t t = 0; t < hsize; ++t) {
+      const int64_t ix = FWD ? (n + hh - t) : (n - hh + t);
+      if (ix < 0 || ix >= d) continue;
+      const T h = nsc_h<T>(hs, nf, hsize, oh, dh, FWD ? ix : n, t);
+      acc += Xb[ix * m + j] * h;
+    }
+    y[idx] = acc;
+  }
+}
+
+extern "C" int pam_nsconv(void* stream, int forward, const void* x, void* y,
+                          const void* hs, int64_t batch, int64_t d,
+                          int64_t m, int64_t nf, int64_t hsize, double oh,
+                          double dh, int dtype) {
+  if (batch <= 0 || d <= 0 || m <= 0 || nf <= 0 || hsize <= 0 || !x || !y ||
+      !hs)
+    return PAM_EARG;
+  hipStream_t s = (hipStream_t)stream;
+#define NSC_CASE(T)                                                           \
+  do {                                                                        \
+    if (forward)                                                              \
+      hipLaunchKernelGGL((nsconv_kernel<T, true>),                            \
+                         dim3(grid_1d_s(batch * d * m)), dim3(BLK), 0, s,     \
+                         (const T*)x, (T*)y, (const T*)hs, batch, d, m, nf,   \
+                         hsize, oh, dh);                                      \
+    else                                                                      \
+      hipLaunchKernelGGL((nsconv_kernel<T, false>),                           \
+                         dim3(grid_1d_s(batch * d * m)), dim3(BLK), 0, s,     \
+                         (const T*)x, (T*)y, (const T*)hs, batch, d, m, nf,   \
+                         hsize, oh, dh);                                      \
+    return check_s(hipGetLastError());                                        \
+  } while (0)
+  if (dtype == PAM_F64) NSC_CASE(double);
+  if (dtype == PAM_F32) NSC_CASE(float);
+#undef NSC_CASE
+  return PAM_EDTYPE;
+}

@@ -132,3 +132,40 @@ def test_laplacian_vs_serial():
     u = pm.DistributedArray.to_dist(dev(rng.standard_normal(n)))
     v = pm.DistributedArray.to_dist(dev(rng.standard_normal(n)))
     assert pm.dottest(op, u, v, rtol=1e-10)
+
+
+# ------------------------------------------------- non-stationary convolution
+@pytest.mark.parametrize("dims,axis", [((24,), 0), ((3, 24), 1),
+                                       ((24, 4), 0)])
+def test_nsconv_local_vs_serial(dims, axis):
+    from oracle import serial_nsconv_mv, serial_nsconv_rmv
+    rng = np.random.default_rng(20)
+    hsize = 7
+    ih = np.array([3, 9, 15, 21])
+    hs = rng.standard_normal((len(ih), hsize))
+    n = int(np.prod(dims))
+    xg = rng.standard_normal(n)
+    op = pm.NonStationaryConvolve1DLocal(dims, dev(hs), ih, axis=axis)
+    assert_allclose(host(op.matvec(dev(xg))),
+                    serial_nsconv_mv(xg, dims, hs, ih, axis),
+                    rtol=1e-12, atol=1e-13)
+    assert_allclose(host(op.rmatvec(dev(xg))),
+                    serial_nsconv_rmv(xg, dims, hs, ih, axis),
+                    rtol=1e-12, atol=1e-13)
+
+
+def test_mpinonstatconv_world1():
+    from oracle import serial_nsconv_mv, serial_nsconv_rmv
+    rng = np.random.default_rng(21)
+    dims = (32,)
+    ih = np.array([4, 12, 20, 28])
+    hs = rng.standard_normal((len(ih), 5))
+    op = pm.MPINonStationaryConvolve1D(dims, dev(hs), ih, axis=0)
+    xg = rng.standard_normal(32)
+    x = pm.DistributedArray.to_dist(dev(xg))
+    assert_allclose(host(op.matvec(x).asarray()),
+                    serial_nsconv_mv(xg, dims, hs, ih, 0),
+                    rtol=1e-12, atol=1e-13)
+    u = pm.DistributedArray.to_dist(dev(rng.standard_normal(32)))
+    v = pm.DistributedArray.to_dist(dev(rng.standard_normal(32)))
+    assert pm.dottest(op, u, v, rtol=1e-10)

@@ -262,3 +262,26 @@ def test_fd1_complex_vs_serial(P):
     y = op.matvec(to_dist(xg, P))
     want = serial_fd1_matvec(xg.reshape(dims), 1.5, "centered", True, 5)
     assert_allclose(y.asarray().reshape(dims), want, rtol=1e-14)
+
+
+def test_nsconv_adjoint_vs_dense():
+    """Non-stationary convolution restatement: rmatvec == A.T @ x where
+    A is the explicit forward matrix (locks the re-derived pylops
+    interpolation convention)."""
+    from oracle import serial_nsconv_mv, serial_nsconv_rmv
+    rng = np.random.default_rng(15)
+    dims, hsize = (24,), 7
+    ih = np.array([3, 9, 15, 21])
+    hs = rng.standard_normal((len(ih), hsize))
+    A = dense_matrix_from_matvec(
+        lambda v: serial_nsconv_mv(v, dims, hs, ih), 24)
+    v = rng.standard_normal(24)
+    assert_allclose(serial_nsconv_rmv(v, dims, hs, ih), A.T @ v,
+                    rtol=1e-12, atol=1e-13)
+    # batched axis=-1 of a 2-D block
+    dims2 = (3, 24)
+    A2 = dense_matrix_from_matvec(
+        lambda v: serial_nsconv_mv(v, dims2, hs, ih, axis=-1), 72)
+    v2 = rng.standard_normal(72)
+    assert_allclose(serial_nsconv_rmv(v2, dims2, hs, ih, axis=-1),
+                    A2.T @ v2, rtol=1e-12, atol=1e-13)
