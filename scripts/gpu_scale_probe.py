@@ -85,5 +85,60 @@ def main():
                       "GB/s": gbs}), flush=True)
 
 
-if __name__ == "__main__":
+if __name__ == "__main__" and not os.environ.get("PAM_PROBE_EXTRA"):
     main()
+
+
+def extra():
+    """Mix-ceiling references + the full cfg2 cgls(50)."""
+    init_default_comm(torch.device("cuda:0"))
+    n = 1 << 29  # 4 GiB fp64 working set
+    x = torch.rand(n, dtype=torch.float64, device="cuda")
+    y = torch.rand(n, dtype=torch.float64, device="cuda")
+    s = torch.cuda.current_stream().cuda_stream
+
+    def axpy():
+        _ffi.checked(_ffi.lib().pam_axpy(s, y.data_ptr(), x.data_ptr(),
+                                         0.5, n, 0), "axpy")
+    sec = timeit(axpy, iters=10)
+    print(json.dumps({"probe": "axpy_f64_2R1W", "ms": sec * 1e3,
+                      "GB/s": 24.0 * n / sec / 1e9}), flush=True)
+
+    import pylops_mpi_amd as pm
+    ws = torch.empty(2048, dtype=torch.float64, device="cuda")
+    out = torch.empty(2, dtype=torch.float64, device="cuda")
+
+    def dot():
+        _ffi.checked(_ffi.lib().pam_dot(s, x.data_ptr(), y.data_ptr(), n,
+                                        ws.data_ptr(), out.data_ptr(), 0),
+                     "dot")
+    sec = timeit(dot, iters=10)
+    print(json.dumps({"probe": "dot_f64_2R", "ms": sec * 1e3,
+                      "GB/s": 16.0 * n / sec / 1e9}), flush=True)
+    del x, y
+    torch.cuda.empty_cache()
+
+    # cfg2: MPIFirstDerivative + cgls(niter=50) end to end (1 GPU)
+    import numpy as np
+    dims = (2048, 2048, 128)
+    ntot = int(np.prod(dims))
+    op = pm.MPIFirstDerivative(dims, kind="centered", order=3)
+    g = torch.Generator(device="cuda").manual_seed(42)
+    yd = pm.DistributedArray((ntot,))
+    yd[:] = torch.randn(ntot, generator=g, dtype=torch.float64,
+                        device="cuda")
+    x0 = pm.DistributedArray((ntot,))
+    x0[:] = 0.0
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    xs, istop, iit, r1, r2, cost = pm.cgls(op, yd, x0, niter=50, damp=0.1,
+                                           tol=0.0)
+    torch.cuda.synchronize()
+    wall = time.perf_counter() - t0
+    print(json.dumps({"probe": "cfg2_cgls50", "s": wall,
+                      "iters": int(iit), "s_per_iter": wall / max(iit, 1)}),
+          flush=True)
+
+
+if __name__ == "__main__" and os.environ.get("PAM_PROBE_EXTRA"):
+    extra()
