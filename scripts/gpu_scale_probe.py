@@ -91,6 +91,7 @@ if __name__ == "__main__" and not os.environ.get("PAM_PROBE_EXTRA"):
 
 def extra():
     """Mix-ceiling references + the full cfg2 cgls(50)."""
+    from pylops_mpi_amd import _ffi
     init_default_comm(torch.device("cuda:0"))
     n = 1 << 29  # 4 GiB fp64 working set
     x = torch.rand(n, dtype=torch.float64, device="cuda")
