@@ -632,9 +632,12 @@ class DistributedArray:
 
     # ------------------------------------------------------------ structure
     def conj(self):
-        # ref :840-854
+        # ref :840-854.  Real dtypes: conj is the identity, so (following
+        # torch.conj) the result SHARES storage instead of copying — the
+        # reference clones 4.3 GB per CGLS dot here (r.dot(r.conj()),
+        # ref cls_basic.py:389-401); treat the result as read-only.
         if not self._is_cplx():
-            return self.copy()
+            return self._like(self._local_array)
         self._require_compute()
         out = self._like()
         _ffi.checked(_ffi.lib().pam_conj(

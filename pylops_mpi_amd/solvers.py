@@ -89,8 +89,9 @@ class CGLS:
         self.niter = niter
         x = x0.copy()
         self.s = self.y - self.Op.matvec(x)
-        damped_x = x * damp
-        r = self.Op.rmatvec(self.s) - damped_x
+        r = self.Op.rmatvec(self.s)
+        if damp != 0.0:
+            r.iaxpy_(-damp, x)  # r -= damp * x, fused (ref :347-348)
         self.rank = x.rank
         self.c = r.copy()
         self.q = self.Op.matvec(self.c)
@@ -109,10 +110,9 @@ class CGLS:
                                       + self.damp * self.c.dot(self.c.conj()))))
         x.iaxpy_(a, self.c)          # x += a * c        (ref :390)
         self.s.iaxpy_(-a, self.q)    # s -= a * q        (ref :391)
+        r = self.Op.rmatvec(self.s)
         if self.damp != 0.0:
-            r = self.Op.rmatvec(self.s) - self.damp * x
-        else:
-            r = self.Op.rmatvec(self.s)
+            r.iaxpy_(-self.damp, x)  # r -= damp^2 * x, fused (ref :392-393)
         k = float(np.abs(r.dot(r.conj())))
         b = float(k / self.kold)
         self.c.xpby_(r, b)           # c = r + b * c     (ref :396)
