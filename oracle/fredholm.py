@@ -11,7 +11,7 @@ pylops_mpi_amd/fftlocal.py) and locked by the adjoint/round-trip tests in
 tests/test_oracle_fredholm.py.
 """
 import math
-from typing import List, Sequence
+from typing import Sequence
 
 import numpy as np
 

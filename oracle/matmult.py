@@ -11,7 +11,7 @@ arithmetic of /root/reference/pylops_mpi/basicoperators/MatrixMult.py:
   summa tiles       -> ref :82-129 local_block_split (ceil blocks)
 """
 import math
-from typing import List, Tuple
+from typing import List
 
 import numpy as np
 

@@ -10,8 +10,6 @@ code — so they pin oracle/stencils.py independently.
 All stencils act along axis 0 of an array of shape ``dims`` and broadcast
 over the remaining axes.
 """
-from typing import Tuple
-
 import numpy as np
 
 

@@ -14,13 +14,11 @@ and SecondDerivative.py, re-designed MI355X-first:
     the reference's two object allgathers per apply disappear, and the
     balanced case is a zero-copy view.
 """
-from typing import Union
-
 import numpy as np
 import torch
 
 from . import _ffi
-from .distributedarray import (DistributedArray, Partition, as_torch_dtype,
+from .distributedarray import (DistributedArray, Partition,
                                local_split)
 from .linearoperator import MPILinearOperator
 from .rebalance import rebalance_1d

@@ -4,8 +4,6 @@ for Gradient/Laplacian axes >= 1, ref basicoperators/Gradient.py:109-116,
 Laplacian.py:107-125).  Same stencil tables/edge semantics as the
 distributed kernels (csrc fd_serial_kernel), along any axis of a local
 block of shape ``dims``."""
-from typing import Tuple, Union
-
 import numpy as np
 import torch
 

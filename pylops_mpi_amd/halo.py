@@ -146,7 +146,6 @@ class MPIHalo(MPILinearOperator):
     def _exchange_along_axis(self, arr: torch.Tensor, axis: int,
                              before: int, after: int) -> None:
         """ref :320-360 — two pairwise swaps per axis, batched isend/irecv."""
-        import torch.distributed as dist
         minus, plus = self.neigh[("-", axis)], self.neigh[("+", axis)]
         comm = self.comm
         sl = [slice(None)] * self.ndim

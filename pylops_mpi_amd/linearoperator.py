@@ -16,8 +16,9 @@ from .distributedarray import DistributedArray
 
 
 class MPILinearOperator:
-    """ref LinearOperator.py:16-405 (without the serial-pylops wrapping —
-    local serial operators arrive with MPIBlockDiag in a later round)."""
+    """ref LinearOperator.py:16-405.  ``Op=`` wraps a serial
+    :class:`~pylops_mpi_amd.localops.LocalOperator` applied identically on
+    every rank (the reference's pylops-operator wrap, ref :22-27)."""
 
     def __init__(self, Op=None, shape=None, dims=None, dimsd=None,
                  dtype=None, base_comm: Optional[PamComm] = None):

@@ -36,9 +36,6 @@ from .comm import PamComm, get_default_comm
 from .distributedarray import DistributedArray, Partition, as_torch_dtype
 from .linearoperator import MPILinearOperator
 
-_T2NP = {torch.float64: np.float64, torch.float32: np.float32}
-
-
 def _stream(t: torch.Tensor):
     return torch.cuda.current_stream(t.device).cuda_stream
 

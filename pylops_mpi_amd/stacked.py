@@ -6,7 +6,7 @@ the operator side of StackedLinearOperator.py (thin Python over the member
 arrays — no new kernels).  Adds the fused iaxpy_/xpby_
 used by our CG/CGLS so stacked systems drive the same solver loop.
 """
-from typing import List, Optional, Union
+from typing import List, Optional
 
 import numpy as np
 
