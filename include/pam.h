@@ -71,6 +71,26 @@ int pam_axpy(void* stream, void* y, const void* x, double alpha, int64_t n,
 int pam_xpby(void* stream, void* y, const void* x, double beta, int64_t n,
              int dtype);
 
+/* Device-scalar solver fast path.  alpha/beta are 1-element f64 DEVICE
+ * buffers written by a prior pam_dot / pam_scalar_* launch on the same
+ * stream; scale is a host constant (+-1.0, exact).  Together with the two
+ * recurrence-scalar kernels below, a whole CG/CGLS iteration
+ * (ref optimization/cls_basic.py:370-404) is launched with a single host
+ * synchronization — the stop-test readback of k — instead of one blocking
+ * readback per dot/norm. */
+/* y += scale * (*alpha) * x */
+int pam_axpy_d(void* stream, void* y, const void* x, const void* alpha,
+               double scale, int64_t n, int dtype);
+/* y = x + scale * (*beta) * y */
+int pam_xpby_d(void* stream, void* y, const void* x, const void* beta,
+               double scale, int64_t n, int dtype);
+/* *out = |num[0] / (den[0] + damp * den[1])| — a = |kold / (q.q + damp^2
+ * c.c)|, ref cls_basic.py:379-383 (CG: damp == 0, ref :121-123). */
+int pam_scalar_alpha(void* stream, void* out, const void* num,
+                     const void* den, double damp);
+/* *out = |num[0] / den[0]| — b = k / kold, ref cls_basic.py:394-395. */
+int pam_scalar_div(void* stream, void* out, const void* num, const void* den);
+
 /* Complex element-wise ops (interleaved storage).  add/sub/neg/fill and
  * real-alpha axpy/xpby on complex arrays are the REAL kernels above on the
  * 2n-float view (the solvers' a,b scalars are real, ref cls_basic.py:389).

@@ -36,6 +36,20 @@ _SIGS = {
     "pam_xpby": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                   ctypes.c_double, ctypes.c_int64, ctypes.c_int],
                  ctypes.c_int),
+    "pam_axpy_d": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                    ctypes.c_void_p, ctypes.c_double, ctypes.c_int64,
+                    ctypes.c_int],
+                   ctypes.c_int),
+    "pam_xpby_d": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                    ctypes.c_void_p, ctypes.c_double, ctypes.c_int64,
+                    ctypes.c_int],
+                   ctypes.c_int),
+    "pam_scalar_alpha": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                          ctypes.c_void_p, ctypes.c_double],
+                         ctypes.c_int),
+    "pam_scalar_div": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                        ctypes.c_void_p],
+                       ctypes.c_int),
     "pam_dot": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                  ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
                  ctypes.c_int], ctypes.c_int),

@@ -59,10 +59,10 @@ static inline int64_t grid_1d(int64_t work) {
 // EW op codes: 0 fill, 1 neg, 2 add, 3 sub, 4 mul, 5 scale, 6 axpy, 7 xpby
 // ---------------------------------------------------------------------------
 template <typename T, int OP, int V>
-__global__ void __launch_bounds__(BLK) ew_kernel(T* __restrict__ y,
-                                                 const T* __restrict__ a,
-                                                 const T* __restrict__ b,
-                                                 T alpha, int64_t n) {
+__device__ __forceinline__ void ew_body(T* __restrict__ y,
+                                        const T* __restrict__ a,
+                                        const T* __restrict__ b,
+                                        T alpha, int64_t n) {
   const int64_t nv = n / V;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
@@ -106,6 +106,27 @@ __global__ void __launch_bounds__(BLK) ew_kernel(T* __restrict__ y,
     else if constexpr (OP == 6) y[i] = y[i] + alpha * va;
     else if constexpr (OP == 7) y[i] = va + alpha * y[i];
   }
+}
+
+template <typename T, int OP, int V>
+__global__ void __launch_bounds__(BLK) ew_kernel(T* __restrict__ y,
+                                                 const T* __restrict__ a,
+                                                 const T* __restrict__ b,
+                                                 T alpha, int64_t n) {
+  ew_body<T, OP, V>(y, a, b, alpha, n);
+}
+
+// device-scalar variant: alpha lives in HBM (1-elem f64 written by a prior
+// reduction / pam_scalar_* kernel on the same stream) so the CG/CGLS
+// recurrence (ref cls_basic.py:370-404) runs with no host round-trip per
+// update.  scale is a host constant (+-1.0): (T)(scale * *ap) is exact for
+// sign flips, so results are bit-identical to the host-scalar kernels.
+template <typename T, int OP, int V>
+__global__ void __launch_bounds__(BLK) ewd_kernel(T* __restrict__ y,
+                                                  const T* __restrict__ a,
+                                                  const double* __restrict__ ap,
+                                                  double scale, int64_t n) {
+  ew_body<T, OP, V>(y, a, nullptr, (T)(scale * ap[0]), n);
 }
 
 template <typename T, int OP>
@@ -159,6 +180,85 @@ EW_ENTRY(pam_axpy(void* stream, void* y, const void* x, double alpha,
 EW_ENTRY(pam_xpby(void* stream, void* y, const void* x, double beta,
                   int64_t n, int dtype),
          7, x, nullptr, beta)
+
+// ---------------------------------------------------------------------------
+// device-scalar solver fast path (ref optimization/cls_basic.py:370-404):
+// axpy/xpby whose scalar is read from device memory, plus the two tiny
+// recurrence-scalar kernels.  Together these let a whole CG/CGLS iteration
+// be launched with a single host synchronization (the stop-test readback).
+// ---------------------------------------------------------------------------
+template <typename T, int OP>
+static int ewd_launch(void* stream, void* y, const void* a, const void* ap,
+                      double scale, int64_t n) {
+  if (n < 0 || !y || !a || !ap) return PAM_EARG;
+  if (n == 0) return 0;
+  constexpr int V = VecW<T>::value;
+  const bool aligned =
+      ((uintptr_t)y % 16 == 0) && ((uintptr_t)a % 16 == 0);
+  hipStream_t s = (hipStream_t)stream;
+  const int64_t grid = grid_1d(n / (aligned ? V : 1) + 1);
+  if (aligned)
+    hipLaunchKernelGGL((ewd_kernel<T, OP, V>), dim3(grid), dim3(BLK), 0, s,
+                       (T*)y, (const T*)a, (const double*)ap, scale, n);
+  else
+    hipLaunchKernelGGL((ewd_kernel<T, OP, 1>), dim3(grid), dim3(BLK), 0, s,
+                       (T*)y, (const T*)a, (const double*)ap, scale, n);
+  return check(hipGetLastError());
+}
+
+// y += scale * (*alpha) * x
+extern "C" int pam_axpy_d(void* stream, void* y, const void* x,
+                          const void* alpha, double scale, int64_t n,
+                          int dtype) {
+  if (dtype == PAM_F64) return ewd_launch<double, 6>(stream, y, x, alpha, scale, n);
+  if (dtype == PAM_F32) return ewd_launch<float, 6>(stream, y, x, alpha, scale, n);
+  return PAM_EDTYPE;
+}
+
+// y = x + scale * (*beta) * y
+extern "C" int pam_xpby_d(void* stream, void* y, const void* x,
+                          const void* beta, double scale, int64_t n,
+                          int dtype) {
+  if (dtype == PAM_F64) return ewd_launch<double, 7>(stream, y, x, beta, scale, n);
+  if (dtype == PAM_F32) return ewd_launch<float, 7>(stream, y, x, beta, scale, n);
+  return PAM_EDTYPE;
+}
+
+// *out = |num[0] / (den[0] + damp * den[1])| — the CG/CGLS step scalars
+// a = |kold / (q.q + damp^2 c.c)| (ref cls_basic.py:379-383; CG a with
+// damp==0, ref :121-123).  |x/y| == |x|/|y| in IEEE, so this matches the
+// reference's np.abs placement for the non-negative numerators involved.
+__global__ void scalar_alpha_kernel(double* __restrict__ out,
+                                    const double* __restrict__ num,
+                                    const double* __restrict__ den,
+                                    double damp) {
+  out[0] = fabs(num[0] / (den[0] + damp * den[1]));
+}
+
+// *out = |num[0] / den[0]| — b = k/kold (ref cls_basic.py:394-395)
+__global__ void scalar_div_kernel(double* __restrict__ out,
+                                  const double* __restrict__ num,
+                                  const double* __restrict__ den) {
+  out[0] = fabs(num[0] / den[0]);
+}
+
+extern "C" int pam_scalar_alpha(void* stream, void* out, const void* num,
+                                const void* den, double damp) {
+  if (!out || !num || !den) return PAM_EARG;
+  hipLaunchKernelGGL(scalar_alpha_kernel, dim3(1), dim3(1), 0,
+                     (hipStream_t)stream, (double*)out, (const double*)num,
+                     (const double*)den, damp);
+  return check(hipGetLastError());
+}
+
+extern "C" int pam_scalar_div(void* stream, void* out, const void* num,
+                              const void* den) {
+  if (!out || !num || !den) return PAM_EARG;
+  hipLaunchKernelGGL(scalar_div_kernel, dim3(1), dim3(1), 0,
+                     (hipStream_t)stream, (double*)out, (const double*)num,
+                     (const double*)den);
+  return check(hipGetLastError());
+}
 
 // ---------------------------------------------------------------------------
 // complex element-wise (interleaved re,im).  EWC op: 0 cmul, 1 cscale, 2 conj

@@ -6,6 +6,10 @@ pylops_mpi/utils/deps.py:58-66 (NCCL_PYLOPS_MPI / PYLOPS_MPI_CUDA_AWARE):
                           interior rows (debug aid)
   PAM_FD_VEC={1,2,4}      force the stencil kernel's per-lane vector width
                           (default: measured optimum, 16 B/lane)
+  PAM_DISABLE_DEVSCALARS=1  run CG/CGLS with host-side scalars (one
+                          blocking readback per dot) instead of the
+                          single-sync device-scalar iteration (debug aid;
+                          both paths are bit-identical)
 
 The RCCL data plane itself has no gate: it IS the backend (there is no
 MPI in this stack to fall back to)."""
@@ -18,3 +22,4 @@ def env_flag(name: str, default: bool = False) -> bool:
 
 
 overlap_enabled = not env_flag("PAM_DISABLE_OVERLAP")
+devscalars_enabled = not env_flag("PAM_DISABLE_DEVSCALARS")

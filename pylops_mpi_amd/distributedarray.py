@@ -547,6 +547,45 @@ class DistributedArray:
             "xpby")
         return self
 
+    # device-scalar solver fast path: the scalar lives in a caller-owned
+    # 1-element f64 CUDA buffer written by a prior dot_into / pam_scalar_*
+    # launch on the same stream, so a CG/CGLS iteration needs one host sync
+    # (the stop test) instead of one per dot (see solvers.py).
+    def iaxpy_dev_(self, alpha_t: torch.Tensor, x: "DistributedArray",
+                   scale: float = 1.0) -> "DistributedArray":
+        """self += scale * alpha_t[0] * x (bit-identical to iaxpy_)."""
+        self._require_compute()
+        a, n, dt = self._ew()
+        b, _, _ = self._ew(x._local_array)
+        _ffi.checked(_ffi.lib().pam_axpy_d(
+            self._stream(), a.data_ptr(), b.data_ptr(), alpha_t.data_ptr(),
+            float(scale), n, dt), "axpy_d")
+        return self
+
+    def xpby_dev_(self, x: "DistributedArray", beta_t: torch.Tensor,
+                  scale: float = 1.0) -> "DistributedArray":
+        """self = x + scale * beta_t[0] * self (bit-identical to xpby_)."""
+        self._require_compute()
+        a, n, dt = self._ew()
+        b, _, _ = self._ew(x._local_array)
+        _ffi.checked(_ffi.lib().pam_xpby_d(
+            self._stream(), a.data_ptr(), b.data_ptr(), beta_t.data_ptr(),
+            float(scale), n, dt), "xpby_d")
+        return self
+
+    def dot_into(self, other: "DistributedArray",
+                 out_t: torch.Tensor) -> torch.Tensor:
+        """Local dot reduced into ``out_t`` (1-elem f64 device slice); the
+        global allreduce is the CALLER's, so the solver can batch several
+        dots into one collective.  Real SCATTER arrays only."""
+        self._require_compute()
+        ws, _ = _reduce_buffers(self._local_array.device)
+        _ffi.checked(_ffi.lib().pam_dot(
+            self._stream(), self._flat().data_ptr(), other._flat().data_ptr(),
+            self._local_array.numel(), ws.data_ptr(), out_t.data_ptr(),
+            self._dt()), "dot")
+        return out_t
+
     def __add__(self, x):
         return self.add(x)
 

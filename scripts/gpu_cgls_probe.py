@@ -1,0 +1,49 @@
+"""Time CGLS end-to-end at the bench config (1 GPU), device-scalar vs
+host-scalar iteration (solvers.py module docstring), and check the two
+cost traces agree bitwise.  Run via gpurun; prints one line per path.
+"""
+import os
+import time
+
+import numpy as np
+import torch
+
+import pylops_mpi_amd as pm
+from pylops_mpi_amd.comm import init_default_comm
+
+
+def main():
+    init_default_comm(torch.device("cuda:0"))
+    dims = (2048, 2048, 128)
+    n = int(np.prod(dims))
+    op = pm.MPIFirstDerivative(dims, kind="centered", order=3)
+    g = torch.Generator(device="cuda").manual_seed(1)
+    y = pm.DistributedArray((n,))
+    y[:] = torch.randn(n, generator=g, dtype=torch.float64, device="cuda")
+    x0 = pm.DistributedArray((n,))
+    x0[:] = 0.0
+
+    traces = {}
+    for tag in ("dev", "host"):
+        if tag == "host":
+            os.environ["PAM_DISABLE_DEVSCALARS"] = "1"
+        else:
+            os.environ.pop("PAM_DISABLE_DEVSCALARS", None)
+        pm.cgls(op, y, x0.copy(), niter=3, damp=0.1, tol=0.0)  # warmup
+        torch.cuda.synchronize()
+        t = time.perf_counter()
+        _, _, _, _, _, cost = pm.cgls(op, y, x0.copy(), niter=50, damp=0.1,
+                                      tol=0.0)
+        torch.cuda.synchronize()
+        dt = time.perf_counter() - t
+        traces[tag] = np.asarray(cost)
+        print(f"cgls[{tag:4s}] niter=50 dims={dims}: {dt:.3f} s total, "
+              f"{dt / 50 * 1e3:.2f} ms/iter, cost[-1]={cost[-1]:.17e}")
+    os.environ.pop("PAM_DISABLE_DEVSCALARS", None)
+    same = np.array_equal(traces["dev"], traces["host"])
+    print(f"traces bitwise equal: {same}")
+    assert same
+
+
+if __name__ == "__main__":
+    main()

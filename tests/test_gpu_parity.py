@@ -363,3 +363,59 @@ def test_fd1_complex_vs_oracle(kind, order):
     v = pm.DistributedArray.to_dist(
         dev(rng.standard_normal(n) + 1j * rng.standard_normal(n)))
     assert pm.dottest(op, u, v, rtol=1e-10)
+
+
+def test_devscalar_solver_bitwise_vs_host_path():
+    """The single-sync device-scalar CG/CGLS iteration (solvers.py
+    _step_dev) must be BIT-IDENTICAL to the host-scalar reference
+    recurrence — same dots, same |x/y| scalar algebra, same fused
+    updates — so toggling PAM_DISABLE_DEVSCALARS never changes a trace."""
+    dims = (64, 12)
+    n = int(np.prod(dims))
+    rng = np.random.default_rng(29)
+    yg = rng.standard_normal(n)
+
+    def run_cgls(damp):
+        op = pm.MPIFirstDerivative(dims, kind="centered", order=5)
+        y = pm.DistributedArray.to_dist(dev(yg))
+        x0 = pm.DistributedArray((n,))
+        x0[:] = 0.0
+        return pm.cgls(op, y, x0, niter=25, damp=damp, tol=0.0)
+
+    def run_cg():
+        op = pm.MPIFirstDerivative(dims, kind="centered", order=5)
+
+        class Normal(pm.MPILinearOperator):
+            def __init__(self):
+                super().__init__(shape=(n, n), dtype=np.float64)
+
+            def _matvec(self, x):
+                return op.rmatvec(op.matvec(x)) + 0.05 * x
+
+            _rmatvec = _matvec
+
+        y = pm.DistributedArray.to_dist(dev(yg))
+        x0 = pm.DistributedArray((n,))
+        x0[:] = 0.0
+        return pm.cg(Normal(), y, x0, niter=25, tol=0.0)
+
+    for damp in (0.0, 0.7):
+        os.environ.pop("PAM_DISABLE_DEVSCALARS", None)
+        xd, _, _, r1d, r2d, costd = run_cgls(damp)
+        os.environ["PAM_DISABLE_DEVSCALARS"] = "1"
+        try:
+            xh, _, _, r1h, r2h, costh = run_cgls(damp)
+        finally:
+            os.environ.pop("PAM_DISABLE_DEVSCALARS", None)
+        assert np.array_equal(np.asarray(costd), np.asarray(costh)), damp
+        assert r1d == r1h and r2d == r2h
+        assert torch.equal(xd.local_array, xh.local_array)
+
+    xd, _, costd = run_cg()
+    os.environ["PAM_DISABLE_DEVSCALARS"] = "1"
+    try:
+        xh, _, costh = run_cg()
+    finally:
+        os.environ.pop("PAM_DISABLE_DEVSCALARS", None)
+    assert np.array_equal(np.asarray(costd), np.asarray(costh))
+    assert torch.equal(xd.local_array, xh.local_array)
