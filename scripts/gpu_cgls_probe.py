@@ -3,10 +3,13 @@ host-scalar iteration (solvers.py module docstring), and check the two
 cost traces agree bitwise.  Run via gpurun; prints one line per path.
 """
 import os
+import sys
 import time
 
 import numpy as np
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import pylops_mpi_amd as pm
 from pylops_mpi_amd.comm import init_default_comm

@@ -419,3 +419,40 @@ def test_devscalar_solver_bitwise_vs_host_path():
         os.environ.pop("PAM_DISABLE_DEVSCALARS", None)
     assert np.array_equal(np.asarray(costd), np.asarray(costh))
     assert torch.equal(xd.local_array, xh.local_array)
+
+
+def test_composite_operator_algebra_vs_dense():
+    """Mirror of the reference's composite-operator numerics tests
+    (ref tests/test_linearop.py:75-292: transpose/scaled/power/sum/
+    product/conj) against the explicit dense serial matrices."""
+    n = 40
+    dims = (n,)
+    op1 = pm.MPIFirstDerivative(dims, kind="centered", order=3)
+    op2 = pm.MPISecondDerivative(dims, kind="centered")
+    D1 = oracle.dense_matrix_from_matvec(oracle.serial_fd1_matvec, n)
+    D2 = oracle.dense_matrix_from_matvec(oracle.serial_fd2_matvec, n)
+    rng = np.random.default_rng(31)
+    x = rng.standard_normal(n)
+    cases = [
+        ("scaled", 2.5 * op1, 2.5 * D1),
+        ("neg", -op1, -D1),
+        ("sum", op1 + op2, D1 + D2),
+        ("sub", op1 - op2, D1 - D2),
+        ("product", op1 @ op2, D1 @ D2),
+        ("power", op1 ** 2, D1 @ D1),
+        ("adjoint", op1.H, D1.T),
+        ("transpose", op1.T, D1.T),
+        ("conj", op1.conj(), D1),
+    ]
+    for name, o, Dm in cases:
+        xd = pm.DistributedArray.to_dist(dev(x))
+        assert_allclose(host(o.matvec(xd).asarray()), Dm @ x,
+                        rtol=1e-12, atol=1e-13, err_msg=f"{name} fwd")
+        xd = pm.DistributedArray.to_dist(dev(x))
+        assert_allclose(host(o.rmatvec(xd).asarray()), Dm.T @ x,
+                        rtol=1e-12, atol=1e-13, err_msg=f"{name} adj")
+    # dottest on every composite (the reference's own gate)
+    for name, o, _ in cases:
+        u = pm.DistributedArray.to_dist(dev(rng.standard_normal(n)))
+        v = pm.DistributedArray.to_dist(dev(rng.standard_normal(n)))
+        assert pm.dottest(o, u, v, rtol=1e-10), name
