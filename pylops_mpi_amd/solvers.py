@@ -18,9 +18,13 @@ MI355X-first differences (both bit-identical to the reference recurrence):
   pam_xpby_d).  The whole iteration is launched back-to-back with a SINGLE
   host synchronization — the k readback the reference's stop test
   (ref cls_basic.py:433) requires — instead of one blocking readback per
-  dot/norm (5 per CGLS iteration).  Measured r01: 31.7 -> ~22 ms/iter at
-  the 512x4096x256 config, the removed ~10 ms being host stall.
-  The device path covers real SCATTER mask-free CUDA arrays (the north
+  dot/norm (5 per CGLS iteration).  Measured r01 (scripts/
+  gpu_cgls_probe.py): at the bench config the iteration is GPU-bound
+  (~20 ms of kernels) and both paths time ~20.4-20.5 ms/iter, but at
+  small sizes where sync latency dominates the device path is ~2x
+  faster/iter, and at N>1 it keeps collective latency off the host
+  critical path.  The device path covers real SCATTER mask-free CUDA
+  arrays (the north
   star); anything else (complex, BROADCAST, masked, stacked, CPU/gloo
   tests) takes the host-scalar path.  PAM_DISABLE_DEVSCALARS=1 forces the
   host path; the two produce bit-identical iterates and cost traces
