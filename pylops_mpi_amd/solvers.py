@@ -20,11 +20,11 @@ MI355X-first differences (both bit-identical to the reference recurrence):
   (ref cls_basic.py:433) requires — instead of one blocking readback per
   dot/norm (5 per CGLS iteration).  Measured r01 (scripts/
   gpu_cgls_probe.py): at the bench config the iteration is GPU-bound
-  (~20 ms of kernels) and both paths time ~20.4-20.5 ms/iter, but at
-  small sizes where sync latency dominates the device path is ~2x
-  faster/iter, and at N>1 it keeps collective latency off the host
-  critical path.  The device path covers real SCATTER mask-free CUDA
-  arrays (the north
+  (~20 ms of kernels) and the paths are within a few percent
+  (20.3 vs 21.1 ms/iter); at latency-bound sizes (256x256x64 and below)
+  the device path is ~35% faster/iter (0.23 vs 0.31 ms), and at N>1 it
+  keeps collective latency off the host critical path.  The device path
+  covers real SCATTER mask-free CUDA arrays (the north
   star); anything else (complex, BROADCAST, masked, stacked, CPU/gloo
   tests) takes the host-scalar path.  PAM_DISABLE_DEVSCALARS=1 forces the
   host path; the two produce bit-identical iterates and cost traces
