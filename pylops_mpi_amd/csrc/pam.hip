@@ -766,7 +766,7 @@ extern "C" int64_t pam_fd_halo_width(int op) {
   }
 }
 
-template <typename T, int OP, int V>
+template <typename T, int OP, int V, bool NTS = false>
 __global__ void __launch_bounds__(BLK) fd_kernel(Rows<T> R, T* __restrict__ y,
                                                  int64_t row0, int64_t N, T c,
                                                  int edge, int64_t rbegin,
@@ -795,7 +795,7 @@ __global__ void __launch_bounds__(BLK) fd_kernel(Rows<T> R, T* __restrict__ y,
       if (edge) fd_edge<T, OP, V>(R, i, j, g, N, acc);
 #pragma unroll
       for (int k = 0; k < V; ++k) acc[k] *= c;
-      storev<T, V>(yrow + j, acc);
+      storev_p<T, V, NTS>(yrow + j, acc);
     }
     // scalar tail columns (m not divisible by V)
     if constexpr (V > 1) {
@@ -809,7 +809,11 @@ __global__ void __launch_bounds__(BLK) fd_kernel(Rows<T> R, T* __restrict__ y,
             acc[0] += (T)tm.coeff * R.row(i + tm.off)[j];
         }
         if (edge) fd_edge<T, OP, 1>(R, i, j, g, N, acc);
-        yrow[j] = acc[0] * c;
+        const T o = acc[0] * c;
+        if constexpr (NTS)
+          __builtin_nontemporal_store(o, yrow + j);
+        else
+          yrow[j] = o;
       }
     }
   }
@@ -818,6 +822,20 @@ __global__ void __launch_bounds__(BLK) fd_kernel(Rows<T> R, T* __restrict__ y,
 static int fd_vec_override() {
   static int v = [] {
     const char* e = getenv("PAM_FD_VEC");
+    return e ? atoi(e) : 0;
+  }();
+  return v;
+}
+
+// PAM_FD_NT=1 opts the output stores into the nt cache hint.  Measured
+// NEGATIVE on the production kernel (bench A/B r01: 291.2 plain vs 290.5
+// nt pairs/s — no change): the +9% the standalone probe saw
+// (scripts/probe_nt_store.hip) does not transfer once the launch geometry
+// gives the row re-use its L2 locality; the kernel is already at the
+// platform's read+write mix ceiling.  Default: plain stores.
+static int fd_nt_override() {
+  static int v = [] {
+    const char* e = getenv("PAM_FD_NT");
     return e ? atoi(e) : 0;
   }();
   return v;
@@ -854,7 +872,18 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
   if (gx64 > cap) gx64 = cap;
   dim3 grid((uint32_t)gx64, (uint32_t)gy);
   hipStream_t s = (hipStream_t)stream;
-  if (V == 4)
+  const bool nt = fd_nt_override() != 0;
+  if (nt) {
+    if (V == 4)
+      hipLaunchKernelGGL((fd_kernel<T, OP, 4, true>), grid, dim3(BLK), 0, s,
+                         R, (T*)y, row0, nglob, (T)coeff, edge, rbegin, rend);
+    else if (V == 2 && sizeof(T) == 8)
+      hipLaunchKernelGGL((fd_kernel<T, OP, 2, true>), grid, dim3(BLK), 0, s,
+                         R, (T*)y, row0, nglob, (T)coeff, edge, rbegin, rend);
+    else
+      hipLaunchKernelGGL((fd_kernel<T, OP, 1, true>), grid, dim3(BLK), 0, s,
+                         R, (T*)y, row0, nglob, (T)coeff, edge, rbegin, rend);
+  } else if (V == 4)
     hipLaunchKernelGGL((fd_kernel<T, OP, 4>), grid, dim3(BLK), 0, s, R, (T*)y,
                        row0, nglob, (T)coeff, edge, rbegin, rend);
   else if (V == 2 && sizeof(T) == 8)

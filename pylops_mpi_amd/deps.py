@@ -6,6 +6,9 @@ pylops_mpi/utils/deps.py:58-66 (NCCL_PYLOPS_MPI / PYLOPS_MPI_CUDA_AWARE):
                           interior rows (debug aid)
   PAM_FD_VEC={1,2,4}      force the stencil kernel's per-lane vector width
                           (default: measured optimum, 16 B/lane)
+  PAM_FD_NT=1             nt cache hint on the stencil's output stores
+                          (measured no-change on the production kernel —
+                          documented negative, csrc/pam.hip)
   PAM_DISABLE_DEVSCALARS=1  run CG/CGLS with host-side scalars (one
                           blocking readback per dot) instead of the
                           single-sync device-scalar iteration (debug aid;
