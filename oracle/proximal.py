@@ -1,0 +1,260 @@
+"""Serial restatements of the proximal subpackage — TEST INFRASTRUCTURE
+ONLY.
+
+Restates, in plain NumPy, ref pylops_mpi/proximal/: the separable local
+operators the reference takes from pyproximal (an UNPINNED dependency
+absent from /root/reference — Box/L0/L1 restated from pyproximal's
+published definitions, anchored on ref tests/test_prox.py:64-228), the
+MPIL2 operator (ref proximal/proximal/L2.py:15-192) and the
+ProximalGradient/ADMML2 solvers (ref proximal/optimization/
+primal.py:22-347).  Serial == world-1 distributed, the comparison the
+reference's own tests make on rank 0 (ref tests/test_proxsolver.py:
+143-161).
+"""
+from math import sqrt
+from typing import Optional
+
+import numpy as np
+
+
+# ------------------------------------------------------- local operators
+def soft_threshold(x: np.ndarray, thresh: float) -> np.ndarray:
+    """pylops/pyproximal soft rule (matches pam_thresh kind=0)."""
+    if np.iscomplexobj(x):
+        a = np.abs(x)
+        with np.errstate(divide="ignore", invalid="ignore"):
+            s = np.where(a > 0, np.maximum(a - thresh, 0.0) / a, 0.0)
+        return x * s
+    return np.sign(x) * np.maximum(np.abs(x) - thresh, 0.0)
+
+
+def hard_threshold(x: np.ndarray, thresh: float) -> np.ndarray:
+    """pylops/pyproximal hard rule (matches pam_thresh kind=1)."""
+    return np.where(np.abs(x) >= np.sqrt(2.0 * thresh), x, 0.0)
+
+
+class SerBox:
+    hasgrad = False
+
+    def __init__(self, lower=-np.inf, upper=np.inf):
+        self.lower, self.upper = lower, upper
+
+    def __call__(self, x):
+        return bool(np.all((x >= self.lower) & (x <= self.upper)))
+
+    def prox(self, x, tau):
+        return np.clip(x, self.lower, self.upper)
+
+
+class SerL0:
+    hasgrad = False
+
+    def __init__(self, sigma=1.0):
+        self.sigma = sigma
+
+    def __call__(self, x):
+        return self.sigma * float(np.count_nonzero(x))
+
+    def prox(self, x, tau):
+        return hard_threshold(x, tau * self.sigma)
+
+
+class SerL1:
+    hasgrad = False
+
+    def __init__(self, sigma=1.0):
+        self.sigma = sigma
+
+    def __call__(self, x):
+        return self.sigma * float(np.sum(np.abs(x)))
+
+    def prox(self, x, tau):
+        return soft_threshold(x, tau * self.sigma)
+
+
+# ------------------------------------------------------------ dense CGLS
+def dense_cg(A, y, x0, niter=10, tol=1e-4):
+    """The CG recurrence (ref cls_basic.py:86-141) on dense arrays."""
+    x = x0.copy()
+    r = y - A @ x
+    c = r.copy()
+    kold = float(np.abs(np.dot(r.conj(), r)))
+    iiter = 0
+    while iiter < niter and kold > tol:
+        Ac = A @ c
+        cAc = np.abs(np.dot(c.conj(), Ac))
+        a = float(kold / cAc)
+        x = x + a * c
+        r = r - a * Ac
+        k = float(np.abs(np.dot(r.conj(), r)))
+        b = float(k / kold)
+        c = r + b * c
+        kold = k
+        iiter += 1
+    return x
+
+
+def dense_cgls(A, y, x0, niter=10, damp=0.0, tol=1e-4):
+    """The CGLS recurrence (ref cls_basic.py:308-404) on dense arrays."""
+    damp2 = damp ** 2
+    x = x0.copy()
+    s = y - A @ x
+    r = A.conj().T @ s - damp * x
+    c = r.copy()
+    q = A @ c
+    kold = float(np.abs(np.dot(r.conj(), r)))
+    iiter = 0
+    while iiter < niter and kold > tol:
+        a = float(np.abs(kold / (np.dot(q.conj(), q)
+                                 + damp2 * np.dot(c.conj(), c))))
+        x = x + a * c
+        s = s - a * q
+        r = A.conj().T @ s - damp2 * x
+        k = float(np.abs(np.dot(r.conj(), r)))
+        b = float(k / kold)
+        c = r + b * c
+        q = A @ c
+        kold = k
+        iiter += 1
+    return x
+
+
+# ------------------------------------------------------------------- L2
+class SerL2:
+    """ref proximal/proximal/L2.py:15-192 with a dense Op."""
+
+    hasgrad = True
+
+    def __init__(self, Op: Optional[np.ndarray] = None, b=None, q=None,
+                 sigma=1.0, alpha=1.0, qgrad=True, niter=10, x0=None,
+                 warm=True, solver="cgls", kwargs_solver=None):
+        if Op is not None and x0 is None:
+            raise ValueError("x0 must be passed when Op is not None")
+        self.Op, self.b, self.q = Op, b, q
+        self.sigma, self.alpha, self.qgrad = sigma, alpha, qgrad
+        self.niter, self.x0, self.warm = niter, x0, warm
+        self.normaleqs = solver == "cg"
+        self.kwargs_solver = {} if kwargs_solver is None else kwargs_solver
+        self.count = 0
+        if Op is not None and b is not None and self.normaleqs:
+            self.OpTb = sigma * (Op.conj().T @ b)
+
+    def __call__(self, x):
+        if self.Op is not None and self.b is not None:
+            f = (self.sigma / 2.0) * float(
+                np.linalg.norm(self.Op @ x - self.b)) ** 2
+        elif self.b is not None:
+            f = (self.sigma / 2.0) * float(np.linalg.norm(x - self.b)) ** 2
+        else:
+            f = (self.sigma / 2.0) * float(np.linalg.norm(x)) ** 2
+        if self.q is not None:
+            f += self.alpha * float(np.dot(self.q, x))
+        return float(f)
+
+    def prox(self, x, tau):
+        self.count += 1
+        niter = self.niter if isinstance(self.niter, int) \
+            else self.niter(self.count)
+        if self.Op is not None and self.b is not None:
+            n = x.shape[0]
+            if self.normaleqs:
+                y = x + tau * self.OpTb
+                if self.q is not None:
+                    y = y - (tau * self.alpha) * self.q
+                Op1 = np.eye(n) + float(tau * self.sigma) * (
+                    self.Op.conj().T @ self.Op)
+                x = dense_cg(Op1, y, self.x0, niter=niter,
+                             **self.kwargs_solver)
+            else:
+                y = x
+                if self.q is not None:
+                    y = y - (tau * self.alpha) * self.q
+                Opreg = np.vstack([sqrt(tau * self.sigma) * self.Op,
+                                   np.eye(n)])
+                breg = np.concatenate([sqrt(tau * self.sigma) * self.b, y])
+                x = dense_cgls(Opreg, breg, self.x0, niter=niter,
+                               **self.kwargs_solver)
+            if self.warm:
+                self.x0 = x
+        elif self.b is not None:
+            num = x + (tau * self.sigma) * self.b
+            if self.q is not None:
+                num = num - (tau * self.alpha) * self.q
+            x = (1.0 / (1.0 + tau * self.sigma)) * num
+        else:
+            num = x
+            if self.q is not None:
+                num = num - (tau * self.alpha) * self.q
+            x = (1.0 / (1.0 + tau * self.sigma)) * num
+        return x
+
+    def grad(self, x):
+        if self.Op is not None and self.b is not None:
+            g = self.sigma * (self.Op.conj().T @ (self.Op @ x - self.b))
+        elif self.b is not None:
+            g = self.sigma * (x - self.b)
+        else:
+            g = self.sigma * x
+        if self.q is not None and self.qgrad:
+            g = g + self.alpha * self.q
+        return g
+
+
+# ---------------------------------------------------------------- solvers
+def ser_proximal_gradient(proxf, proxg, x0, epsg=1.0, tau=None, eta=1.0,
+                          niter=10, acceleration=None, tol=None):
+    """ref primal.py:22-201, serial."""
+    epsg = np.asarray(epsg, dtype=float)
+    if epsg.size == 1:
+        epsg = epsg * np.ones(niter)
+    t = 1.0
+    x = x0.copy()
+    y = x.copy()
+    pfg = np.inf
+    for iiter in range(niter):
+        xold = x.copy()
+        if eta == 1.0:
+            x = proxg.prox(y - tau * proxf.grad(y), epsg[iiter] * tau)
+        else:
+            x = x + eta * (
+                proxg.prox(x - tau * proxf.grad(x), epsg[iiter] * tau) - x)
+        if acceleration == "vandenberghe":
+            omega = iiter / (iiter + 3)
+        elif acceleration == "fista":
+            told = t
+            t = (1.0 + np.sqrt(1.0 + 4.0 * t ** 2)) / 2.0
+            omega = (told - 1.0) / t
+        else:
+            omega = 0
+        y = x + omega * (x - xold)
+        if tol is not None:
+            pfgold = pfg
+            pfg = proxf(x) + np.sum(epsg[iiter] * proxg(x))
+            if np.abs(1.0 - pfg / pfgold) < tol:
+                break
+    return x
+
+
+def ser_admml2(proxg, Op, b, A, x0, tau, niter=10, z0=None, gfirst=False,
+               kwargs_solver=None):
+    """ref primal.py:209-347, serial (dense Op and A)."""
+    kwargs_solver = {} if kwargs_solver is None else dict(kwargs_solver)
+    x = x0.copy()
+    z = z0.copy() if z0 is not None else A @ x
+    u = np.zeros_like(z)
+    sqrttau = 1.0 / sqrt(tau)
+    for _ in range(niter):
+        if gfirst:
+            Ax = A @ x
+            z = proxg.prox(Ax + u, tau)
+            Opreg = np.vstack([Op, sqrttau * A])
+            breg = np.concatenate([b, sqrttau * (z - u)])
+            x = dense_cgls(Opreg, breg, x, **kwargs_solver)
+        else:
+            Opreg = np.vstack([Op, sqrttau * A])
+            breg = np.concatenate([b, sqrttau * (z - u)])
+            x = dense_cgls(Opreg, breg, x, **kwargs_solver)
+            Ax = A @ x
+            z = proxg.prox(Ax + u, tau)
+        u = u + Ax - z
+    return x, z

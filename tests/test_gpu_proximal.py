@@ -1,0 +1,229 @@
+"""GPU parity for the proximal subpackage vs the serial oracle
+(world-1 == serial, the comparison the reference's own tests make on
+rank 0 — ref tests/test_prox.py:64-228, tests/test_proxsolver.py:
+103-300)."""
+import numpy as np
+import pytest
+import torch
+from numpy.testing import assert_allclose
+
+import oracle
+import pylops_mpi_amd as pm
+from pylops_mpi_amd.proximal import Box, L0, L1, MPIL2, MPIProxOperator
+from pylops_mpi_amd.proximal.optimization.primal import (ADMML2,
+                                                         ProximalGradient)
+
+pytestmark = pytest.mark.gpu
+
+
+def dev(a):
+    return torch.as_tensor(np.ascontiguousarray(a), device="cuda")
+
+
+def host(t):
+    return t.cpu().numpy()
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _init():
+    from pylops_mpi_amd.comm import init_default_comm
+    init_default_comm(torch.device("cuda:0"))
+
+
+# ------------------------------------------------------- local operators
+def test_local_prox_ops_vs_oracle():
+    """Mirror of ref test_prox.py:64-151 (call/prox per operator)."""
+    rng = np.random.default_rng(7)
+    x = rng.standard_normal(257) * 2.0
+
+    box = Box(lower=0.0, upper=1.0)
+    sb = oracle.SerBox(lower=0.0, upper=1.0)
+    assert box(dev(x)) == sb(x)
+    assert_allclose(host(box.prox(dev(x), 0.4)), sb.prox(x, 0.4))
+
+    l0 = L0(sigma=2.0)
+    s0 = oracle.SerL0(sigma=2.0)
+    assert l0(dev(x)) == s0(x)
+    assert_allclose(host(l0.prox(dev(x), 0.6)), s0.prox(x, 0.6))
+
+    l1 = L1(sigma=2.0)
+    s1 = oracle.SerL1(sigma=2.0)
+    assert_allclose(l1(dev(x)), s1(x), rtol=1e-12)
+    assert_allclose(host(l1.prox(dev(x), 0.6)), s1.prox(x, 0.6),
+                    rtol=1e-13, atol=1e-15)
+
+    # complex soft/hard thresholds (magnitude rule)
+    z = rng.standard_normal(100) + 1j * rng.standard_normal(100)
+    zt = dev(z)
+    assert_allclose(host(l1.prox(zt, 0.3)), oracle.soft_threshold(
+        z, 0.3 * 2.0), rtol=1e-13, atol=1e-15)
+    assert_allclose(host(l0.prox(zt, 0.3)),
+                    np.where(np.abs(z) >= np.sqrt(2 * 0.3 * 2.0), z, 0.0),
+                    rtol=1e-13, atol=1e-15)
+
+    # proxdual via Moreau (base-class rule)
+    pd = host(l1.proxdual(dev(x), 0.7))
+    pd_ref = x - 0.7 * s1.prox(x / 0.7, 1.0 / 0.7)
+    assert_allclose(pd, pd_ref, rtol=1e-12, atol=1e-14)
+
+
+def test_mpiprox_call_and_prox_scatter():
+    """MPIProxOperator on a SCATTER array (world 1): reduction semantics,
+    ref ProxOperator.py:56-121."""
+    rng = np.random.default_rng(8)
+    x = rng.standard_normal(300)
+    xd = pm.DistributedArray.to_dist(dev(x))
+
+    l1d = MPIProxOperator(L1(sigma=0.5))
+    assert_allclose(l1d(xd), oracle.SerL1(0.5)(x), rtol=1e-12)
+    y = l1d.prox(xd, 0.2)
+    assert isinstance(y, pm.DistributedArray)
+    assert_allclose(host(y.asarray()), oracle.SerL1(0.5).prox(x, 0.2),
+                    rtol=1e-13, atol=1e-15)
+
+    boxd = MPIProxOperator(Box(lower=-10.0, upper=10.0))
+    assert boxd(xd) is True
+    boxd2 = MPIProxOperator(Box(lower=0.0, upper=0.1))
+    assert boxd2(xd) is False
+
+    # stacked arrays reduce with python-all / python-sum (ref :106-110)
+    x2 = rng.standard_normal(120)
+    xs = pm.StackedDistributedArray([pm.DistributedArray.to_dist(dev(x)),
+                                     pm.DistributedArray.to_dist(dev(x2))])
+    assert_allclose(l1d(xs), oracle.SerL1(0.5)(x)
+                    + oracle.SerL1(0.5)(x2), rtol=1e-12)
+    ys = l1d.prox(xs, 0.2)
+    assert_allclose(host(ys[1].asarray()),
+                    oracle.SerL1(0.5).prox(x2, 0.2), rtol=1e-13, atol=1e-15)
+
+
+# ------------------------------------------------------------------- L2
+def test_mpil2_closed_forms_and_grad():
+    rng = np.random.default_rng(9)
+    n = 80
+    x = rng.standard_normal(n)
+    b = rng.standard_normal(n)
+    q = rng.standard_normal(n)
+    xd = pm.DistributedArray.to_dist(dev(x))
+    bd = pm.DistributedArray.to_dist(dev(b))
+    qd = pm.DistributedArray.to_dist(dev(q))
+
+    for kw_d, kw_s in [
+        (dict(b=bd, sigma=2.0), dict(b=b, sigma=2.0)),
+        (dict(sigma=1.5), dict(sigma=1.5)),
+        (dict(b=bd, q=qd, sigma=2.0, alpha=0.5),
+         dict(b=b, q=q, sigma=2.0, alpha=0.5)),
+    ]:
+        l2d = MPIL2(**kw_d)
+        l2s = oracle.SerL2(**kw_s)
+        assert_allclose(l2d(xd), l2s(x), rtol=1e-10)
+        assert_allclose(host(l2d.prox(xd, 0.3).asarray()),
+                        l2s.prox(x, 0.3), rtol=1e-12, atol=1e-14)
+        assert_allclose(host(l2d.grad(xd).asarray()), l2s.grad(x),
+                        rtol=1e-12, atol=1e-14)
+
+
+@pytest.mark.parametrize("solver", ["cg", "cgls"])
+def test_mpil2_prox_with_operator(solver):
+    """Mirror of ref test_prox.py:153-228 (L2 with Op, both solvers)."""
+    rng = np.random.default_rng(10)
+    ny, nx = 36, 24
+    A = rng.standard_normal((ny, nx))
+    x = rng.standard_normal(nx)
+    b = rng.standard_normal(ny)
+
+    Op = pm.MPIBlockDiag([pm.DenseLocal(dev(A))])
+    bd = pm.DistributedArray.to_dist(dev(b))
+    x0 = pm.DistributedArray((nx,))
+    x0[:] = 0.0
+    l2d = MPIL2(Op=Op, b=bd, x0=x0, sigma=1.3, niter=15, solver=solver,
+                kwargs_solver={"tol": 0.0})
+    l2s = oracle.SerL2(Op=A, b=b, x0=np.zeros(nx), sigma=1.3, niter=15,
+                       solver=solver, kwargs_solver={"tol": 0.0})
+
+    xd = pm.DistributedArray.to_dist(dev(x))
+    got = host(l2d.prox(xd, 0.4).asarray())
+    want = l2s.prox(x, 0.4)
+    assert_allclose(got, want, rtol=1e-8, atol=1e-10)
+    # warm start: second call starts from the previous solution
+    got2 = host(l2d.prox(xd, 0.4).asarray())
+    want2 = l2s.prox(x, 0.4)
+    assert_allclose(got2, want2, rtol=1e-8, atol=1e-10)
+    # functional value with Op
+    assert_allclose(l2d(xd), l2s(x), rtol=1e-10)
+
+
+# --------------------------------------------------------------- solvers
+@pytest.mark.parametrize("acceleration", [None, "vandenberghe", "fista"])
+def test_proximalgradient_vs_serial(acceleration):
+    """Mirror of ref test_proxsolver.py:166-225 (scatter model)."""
+    rng = np.random.default_rng(11)
+    ny, nx = 40, 30
+    A = rng.standard_normal((ny, nx))
+    xtrue = np.zeros(nx)
+    xtrue[[3, 12, 25]] = [2.0, -1.0, 1.5]
+    b = A @ xtrue
+
+    Op = pm.MPIBlockDiag([pm.DenseLocal(dev(A))])
+    bd = pm.DistributedArray.to_dist(dev(b))
+    x0 = pm.DistributedArray((nx,))
+    x0[:] = 0.0
+    l2d = MPIL2(Op=Op, b=bd, x0=x0)
+    l1d = MPIProxOperator(L1(sigma=1e-1))
+    xinv = ProximalGradient(l2d, l1d, x0=x0, tau=1e-3, niter=120,
+                            acceleration=acceleration)
+    assert isinstance(xinv, pm.DistributedArray)
+
+    l2s = oracle.SerL2(Op=A, b=b, x0=np.zeros(nx))
+    l1s = oracle.SerL1(sigma=1e-1)
+    xref = oracle.ser_proximal_gradient(l2s, l1s, np.zeros(nx), tau=1e-3,
+                                        niter=120,
+                                        acceleration=acceleration)
+    assert_allclose(host(xinv.asarray()), xref, rtol=1e-10, atol=1e-12)
+
+
+def test_proximalgradient_eta_and_tol():
+    rng = np.random.default_rng(12)
+    ny, nx = 30, 20
+    A = rng.standard_normal((ny, nx))
+    b = rng.standard_normal(ny)
+    Op = pm.MPIBlockDiag([pm.DenseLocal(dev(A))])
+    bd = pm.DistributedArray.to_dist(dev(b))
+    x0 = pm.DistributedArray((nx,))
+    x0[:] = 0.0
+    l2d = MPIL2(Op=Op, b=bd, x0=x0)
+    l1d = MPIProxOperator(L1(sigma=1e-1))
+    xinv = ProximalGradient(l2d, l1d, x0=x0, tau=1e-3, niter=80, eta=0.7,
+                            tol=1e-10)
+    l2s = oracle.SerL2(Op=A, b=b, x0=np.zeros(nx))
+    l1s = oracle.SerL1(sigma=1e-1)
+    xref = oracle.ser_proximal_gradient(l2s, l1s, np.zeros(nx), tau=1e-3,
+                                        niter=80, eta=0.7, tol=1e-10)
+    assert_allclose(host(xinv.asarray()), xref, rtol=1e-9, atol=1e-11)
+
+
+@pytest.mark.parametrize("gfirst", [False, True])
+def test_admml2_vs_serial(gfirst):
+    """Mirror of ref test_proxsolver.py:226-300."""
+    rng = np.random.default_rng(13)
+    ny, nx = 36, 24
+    A = rng.standard_normal((ny, nx))
+    b = rng.standard_normal(ny)
+    R = rng.standard_normal((nx, nx)) * 0.1 + np.eye(nx)
+
+    Op = pm.MPIBlockDiag([pm.DenseLocal(dev(A))])
+    Rop = pm.MPIBlockDiag([pm.DenseLocal(dev(R))])
+    bd = pm.DistributedArray.to_dist(dev(b))
+    x0 = pm.DistributedArray((nx,))
+    x0[:] = 0.0
+    l1d = MPIProxOperator(L1(sigma=1e-1))
+    xinv, zinv = ADMML2(l1d, Op, bd, Rop, x0, tau=1.0, niter=8,
+                        gfirst=gfirst,
+                        kwargs_solver={"niter": 6, "tol": 0.0})
+
+    l1s = oracle.SerL1(sigma=1e-1)
+    xref, zref = oracle.ser_admml2(l1s, A, b, R, np.zeros(nx), tau=1.0,
+                                   niter=8, gfirst=gfirst,
+                                   kwargs_solver={"niter": 6, "tol": 0.0})
+    assert_allclose(host(xinv.asarray()), xref, rtol=1e-8, atol=1e-10)
+    assert_allclose(host(zinv.asarray()), zref, rtol=1e-8, atol=1e-10)

@@ -1,0 +1,128 @@
+"""CPU pins for the proximal subpackage: the oracle restatements
+(oracle/proximal.py) against independent mathematical properties, and
+the package's fail-loud / interface checks that need no GPU.
+Mirrors the scheme of ref tests/test_prox.py, test_proxsolver.py."""
+import numpy as np
+import pytest
+import torch
+from numpy.testing import assert_allclose
+
+import oracle
+from pylops_mpi_amd.proximal import Box, L1, MPIProxOperator, ProxOperator
+
+
+# ------------------------------------------------------------- formulas
+def test_soft_threshold_properties():
+    rng = np.random.default_rng(0)
+    x = rng.standard_normal(200) * 3
+    t = 0.7
+    y = oracle.soft_threshold(x, t)
+    # shrinkage by exactly t where |x| > t, zero elsewhere
+    assert_allclose(y[np.abs(x) > t], x[np.abs(x) > t]
+                    - np.sign(x[np.abs(x) > t]) * t)
+    assert np.all(y[np.abs(x) <= t] == 0)
+    assert_allclose(oracle.soft_threshold(x, 0.0), x)
+    # complex: magnitude shrinks by t, phase preserved
+    z = rng.standard_normal(50) + 1j * rng.standard_normal(50)
+    w = oracle.soft_threshold(z, 0.5)
+    big = np.abs(z) > 0.5
+    assert_allclose(np.abs(w[big]), np.abs(z[big]) - 0.5, rtol=1e-13)
+    assert_allclose(np.angle(w[big]), np.angle(z[big]), rtol=1e-13)
+    assert np.all(w[~big] == 0)
+
+
+def test_hard_threshold_properties():
+    rng = np.random.default_rng(1)
+    x = rng.standard_normal(200) * 3
+    t = 0.8
+    y = oracle.hard_threshold(x, t)
+    keep = np.abs(x) >= np.sqrt(2 * t)
+    assert_allclose(y[keep], x[keep])
+    assert np.all(y[~keep] == 0)
+
+
+def test_serl2_closed_form_optimality():
+    # prox_{tau f}(x) with f = (sigma/2)||v-b||^2 must satisfy
+    # sigma*(v-b) + (v-x)/tau = 0
+    rng = np.random.default_rng(2)
+    x = rng.standard_normal(64)
+    b = rng.standard_normal(64)
+    sigma, tau = 2.0, 0.3
+    l2 = oracle.SerL2(b=b, sigma=sigma)
+    v = l2.prox(x, tau)
+    assert_allclose(sigma * (v - b) + (v - x) / tau, 0.0, atol=1e-12)
+    # functional value
+    assert_allclose(l2(x), sigma / 2 * np.sum((x - b) ** 2), rtol=1e-12)
+
+
+def test_dense_cgls_solves_lstsq():
+    rng = np.random.default_rng(3)
+    A = rng.standard_normal((40, 12))
+    y = rng.standard_normal(40)
+    x = oracle.dense_cgls(A, y, np.zeros(12), niter=60, tol=0.0)
+    xref = np.linalg.lstsq(A, y, rcond=None)[0]
+    assert_allclose(x, xref, rtol=1e-8, atol=1e-10)
+
+
+def test_ser_proximal_gradient_lasso():
+    # LASSO objective decreases and solution is sparse-ish
+    rng = np.random.default_rng(4)
+    A = rng.standard_normal((30, 20))
+    xtrue = np.zeros(20)
+    xtrue[[2, 7, 11]] = [3.0, -2.0, 1.5]
+    b = A @ xtrue
+    proxf = oracle.SerL2(Op=None, b=None)  # placeholder, not used below
+    l2 = oracle.SerL2(Op=A, b=b, x0=np.zeros(20), niter=10)
+
+    class GradOnly:
+        def grad(self, x):
+            return A.T @ (A @ x - b)
+
+        def __call__(self, x):
+            return 0.5 * np.sum((A @ x - b) ** 2)
+
+    g = oracle.SerL1(sigma=0.05)
+    x = oracle.ser_proximal_gradient(GradOnly(), g, np.zeros(20),
+                                     tau=1e-2, niter=300)
+    obj = 0.5 * np.sum((A @ x - b) ** 2) + 0.05 * np.sum(np.abs(x))
+    obj0 = 0.5 * np.sum(b ** 2)
+    assert obj < 0.05 * obj0
+    assert l2 is not None and proxf is not None  # silence linters
+
+
+def test_ser_admml2_fixed_point():
+    rng = np.random.default_rng(5)
+    A = rng.standard_normal((25, 15))
+    b = rng.standard_normal(25)
+    R = np.eye(15)
+    g = oracle.SerL1(sigma=0.1)
+    x, z = oracle.ser_admml2(g, A, b, R, np.zeros(15), tau=1.0, niter=40,
+                             kwargs_solver={"niter": 20, "tol": 0.0})
+    # z is the soft-thresholded copy of x (consensus up to ADMM tolerance)
+    assert np.linalg.norm(x - z) < 1e-2 * max(1.0, np.linalg.norm(x))
+
+
+# ------------------------------------------------- package CPU behavior
+def test_mpiprox_rejects_nonseparable():
+    class Weird(ProxOperator):
+        pass
+
+    with pytest.raises(NotImplementedError, match="not a separable"):
+        MPIProxOperator(Weird())
+
+
+def test_box_matches_oracle_on_cpu_tensors():
+    # Box uses torch ops only (device-agnostic plumbing)
+    rng = np.random.default_rng(6)
+    x = rng.standard_normal(100)
+    box = Box(lower=-0.5, upper=0.5)
+    sbox = oracle.SerBox(lower=-0.5, upper=0.5)
+    xt = torch.from_numpy(x)
+    assert box(xt) == sbox(x)
+    assert box(torch.clamp(xt, -0.5, 0.5)) is True
+    assert_allclose(box.prox(xt, 0.3).numpy(), sbox.prox(x, 0.3))
+
+
+def test_l1_prox_fails_loudly_on_cpu():
+    with pytest.raises(RuntimeError, match="CUDA"):
+        L1(0.1).prox(torch.randn(8, dtype=torch.float64), 0.5)
