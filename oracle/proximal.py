@@ -74,14 +74,17 @@ class SerL1:
 
 # ------------------------------------------------------------ dense CGLS
 def dense_cg(A, y, x0, niter=10, tol=1e-4):
-    """The CG recurrence (ref cls_basic.py:86-141) on dense arrays."""
+    """The CG recurrence (ref cls_basic.py:86-141) on dense arrays.
+    ``A`` may be a matrix or a callable matvec (to mirror op-by-op
+    application of composite operators, as pylops/pyproximal do)."""
+    mv = A if callable(A) else (lambda v: A @ v)
     x = x0.copy()
-    r = y - A @ x
+    r = y - mv(x)
     c = r.copy()
     kold = float(np.abs(np.dot(r.conj(), r)))
     iiter = 0
     while iiter < niter and kold > tol:
-        Ac = A @ c
+        Ac = mv(c)
         cAc = np.abs(np.dot(c.conj(), Ac))
         a = float(kold / cAc)
         x = x + a * c
@@ -95,13 +98,20 @@ def dense_cg(A, y, x0, niter=10, tol=1e-4):
 
 
 def dense_cgls(A, y, x0, niter=10, damp=0.0, tol=1e-4):
-    """The CGLS recurrence (ref cls_basic.py:308-404) on dense arrays."""
+    """The CGLS recurrence (ref cls_basic.py:308-404) on dense arrays.
+    ``A`` may be a matrix or an ``(mv, rmv)`` callable pair (to mirror
+    op-by-op application of composite/stacked operators)."""
+    if isinstance(A, tuple):
+        mv, rmv = A
+    else:
+        mv = lambda v: A @ v           # noqa: E731
+        rmv = lambda s: A.conj().T @ s  # noqa: E731
     damp2 = damp ** 2
     x = x0.copy()
-    s = y - A @ x
-    r = A.conj().T @ s - damp * x
+    s = y - mv(x)
+    r = rmv(s) - damp * x
     c = r.copy()
-    q = A @ c
+    q = mv(c)
     kold = float(np.abs(np.dot(r.conj(), r)))
     iiter = 0
     while iiter < niter and kold > tol:
@@ -109,11 +119,11 @@ def dense_cgls(A, y, x0, niter=10, damp=0.0, tol=1e-4):
                                  + damp2 * np.dot(c.conj(), c))))
         x = x + a * c
         s = s - a * q
-        r = A.conj().T @ s - damp2 * x
+        r = rmv(s) - damp2 * x
         k = float(np.abs(np.dot(r.conj(), r)))
         b = float(k / kold)
         c = r + b * c
-        q = A @ c
+        q = mv(c)
         kold = k
         iiter += 1
     return x
@@ -161,18 +171,28 @@ class SerL2:
                 y = x + tau * self.OpTb
                 if self.q is not None:
                     y = y - (tau * self.alpha) * self.q
-                Op1 = np.eye(n) + float(tau * self.sigma) * (
-                    self.Op.conj().T @ self.Op)
-                x = dense_cg(Op1, y, self.x0, niter=niter,
+                # functional normal-equations op, mirroring the composite
+                # Iop + tau*sigma*(Op.H @ Op) applied op-by-op (ref
+                # L2.py:149-155, pylops composite semantics)
+                A, ts = self.Op, float(tau * self.sigma)
+                mv = lambda v: v.copy() + ts * (   # noqa: E731
+                    A.conj().T @ (A @ v))
+                x = dense_cg(mv, y, self.x0, niter=niter,
                              **self.kwargs_solver)
             else:
                 y = x
                 if self.q is not None:
                     y = y - (tau * self.alpha) * self.q
-                Opreg = np.vstack([sqrt(tau * self.sigma) * self.Op,
-                                   np.eye(n)])
-                breg = np.concatenate([sqrt(tau * self.sigma) * self.b, y])
-                x = dense_cgls(Opreg, breg, self.x0, niter=niter,
+                # functional stacked op, mirroring MPIStackedVStack
+                # [sqrt(tau*sigma)*Op, Iop] (ref L2.py:156-170)
+                A, c0 = self.Op, sqrt(tau * self.sigma)
+                ny = A.shape[0]
+                mv = lambda v: np.concatenate(    # noqa: E731
+                    [c0 * (A @ v), v.copy()])
+                rmv = lambda s: c0 * (            # noqa: E731
+                    A.conj().T @ s[:ny]) + s[ny:]
+                breg = np.concatenate([c0 * self.b, y])
+                x = dense_cgls((mv, rmv), breg, self.x0, niter=niter,
                                **self.kwargs_solver)
             if self.warm:
                 self.x0 = x
@@ -243,17 +263,21 @@ def ser_admml2(proxg, Op, b, A, x0, tau, niter=10, z0=None, gfirst=False,
     z = z0.copy() if z0 is not None else A @ x
     u = np.zeros_like(z)
     sqrttau = 1.0 / sqrt(tau)
+    ny = Op.shape[0]
+    # functional stacked op, mirroring MPIStackedVStack [Op, sqrttau*A]
+    mv = lambda v: np.concatenate(          # noqa: E731
+        [Op @ v, sqrttau * (A @ v)])
+    rmv = lambda s: Op.conj().T @ s[:ny] + sqrttau * (  # noqa: E731
+        A.conj().T @ s[ny:])
     for _ in range(niter):
         if gfirst:
             Ax = A @ x
             z = proxg.prox(Ax + u, tau)
-            Opreg = np.vstack([Op, sqrttau * A])
             breg = np.concatenate([b, sqrttau * (z - u)])
-            x = dense_cgls(Opreg, breg, x, **kwargs_solver)
+            x = dense_cgls((mv, rmv), breg, x, **kwargs_solver)
         else:
-            Opreg = np.vstack([Op, sqrttau * A])
             breg = np.concatenate([b, sqrttau * (z - u)])
-            x = dense_cgls(Opreg, breg, x, **kwargs_solver)
+            x = dense_cgls((mv, rmv), breg, x, **kwargs_solver)
             Ax = A @ x
             z = proxg.prox(Ax + u, tau)
         u = u + Ax - z
