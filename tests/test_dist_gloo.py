@@ -170,6 +170,7 @@ def body_blockdiag(c):
     "body_to_dist_asarray", "body_ghost_cells", "body_rebalance",
     "body_sendrecv", "body_blockdiag", "body_fredholm", "body_vstack",
     "body_post_neighbors_overlap", "body_mask_subcomm", "body_nonstatconv",
+    "body_proximal_call_reduction",
 ])
 def test_gloo_world2(body):
     _spawn(body)
@@ -314,3 +315,34 @@ def body_nonstatconv(c):
     gotr = op.rmatvec(y).asarray().numpy()
     np.testing.assert_allclose(gotr, serial_nsconv_rmv(yg, dims, hs, ih, 0),
                                rtol=1e-12, atol=1e-13)
+
+
+def body_proximal_call_reduction(c):
+    """MPIProxOperator.__call__ reduction semantics across 2 ranks
+    (ref proximal/ProxOperator.py:56-110: Box -> LAND, L0/L1 -> SUM).
+    The functional evaluations are torch ops, so this runs on gloo/CPU;
+    the prox kernels themselves are GPU-only (test_gpu_proximal)."""
+    import numpy as np
+    from pylops_mpi_amd import DistributedArray
+    from pylops_mpi_amd.proximal import Box, L1, MPIProxOperator
+    rng = np.random.default_rng(80 + c.rank)
+    nloc = 40
+    xl = rng.standard_normal(nloc)
+    x = DistributedArray((2 * nloc,), c)
+    x[:] = torch.from_numpy(xl)
+    # L1: global sum of per-rank sums
+    l1d = MPIProxOperator(L1(sigma=0.5))
+    want = 0.0
+    for r in range(2):
+        want += 0.5 * np.abs(
+            np.random.default_rng(80 + r).standard_normal(nloc)).sum()
+    got = l1d(x)
+    np.testing.assert_allclose(got, want, rtol=1e-12)
+    # Box: logical AND across ranks — make it fail on rank 1 only
+    y = DistributedArray((2 * nloc,), c)
+    y[:] = torch.zeros(nloc, dtype=torch.float64) + (5.0 if c.rank else 0.5)
+    boxd = MPIProxOperator(Box(lower=0.0, upper=1.0))
+    assert boxd(y) is False
+    z = DistributedArray((2 * nloc,), c)
+    z[:] = torch.full((nloc,), 0.5, dtype=torch.float64)
+    assert boxd(z) is True
