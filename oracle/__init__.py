@@ -54,3 +54,4 @@ from .proximal import (soft_threshold, hard_threshold,  # noqa: F401
                        SerBox, SerL0, SerL1, SerL2,
                        dense_cg, dense_cgls,
                        ser_proximal_gradient, ser_admml2)
+from .fftnd import serial_fftnd_mv, serial_fftnd_rmv  # noqa: F401

@@ -37,6 +37,8 @@ from .localops import AdjointLocal  # noqa: F401
 from .dottest import dottest  # noqa: F401
 from .comm import (PamComm, get_default_comm,  # noqa: F401
                    init_default_comm)
+from .fftnd import MPIFFTND, MPIFFT2D  # noqa: F401
+from .fft_helper import fftshift_nd, ifftshift_nd  # noqa: F401
 from . import proximal  # noqa: F401  (ref pylops_mpi/proximal/)
 
 __version__ = "0.1.0"

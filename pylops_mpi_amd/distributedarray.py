@@ -741,11 +741,37 @@ class DistributedArray:
             engine=self._engine, dtype=self.dtype)
 
     def redistribute(self, axis: int):
-        # ref :493-552 — on the hot path the output of every stencil is
-        # already axis-0 so this is the no-op branch (ref :516-517)
+        """ref :493-552 — all-to-all realignment of the distribution axis
+        (the FFT pencil transpose).  The result is balanced along the new
+        axis even if this array is unbalanced (ref :504-506).  The
+        reference serializes P pairwise sendrecvs (ref :534-551); here
+        every (src,dst) block is posted in ONE batched RCCL isend/irecv
+        group so xGMI links run concurrently."""
         if self._axis == axis or self._partition is not Partition.SCATTER:
             return self
-        raise NotImplementedError("redistribute to a new axis deferred")
+        out = DistributedArray(self._global_shape, self._base_comm,
+                               self._partition, axis, mask=self._mask,
+                               engine=self._engine, dtype=self.dtype)
+        counts_from = [s[self._axis] for s in self._all_local_shapes]
+        counts_to = [s[axis] for s in out._all_local_shapes]
+        offs = np.cumsum([0] + counts_to[:-1])
+        sends, recvs = [], []
+        for r in range(self.size):
+            sl = [slice(None)] * self._local_array.ndim
+            sl[axis] = slice(int(offs[r]), int(offs[r]) + counts_to[r])
+            sends.append(self._local_array[tuple(sl)].contiguous())
+            shp = list(self._global_shape)
+            shp[self._axis] = counts_from[r]
+            shp[axis] = counts_to[self.rank]
+            recvs.append(torch.empty(
+                shp, dtype=self._local_array.dtype,
+                device=self._local_array.device))
+        recvs[self.rank].copy_(sends[self.rank])
+        self._base_comm.exchange(
+            [(t, r) for r, t in enumerate(sends) if r != self.rank],
+            [(t, r) for r, t in enumerate(recvs) if r != self.rank])
+        out[:] = torch.cat(recvs, dim=self._axis)
+        return out
 
     def __repr__(self):
         return (f"<DistributedArray with global shape={self.global_shape}, "
