@@ -40,5 +40,7 @@ from .comm import (PamComm, get_default_comm,  # noqa: F401
 from .fftnd import MPIFFTND, MPIFFT2D  # noqa: F401
 from .fft_helper import fftshift_nd, ifftshift_nd  # noqa: F401
 from . import proximal  # noqa: F401  (ref pylops_mpi/proximal/)
+from .plotting import (plot_distributed_array,  # noqa: F401
+                       plot_local_arrays)
 
 __version__ = "0.1.0"

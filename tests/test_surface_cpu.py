@@ -129,3 +129,26 @@ def test_benchmark_decorator(capsys):
     import pytest as _pt
     with _pt.raises(RuntimeError, match="outside of a benchmarked region"):
         pm.mark("stray")
+
+
+def test_plotting_helpers():
+    """ref plotting/plotting.py:13-75 — smoke the matplotlib wrappers
+    (Agg backend, world 1, CPU tensors: visualization is not compute)."""
+    import matplotlib
+    matplotlib.use("Agg")
+    from matplotlib import pyplot as plt
+    import torch
+    import pylops_mpi_amd as pm
+    arr = pm.DistributedArray((6, 5))
+    arr[:] = torch.arange(30, dtype=torch.float64).reshape(6, 5)
+    pm.plot_distributed_array(arr)
+    pm.plot_local_arrays(arr, title="locals", vmin=0, vmax=30)
+    assert plt.get_fignums()
+    plt.close("all")
+    import pytest
+    b = pm.DistributedArray((4, 4), partition=pm.Partition.BROADCAST)
+    b[:] = torch.zeros(4, 4, dtype=torch.float64)
+    with pytest.raises(NotImplementedError, match="Use Scatter"):
+        pm.plot_distributed_array(b)
+    with pytest.raises(TypeError, match="Not a DistributedArray"):
+        pm.plot_distributed_array(torch.zeros(3, 3))
