@@ -171,6 +171,7 @@ def body_blockdiag(c):
     "body_sendrecv", "body_blockdiag", "body_fredholm", "body_vstack",
     "body_post_neighbors_overlap", "body_mask_subcomm", "body_nonstatconv",
     "body_proximal_call_reduction", "body_redistribute", "body_fftnd",
+    "body_norm_axis",
 ])
 def test_gloo_world2(body):
     _spawn(body)
@@ -411,3 +412,21 @@ def body_fftnd(c):
                                         norm=par["norm"], real=par["real"])
         np.testing.assert_allclose(z.asarray().numpy(), z_ref,
                                    rtol=1e-10, atol=1e-11)
+
+
+def body_norm_axis(c):
+    """norm(ord, axis) across 2 ranks vs np.linalg.norm on the global
+    array (ref DistributedArray.py:719-771,828-838), both when the
+    reduced axis is local (allgather+concat) and when it is the
+    distributed axis (elementwise allreduce)."""
+    import numpy as np
+    from pylops_mpi_amd import DistributedArray
+    rng = np.random.default_rng(70)
+    g = rng.standard_normal((7, 5, 4))
+    x = DistributedArray.to_dist(torch.from_numpy(g), c)
+    for ordv in (1, 2, np.inf, 0):
+        for ax in (0, 1, 2):
+            got = x.norm(ord=ordv, axis=ax).numpy()
+            want = np.linalg.norm(g, ord=ordv, axis=ax)
+            np.testing.assert_allclose(got, want, rtol=1e-13,
+                                       err_msg=f"ord={ordv} axis={ax}")

@@ -152,3 +152,23 @@ def test_plotting_helpers():
         pm.plot_distributed_array(b)
     with pytest.raises(TypeError, match="Not a DistributedArray"):
         pm.plot_distributed_array(torch.zeros(3, 3))
+
+
+def test_norm_axis_world1():
+    """norm(ord, axis) == np.linalg.norm(x, ord, axis) — the reference's
+    own pin (ref tests/test_distributedarray.py:215-222)."""
+    import numpy as np
+    import torch
+    import pylops_mpi_amd as pm
+    rng = np.random.default_rng(0)
+    g = rng.standard_normal((7, 5, 4))
+    x = pm.DistributedArray.to_dist(torch.from_numpy(g))
+    for ordv in (1, 2, None, np.inf, -np.inf, 0, 3):
+        for ax in (0, 1, 2):
+            got = x.norm(ord=ordv, axis=ax).numpy()
+            want = np.linalg.norm(g, ord=ordv, axis=ax) if ordv != 3 \
+                else np.sum(np.abs(g ** 3), axis=ax) ** (1 / 3)
+            np.testing.assert_allclose(got, want, rtol=1e-13, err_msg=f"{ordv} {ax}")
+    import pytest
+    with pytest.raises(ValueError, match="out of range"):
+        x.norm(axis=3)
