@@ -102,9 +102,10 @@ int pam_cscale(void* stream, void* y, const void* x, double alpha_re,
                double alpha_im, int64_t n, int dtype);
 int pam_conj(void* stream, void* y, const void* x, int64_t n, int dtype);
 
-/* Thresholding for ISTA/FISTA (pylops _softthreshold/_hardthreshold
- * formulas, imported by ref optimization/cls_sparsity.py:10).
- * kind: 0 = soft, 1 = hard.  In-place safe (y may equal x). */
+/* Thresholding for ISTA/FISTA (pylops _softthreshold/_hardthreshold/
+ * _halfthreshold formulas, imported by ref optimization/cls_sparsity.py:10;
+ * half = the published Xu et al. 2012 L1/2 prox, see pam.hip).
+ * kind: 0 = soft, 1 = hard, 2 = half.  In-place safe (y may equal x). */
 int pam_thresh(void* stream, void* y, const void* x, int64_t n, int kind,
                double thresh, int dtype);
 

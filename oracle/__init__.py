@@ -51,6 +51,7 @@ from .fredholm import (SimFredholm1, SimMDC,  # noqa: F401
 from .nsconv import (serial_nsconv_mv,  # noqa: F401
                      serial_nsconv_rmv)
 from .proximal import (soft_threshold, hard_threshold,  # noqa: F401
+                       half_threshold,
                        SerBox, SerL0, SerL1, SerL2,
                        dense_cg, dense_cgls,
                        ser_proximal_gradient, ser_admml2)

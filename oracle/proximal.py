@@ -282,3 +282,20 @@ def ser_admml2(proxg, Op, b, A, x0, tau, niter=10, z0=None, gfirst=False,
             z = proxg.prox(Ax + u, tau)
         u = u + Ax - z
     return x, z
+
+
+def half_threshold(x: np.ndarray, thresh: float) -> np.ndarray:
+    """L1/2 ('half') threshold — the published Xu et al. (2012) closed
+    form pylops' _halfthreshold implements: the EXACT prox of
+    (thresh/2)*|v|^(1/2) (verified to ~1e-12 against a dense grid scan;
+    pylops is absent from /root/reference, so the formula is pinned by
+    the prox-optimality property test in tests/test_oracle_proximal.py).
+    Complex: magnitude rule."""
+    a = np.abs(x)
+    cut = (54.0 ** (1.0 / 3.0) / 4.0) * thresh ** (2.0 / 3.0)
+    with np.errstate(divide="ignore", invalid="ignore"):
+        phi = np.arccos(np.clip((thresh / 8.0) * (a / 3.0) ** (-1.5),
+                                -1.0, 1.0))
+        fac = (2.0 / 3.0) * (1.0 + np.cos(2.0 * np.pi / 3.0
+                                          - (2.0 / 3.0) * phi))
+    return np.where(a > cut, x * fac, 0.0)

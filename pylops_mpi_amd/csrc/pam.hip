@@ -327,8 +327,26 @@ EWC_ENTRY(pam_conj(void* stream, void* y, const void* a, int64_t n,
 //   soft real   : sign(x) * max(|x|-t, 0)
 //   soft complex: z * max(|z|-t, 0)/|z|
 //   hard        : x * (|x| >= sqrt(2 t))
-// kind: 0 soft, 1 hard.  In-place safe (y may alias x).
+//   half        : L1/2 prox (Xu et al. 2012, the published formula pylops'
+//                 _halfthreshold implements — the EXACT prox of
+//                 (t/2)|v|^(1/2), matching ISTA's thresh = eps*alpha*0.5):
+//                 for |x| > (54^(1/3)/4) t^(2/3),
+//                 y = (2/3) x (1 + cos(2pi/3 - (2/3) acos((t/8)(|x|/3)^-1.5)))
+//                 else 0; complex by the magnitude rule.  Pinned by the prox
+//                 optimality property test (tests/test_oracle_proximal.py) —
+//                 pylops itself is absent from /root/reference.
+// kind: 0 soft, 1 hard, 2 half.  In-place safe (y may alias x).
 // ---------------------------------------------------------------------------
+template <typename T>
+__device__ __forceinline__ T half_factor(T a, T thresh) {
+  // multiplicative factor of the L1/2 prox at magnitude a (>= 0)
+  const T cut = (T)(0.9085602964160698) * pow(thresh, (T)(2.0 / 3.0));
+  if (!(a > cut)) return (T)0;  // 54^(1/3)/4 = 0.90856...
+  const T phi = acos((thresh / (T)8) * pow(a / (T)3, (T)-1.5));
+  return ((T)2 / (T)3) *
+         ((T)1 + cos((T)(2.0 * M_PI / 3.0) - ((T)2 / (T)3) * phi));
+}
+
 template <typename T, bool CPLX, int KIND>
 __global__ void __launch_bounds__(BLK) thresh_kernel(T* __restrict__ y,
                                                      const T* __restrict__ x,
@@ -341,8 +359,10 @@ __global__ void __launch_bounds__(BLK) thresh_kernel(T* __restrict__ y,
       if constexpr (KIND == 0) {
         const T m = fabs(v) - thresh;
         y[i] = m > (T)0 ? copysign(m, v) : (T)0;
-      } else {
+      } else if constexpr (KIND == 1) {
         y[i] = fabs(v) >= sqrt((T)2 * thresh) ? v : (T)0;
+      } else {
+        y[i] = v * half_factor(fabs(v), thresh);
       }
     } else {
       const T zr = x[2 * i], zi = x[2 * i + 1];
@@ -351,8 +371,10 @@ __global__ void __launch_bounds__(BLK) thresh_kernel(T* __restrict__ y,
       if constexpr (KIND == 0) {
         const T m = a - thresh;
         s = (m > (T)0 && a > (T)0) ? m / a : (T)0;
-      } else {
+      } else if constexpr (KIND == 1) {
         s = a >= sqrt((T)2 * thresh) ? (T)1 : (T)0;
+      } else {
+        s = half_factor(a, thresh);
       }
       y[2 * i] = zr * s;
       y[2 * i + 1] = zi * s;
@@ -371,6 +393,9 @@ static int thresh_launch(void* stream, void* y, const void* x, int kind,
                        dim3(BLK), 0, s, (T*)y, (const T*)x, (T)t, n);
   else if (kind == 1)
     hipLaunchKernelGGL((thresh_kernel<T, CPLX, 1>), dim3(grid_1d(n)),
+                       dim3(BLK), 0, s, (T*)y, (const T*)x, (T)t, n);
+  else if (kind == 2)
+    hipLaunchKernelGGL((thresh_kernel<T, CPLX, 2>), dim3(grid_1d(n)),
                        dim3(BLK), 0, s, (T*)y, (const T*)x, (T)t, n);
   else
     return PAM_EOP;

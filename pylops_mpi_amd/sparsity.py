@@ -3,7 +3,7 @@
 Drop-in for /root/reference/pylops_mpi/optimization/cls_sparsity.py:50-718
 and sparsity.py wrappers: proximal-gradient iterations with the pylops
 soft/hard thresholding formulas (HIP pam_thresh kernel, in place) and the
-power-iteration step-size estimate.  'half' thresholding is deferred.
+power-iteration step-size estimate ('soft'/'hard'/'half' kinds).
 """
 from math import sqrt
 from typing import Optional, Tuple, Union
@@ -15,7 +15,7 @@ from .distributedarray import DistributedArray
 from .solvers import power_iteration
 from .stacked import StackedDistributedArray
 
-_THRESH_KINDS = {"soft": 0, "hard": 1}
+_THRESH_KINDS = {"soft": 0, "hard": 1, "half": 2}
 
 
 def _thresh_inplace(d: DistributedArray, kind: int, thresh: float):
@@ -57,8 +57,6 @@ class ISTA:
         if threshkind not in ("hard", "soft", "half"):
             raise ValueError(
                 f"threshkind must be hard, soft, half, got {threshkind}")
-        if threshkind == "half":
-            raise NotImplementedError("half thresholding deferred")
         self.threshkind = _THRESH_KINDS[threshkind]
         self.decay = np.ones(niter) if decay is None else decay
         if alpha is not None:
