@@ -338,13 +338,16 @@ EWC_ENTRY(pam_conj(void* stream, void* y, const void* a, int64_t n,
 // kind: 0 soft, 1 hard, 2 half.  In-place safe (y may alias x).
 // ---------------------------------------------------------------------------
 template <typename T>
-__device__ __forceinline__ T half_factor(T a, T thresh) {
-  // multiplicative factor of the L1/2 prox at magnitude a (>= 0)
-  const T cut = (T)(0.9085602964160698) * pow(thresh, (T)(2.0 / 3.0));
+__device__ __forceinline__ T half_factor(T a_, T thresh_) {
+  // multiplicative factor of the L1/2 prox at magnitude a (>= 0).
+  // Always evaluated in double so the cutoff decision (a discontinuity)
+  // and the acos/pow chain agree with the f64 oracle for every dtype.
+  const double a = (double)a_, thresh = (double)thresh_;
+  const double cut = 0.9085602964160698 * pow(thresh, 2.0 / 3.0);
   if (!(a > cut)) return (T)0;  // 54^(1/3)/4 = 0.90856...
-  const T phi = acos((thresh / (T)8) * pow(a / (T)3, (T)-1.5));
-  return ((T)2 / (T)3) *
-         ((T)1 + cos((T)(2.0 * M_PI / 3.0) - ((T)2 / (T)3) * phi));
+  const double phi = acos((thresh / 8.0) * pow(a / 3.0, -1.5));
+  return (T)((2.0 / 3.0) *
+             (1.0 + cos(2.0 * M_PI / 3.0 - (2.0 / 3.0) * phi)));
 }
 
 template <typename T, bool CPLX, int KIND>
