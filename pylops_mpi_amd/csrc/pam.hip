@@ -343,8 +343,8 @@ __device__ __forceinline__ T half_factor(T a_, T thresh_) {
   // Always evaluated in double so the cutoff decision (a discontinuity)
   // and the acos/pow chain agree with the f64 oracle for every dtype.
   const double a = (double)a_, thresh = (double)thresh_;
-  const double cut = 0.9085602964160698 * pow(thresh, 2.0 / 3.0);
-  if (!(a > cut)) return (T)0;  // 54^(1/3)/4 = 0.90856...
+  const double cut = 0.9449407874211548 * pow(thresh, 2.0 / 3.0);
+  if (!(a > cut)) return (T)0;  // 54^(1/3)/4 = 0.944940787421...
   const double phi = acos((thresh / 8.0) * pow(a / 3.0, -1.5));
   return (T)((2.0 / 3.0) *
              (1.0 + cos(2.0 * M_PI / 3.0 - (2.0 / 3.0) * phi)));
