@@ -800,11 +800,16 @@ class DistributedArray:
         group so xGMI links run concurrently."""
         if self._axis == axis or self._partition is not Partition.SCATTER:
             return self
-        out = DistributedArray(self._global_shape, self._base_comm,
-                               self._partition, axis, mask=self._mask,
-                               engine=self._engine, dtype=self.dtype)
+        if self.size == 1:
+            # same block, new distribution label: zero-copy re-wrap
+            return DistributedArray(
+                self._global_shape, self._base_comm, self._partition, axis,
+                local_array=self._local_array, mask=self._mask,
+                engine=self._engine, dtype=self.dtype)
         counts_from = [s[self._axis] for s in self._all_local_shapes]
-        counts_to = [s[axis] for s in out._all_local_shapes]
+        counts_to = [local_split(self._global_shape, self.size, r,
+                                 Partition.SCATTER, axis)[axis]
+                     for r in range(self.size)]
         offs = np.cumsum([0] + counts_to[:-1])
         sends, recvs = [], []
         for r in range(self.size):
@@ -821,8 +826,10 @@ class DistributedArray:
         self._base_comm.exchange(
             [(t, r) for r, t in enumerate(sends) if r != self.rank],
             [(t, r) for r, t in enumerate(recvs) if r != self.rank])
-        out[:] = torch.cat(recvs, dim=self._axis)
-        return out
+        return DistributedArray(
+            self._global_shape, self._base_comm, self._partition, axis,
+            local_array=torch.cat(recvs, dim=self._axis), mask=self._mask,
+            engine=self._engine, dtype=self.dtype)
 
     def __repr__(self):
         return (f"<DistributedArray with global shape={self.global_shape}, "

@@ -154,17 +154,27 @@ class MPIFFTND(_MPIBaseFFTND):
             else 1.0 / float(np.prod(self.nffts))
 
     # ------------------------------------------------------ reshaped I/O
-    def _reshape_in(self, x: DistributedArray, shape) -> DistributedArray:
+    def _reshape_in(self, x: DistributedArray, shape,
+                    copy: bool) -> DistributedArray:
         """The @reshaped input rebalance (ref utils/decorators.py:44-78):
-        flat 1-D -> ``shape`` with a balanced axis-0 split."""
+        flat 1-D -> ``shape`` with a balanced axis-0 split.  When the
+        split already matches and nothing downstream mutates in place
+        (``copy=False``), the result is a zero-copy view of ``x``."""
         if x.partition is not Partition.SCATTER:
             raise ValueError(f"x should have partition={Partition.SCATTER}, "
                              f"{x.partition} != {Partition.SCATTER}")
-        arr = DistributedArray(tuple(shape), x.base_comm, Partition.SCATTER,
-                               0, engine="hip", dtype=x.dtype)
-        counts = [int(np.prod(s)) for s in arr.local_shapes]
-        arr[:] = rebalance_1d(x, counts).reshape(arr.local_shape)
-        return arr
+        from .distributedarray import local_split
+        size = x.base_comm.size
+        lshapes = [local_split(tuple(shape), size, r, Partition.SCATTER, 0)
+                   for r in range(size)]
+        counts = [int(np.prod(s)) for s in lshapes]
+        t = rebalance_1d(x, counts).reshape(lshapes[x.rank])
+        if copy and t.data_ptr() == x.local_array.data_ptr():
+            t = t.clone()
+        return DistributedArray(tuple(shape), x.base_comm,
+                                Partition.SCATTER, 0, local_array=t,
+                                local_shapes=lshapes, engine="hip",
+                                dtype=x.dtype)
 
     @staticmethod
     def _flatten_out(y: DistributedArray) -> DistributedArray:
@@ -273,7 +283,9 @@ class MPIFFTND(_MPIBaseFFTND):
     # ----------------------------------------------------- matvec paths
     def _matvec(self, x: DistributedArray) -> DistributedArray:
         # ref FFTND.py:214-244
-        arr = self._reshape_in(x, self.dims)
+        arr = self._reshape_in(
+            x, self.dims,
+            copy=bool(self.ifftshift_before.any()) or not self.clinear)
         if self.ifftshift_before.any():
             arr = ifftshift_nd(
                 arr, axes=[int(a) for a in
@@ -294,7 +306,9 @@ class MPIFFTND(_MPIBaseFFTND):
 
     def _rmatvec(self, x: DistributedArray) -> DistributedArray:
         # ref FFTND.py:246-276
-        arr = self._reshape_in(x, self.dimsd)
+        arr = self._reshape_in(
+            x, self.dimsd,
+            copy=bool(self.fftshift_after.any()) or self.real)
         if self.fftshift_after.any():
             arr = ifftshift_nd(
                 arr, axes=[int(a) for a in self.axes[self.fftshift_after]])
