@@ -196,8 +196,10 @@ class MPIFFTND(_MPIBaseFFTND):
         last = int(self.axes[-1])
         pending = [int(a) for a in self.axes]
         t = arr.local_array
+        world1 = arr.base_comm.size == 1
         while pending:
-            cur = arr.axis
+            # at world 1 every axis is local: one fused rocFFT plan
+            cur = -1 if world1 else arr.axis
             if forward and self.real and last in pending:
                 # real transform must run FIRST (np.fft.rfftn order)
                 if cur == last:
