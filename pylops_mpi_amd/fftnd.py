@@ -246,6 +246,13 @@ class MPIFFTND(_MPIBaseFFTND):
         """Re-wrap a transformed local block (possibly new dtype/extent
         along a LOCAL axis) as a DistributedArray with the same
         distribution axis."""
+        if like.base_comm.size == 1:
+            # world 1: the local block IS the global array; the axis label
+            # may even be the axis a real transform resized
+            return DistributedArray(
+                tuple(t.shape), like.base_comm, Partition.SCATTER,
+                like.axis, local_array=t, engine="hip",
+                dtype=_t2np(t.dtype))
         axis = like.axis
         gshape = list(like.global_shape)
         lshapes = [list(s) for s in like.local_shapes]
