@@ -35,6 +35,12 @@ template <> struct GemmCfg<float> {
   static constexpr int TM = 32;   // MFMA tile edge (square)
   static constexpr int TK = 2;    // K per MFMA
   static constexpr int BM = 128, BN = 128, BK = 32;
+  static constexpr int MI = 2, NJ = 2;  // MFMA tiles per wave (rows, cols)
+  // NBUF=2 was A/B-measured for f32: 135.4 -> 123.0 TF at 8192^3 —
+  // doubling the 66 KB LDS footprint halves occupancy (4 -> 2 WG/CU)
+  // and costs more than the hidden staging latency.  f64's smaller
+  // panels (25 -> 50 KB) keep 3 WG/CU and gained +14% (57.2 -> 65.1).
+  static constexpr int NBUF = 1;        // LDS K-panel buffers (see kernel)
   using acc_t = f32x16;
   using vec_t = f32x4;            // 16-B staging vector
   static constexpr int VW = 4;
@@ -51,7 +57,11 @@ template <> struct GemmCfg<float> {
 template <> struct GemmCfg<double> {
   static constexpr int TM = 16;
   static constexpr int TK = 4;
-  static constexpr int BM = 64, BN = 64, BK = 16;
+  // 128x64 block tile (A/B-measured vs the earlier 64x64: the bigger M
+  // tile cuts B-panel HBM traffic 25% and lifted the 8192^3 rate)
+  static constexpr int BM = 128, BN = 64, BK = 16;
+  static constexpr int MI = 4, NJ = 2;
+  static constexpr int NBUF = 2;
   using acc_t = f64x4;
   using vec_t = f64x2;
   static constexpr int VW = 2;
@@ -78,8 +88,8 @@ __global__ void __launch_bounds__(GBLK) gemm_kernel(
   using acc_t = typename CFG::acc_t;
   using vec_t = typename CFG::vec_t;
 
-  __shared__ T As[BK][BM + 1];  // transposed, padded
-  __shared__ T Bs[BK][BN];      // linear
+  __shared__ T As[CFG::NBUF][BK][BM + 1];  // transposed, padded
+  __shared__ T Bs[CFG::NBUF][BK][BN];       // linear
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -95,85 +105,117 @@ __global__ void __launch_bounds__(GBLK) gemm_kernel(
   const int64_t brow = (int64_t)blockIdx.y * BM;
   const int64_t bcol = (int64_t)blockIdx.x * BN;
 
-  acc_t acc[2][2];
+  constexpr int MI = CFG::MI, NJ = CFG::NJ;
+  acc_t acc[MI][NJ];
 #pragma unroll
-  for (int i = 0; i < 2; ++i)
+  for (int i = 0; i < MI; ++i)
 #pragma unroll
-    for (int j = 0; j < 2; ++j) acc[i][j] = {};
+    for (int j = 0; j < NJ; ++j) acc[i][j] = {};
 
-  for (int64_t k0 = 0; k0 < K; k0 += BK) {
-    // ---- stage A[brow:brow+BM, k0:k0+BK] -> As[k][row] (transposed)
-    constexpr int AV = (BM * BK) / VW / GBLK;  // vector loads per thread
+  // ---- K-panel pipeline.  NBUF=2 (f64): next panel's global loads are
+  // issued BEFORE the MFMA loop on the current panel and parked in
+  // registers; the LDS store + single barrier happen after, so the HBM
+  // latency hides under compute (one barrier per panel).  NBUF=1 keeps
+  // the classic 2-barrier structure (f32: doubling LDS would halve
+  // occupancy; measured best single-buffered).
+  constexpr int NBUF = CFG::NBUF;
+  constexpr int AV = (BM * BK) / VW / GBLK;  // A vector loads per thread
+  constexpr int BV = (BK * BN) / VW / GBLK;  // B vector loads per thread
+  vec_t areg[AV], breg[BV];
+
+  auto load_panel = [&](int64_t k0) {
 #pragma unroll
     for (int e = 0; e < AV; ++e) {
       const int vi = tid + GBLK * e;
       const int row = vi / (BK / VW);
       const int kv = vi % (BK / VW);
-      T vals[VW];
       const int64_t gr = brow + row;
       if (!GUARD || (gr < M && k0 + (int64_t)kv * VW + VW <= K)) {
-        vec_t v = *reinterpret_cast<const vec_t*>(A + gr * lda + k0
+        areg[e] = *reinterpret_cast<const vec_t*>(A + gr * lda + k0
                                                   + (int64_t)kv * VW);
-#pragma unroll
-        for (int j = 0; j < VW; ++j) vals[j] = v[j];
       } else {
 #pragma unroll
         for (int j = 0; j < VW; ++j) {
           const int64_t gk = k0 + kv * VW + j;
-          vals[j] = (gr < M && gk < K) ? A[gr * lda + gk] : (T)0;
-        }
-      }
-#pragma unroll
-      for (int j = 0; j < VW; ++j) As[kv * VW + j][row] = vals[j];
-    }
-    // ---- stage B[k0:k0+BK, bcol:bcol+BN] -> Bs[k][col] (linear)
-    // (a 16-B global_load_lds variant was A/B-measured: f32 within noise
-    //  (+1%), f64 -9% — the synchronous 2-barrier structure gets no
-    //  benefit from DMA staging without a counted-vmcnt pipeline)
-    {
-      constexpr int BV = (BK * BN) / VW / GBLK;
-#pragma unroll
-      for (int e = 0; e < BV; ++e) {
-        const int vi = tid + GBLK * e;
-        const int kb = vi / (BN / VW);
-        const int nv = vi % (BN / VW);
-        const int64_t gk = k0 + kb;
-        const int64_t gn = bcol + (int64_t)nv * VW;
-        if (!GUARD || (gk < K && gn + VW <= N)) {
-          *reinterpret_cast<vec_t*>(&Bs[kb][nv * VW]) =
-              *reinterpret_cast<const vec_t*>(B + gk * ldb + gn);
-        } else {
-#pragma unroll
-          for (int j = 0; j < VW; ++j)
-            Bs[kb][nv * VW + j] =
-                (gk < K && gn + j < N) ? B[gk * ldb + gn + j] : (T)0;
+          areg[e][j] = (gr < M && gk < K) ? A[gr * lda + gk] : (T)0;
         }
       }
     }
-    __syncthreads();
-    // ---- MFMA inner loop
 #pragma unroll
-    for (int kk = 0; kk < BK / TK; ++kk) {
-      const int krow = kk * TK + lk;
-      T a0 = As[krow][wr * 2 * TM + li];
-      T a1 = As[krow][wr * 2 * TM + TM + li];
-      T b0 = Bs[krow][wc * 2 * TM + li];
-      T b1 = Bs[krow][wc * 2 * TM + TM + li];
-      acc[0][0] = CFG::mfma(a0, b0, acc[0][0]);
-      acc[0][1] = CFG::mfma(a0, b1, acc[0][1]);
-      acc[1][0] = CFG::mfma(a1, b0, acc[1][0]);
-      acc[1][1] = CFG::mfma(a1, b1, acc[1][1]);
+    for (int e = 0; e < BV; ++e) {
+      const int vi = tid + GBLK * e;
+      const int kb = vi / (BN / VW);
+      const int nv = vi % (BN / VW);
+      const int64_t gk = k0 + kb;
+      const int64_t gn = bcol + (int64_t)nv * VW;
+      if (!GUARD || (gk < K && gn + VW <= N)) {
+        breg[e] = *reinterpret_cast<const vec_t*>(B + gk * ldb + gn);
+      } else {
+#pragma unroll
+        for (int j = 0; j < VW; ++j)
+          breg[e][j] = (gk < K && gn + j < N) ? B[gk * ldb + gn + j] : (T)0;
+      }
     }
+  };
+
+  auto store_panel = [&](int buf) {
+#pragma unroll
+    for (int e = 0; e < AV; ++e) {
+      const int vi = tid + GBLK * e;
+      const int row = vi / (BK / VW);
+      const int kv = vi % (BK / VW);
+#pragma unroll
+      for (int j = 0; j < VW; ++j) As[buf][kv * VW + j][row] = areg[e][j];
+    }
+#pragma unroll
+    for (int e = 0; e < BV; ++e) {
+      const int vi = tid + GBLK * e;
+      const int kb = vi / (BN / VW);
+      const int nv = vi % (BN / VW);
+      *reinterpret_cast<vec_t*>(&Bs[buf][kb][nv * VW]) = breg[e];
+    }
+  };
+
+  const int64_t NP = GUARD ? (K + BK - 1) / BK : K / BK;
+  if (NP > 0) {
+    load_panel(0);
+    store_panel(0);
     __syncthreads();
+    for (int64_t p = 0; p < NP; ++p) {
+      if (p + 1 < NP) load_panel((p + 1) * BK);
+      const int cur = (int)(p % NBUF);
+      // ---- MFMA inner loop on the current panel
+#pragma unroll
+      for (int kk = 0; kk < BK / TK; ++kk) {
+        const int krow = kk * TK + lk;
+        T a[MI], b[NJ];
+#pragma unroll
+        for (int mi = 0; mi < MI; ++mi)
+          a[mi] = As[cur][krow][wr * (BM / 2) + mi * TM + li];
+#pragma unroll
+        for (int nj = 0; nj < NJ; ++nj)
+          b[nj] = Bs[cur][krow][wc * (BN / 2) + nj * TM + li];
+#pragma unroll
+        for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+          for (int nj = 0; nj < NJ; ++nj)
+            acc[mi][nj] = CFG::mfma(a[mi], b[nj], acc[mi][nj]);
+      }
+      if (p + 1 < NP) {
+        if (NBUF == 1) __syncthreads();  // readers done before overwrite
+        store_panel((int)((p + 1) % NBUF));
+        __syncthreads();
+      }
+    }
   }
 
   // ---- epilogue: C/D fragment layout -> global (guarded)
 #pragma unroll
-  for (int mi = 0; mi < 2; ++mi)
+  for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
-    for (int nj = 0; nj < 2; ++nj) {
-      const int64_t r0 = brow + wr * 2 * TM + mi * TM;
-      const int64_t c0 = bcol + wc * 2 * TM + nj * TM;
+    for (int nj = 0; nj < NJ; ++nj) {
+      const int64_t r0 = brow + wr * (BM / 2) + mi * TM;
+      const int64_t c0 = bcol + wc * (BN / 2) + nj * TM;
       const int64_t cc = c0 + li;
       if (GUARD && cc >= N) continue;
 #pragma unroll

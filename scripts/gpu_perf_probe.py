@@ -68,6 +68,7 @@ def main():
     bench_gemm(4096, 4096, 4096, torch.float32, "gemm_f32")
     bench_gemm(8192, 8192, 8192, torch.float32, "gemm_f32")
     bench_gemm(4096, 4096, 4096, torch.float64, "gemm_f64")
+    bench_gemm(8192, 8192, 8192, torch.float64, "gemm_f64")
     bench_gemv(4096, "gemv_f64")
     bench_gemv(8192, "gemv_f64")
 
