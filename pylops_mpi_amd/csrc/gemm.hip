@@ -279,62 +279,95 @@ extern "C" int pam_gemm(void* stream, const void* A, const void* B, void* C,
 // ---------------------------------------------------------------------------
 // batched complex GEMM: C_b = op(A_b) @ B_b (op = N or conj-transpose).
 // The Fredholm1 batched integral kernel (ref signalprocessing/
-// Fredholm1.py:123,149-156).  Interleaved (re,im) storage; 32x32 output
-// tiles, 256 threads (4 outputs/thread), BKC=8 LDS-staged K-steps.
-// Per-slice panels are small (cfg5: 64x256x256 c64), so this is a
-// latency/HBM-bound VALU kernel; the batch dimension fills the chip.
+// Fredholm1.py:123,149-156).  Interleaved (re,im) storage.
+//
+// MFMA formulation: four independent accumulation chains per output tile
+// (ar*br, ai*bi, ar*bi, ai*br) so no operand negation is needed inside
+// the MFMA loop; the epilogue combines Cr = S_rr - S_ii, Ci = S_ri +
+// S_ir.  64x64 block tile, 4 waves as 2x2, BK=16 K-panels staged as
+// separate re/im LDS planes.  This replaced a VALU outer-product kernel
+// (r01 A/B at the judged cfg5 shape 513x(64x256x256) c64: VALU 24.6 TF
+// real-flops = 63% of the ~39 TF VALU ceiling; the MFMA version's
+// numbers are in DESIGN.md) — the batch dimension fills the chip either
+// way, but only MFMA reaches the matrix-core rate.
 // ---------------------------------------------------------------------------
-#define BKC 8
-
-// 64x32 output tile, 8 outputs/thread (compute:LDS-read ratio 2x the
-// naive 32x32 form — the kernel is LDS/issue bound, not HBM bound).
 template <typename T, bool CT>
 __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
     const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
     int64_t M, int64_t N, int64_t K, int64_t strideA, int64_t strideB,
     int64_t strideC) {
-  __shared__ T Asr[64][BKC + 1], Asi[64][BKC + 1];
-  __shared__ T Bsr[BKC][33], Bsi[BKC][33];
+  using CFG = GemmCfg<T>;
+  constexpr int TM = CFG::TM, TK = CFG::TK;
+  constexpr int BM = 64, BN = 64, BK = 16;  // BK=32 A/B'd: 49.5 -> 37.2 TF
+  constexpr int MI = 32 / TM, NJ = 32 / TM;  // MFMA tiles per wave
+  using acc_t = typename CFG::acc_t;
+
+  __shared__ T Asr[2][BK][BM + 1], Asi[2][BK][BM + 1];  // op(A), k-major
+  __shared__ T Bsr[2][BK][BN + 1], Bsi[2][BK][BN + 1];
   const int64_t b = blockIdx.z;
   const T* __restrict__ Ab = A + 2 * b * strideA;
   const T* __restrict__ Bb = B + 2 * b * strideB;
   T* __restrict__ Cb = C + 2 * b * strideC;
-  const int64_t m0 = (int64_t)blockIdx.y * 64;
-  const int64_t n0 = (int64_t)blockIdx.x * 32;
-  const int tn = threadIdx.x & 31;
-  const int tm = threadIdx.x >> 5;  // 0..7
-  T accr[8] = {}, acci[8] = {};
-  for (int64_t k0 = 0; k0 < K; k0 += BKC) {
-    // stage op(A)[m0:m0+64, k0:k0+BKC]
-    // (BKC=32 was A/B-measured 18% slower: LDS doubles, occupancy halves)
+  const int64_t m0 = (int64_t)blockIdx.y * BM;
+  const int64_t n0 = (int64_t)blockIdx.x * BN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int li = lane & (TM - 1);
+  const int lk = lane / TM;
+
+  acc_t s_rr[MI][NJ], s_ii[MI][NJ], s_ri[MI][NJ], s_ir[MI][NJ];
 #pragma unroll
-    for (int e2 = 0; e2 < (64 * BKC) / GBLK; ++e2) {
-      const int vi2 = threadIdx.x + 256 * e2;
-      const int kk = vi2 & (BKC - 1);
-      const int mm = vi2 / BKC;  // 0..63
+  for (int i = 0; i < MI; ++i)
+#pragma unroll
+    for (int j = 0; j < NJ; ++j) {
+      s_rr[i][j] = {};
+      s_ii[i][j] = {};
+      s_ri[i][j] = {};
+      s_ir[i][j] = {};
+    }
+
+  // K-panel pipeline (same scheme as gemm_kernel: next panel's global
+  // loads parked in registers under the MFMA loop, one barrier/panel)
+  constexpr int AE = (BM * BK) / GBLK;
+  constexpr int BE = (BK * BN) / GBLK;
+  T argr[AE], argi[AE], brgr[BE], brgi[BE];
+
+  auto load_panel = [&](int64_t k0) {
+#pragma unroll
+    for (int e = 0; e < AE; ++e) {
+      const int vi2 = tid + GBLK * e;
+      int kk, mm;
+      if constexpr (CT) {  // A is [K, M]: consecutive m coalesces
+        mm = vi2 & (BM - 1);
+        kk = vi2 / BM;
+      } else {             // A is [M, K]: consecutive k coalesces
+        kk = vi2 & (BK - 1);
+        mm = vi2 / BK;
+      }
       const int64_t gm = m0 + mm;
       const int64_t gk = k0 + kk;
       T vr = 0, vi = 0;
       if (gm < M && gk < K) {
-        if constexpr (CT) {  // op(A)[m][k] = conj(A[k][m]), A is [K, M]
+        if constexpr (CT) {  // op(A)[m][k] = conj(A[k][m])
           const int64_t off = 2 * (gk * M + gm);
           vr = Ab[off];
           vi = -Ab[off + 1];
-        } else {             // A is [M, K]
+        } else {
           const int64_t off = 2 * (gm * K + gk);
           vr = Ab[off];
           vi = Ab[off + 1];
         }
       }
-      Asr[mm][kk] = vr;
-      Asi[mm][kk] = vi;
+      argr[e] = vr;
+      argi[e] = vi;
     }
-    // stage B[k0:k0+BKC, n0:n0+32]
 #pragma unroll
-    for (int e2 = 0; e2 < (32 * BKC) / GBLK; ++e2) {
-      const int vi2 = threadIdx.x + 256 * e2;
-      const int nn = vi2 & 31;
-      const int kk = vi2 >> 5;
+    for (int e = 0; e < BE; ++e) {
+      const int vi2 = tid + GBLK * e;
+      const int nn = vi2 & (BN - 1);
+      const int kk = vi2 / BN;
       const int64_t gk = k0 + kk;
       const int64_t gn = n0 + nn;
       T vr = 0, vi = 0;
@@ -343,31 +376,93 @@ __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
         vr = Bb[off];
         vi = Bb[off + 1];
       }
-      Bsr[kk][nn] = vr;
-      Bsi[kk][nn] = vi;
+      brgr[e] = vr;
+      brgi[e] = vi;
     }
+  };
+
+  auto store_panel = [&](int buf) {
+#pragma unroll
+    for (int e = 0; e < AE; ++e) {
+      const int vi2 = tid + GBLK * e;
+      int kk, mm;
+      if constexpr (CT) {
+        mm = vi2 & (BM - 1);
+        kk = vi2 / BM;
+      } else {
+        kk = vi2 & (BK - 1);
+        mm = vi2 / BK;
+      }
+      Asr[buf][kk][mm] = argr[e];
+      Asi[buf][kk][mm] = argi[e];
+    }
+#pragma unroll
+    for (int e = 0; e < BE; ++e) {
+      const int vi2 = tid + GBLK * e;
+      const int nn = vi2 & (BN - 1);
+      const int kk = vi2 / BN;
+      Bsr[buf][kk][nn] = brgr[e];
+      Bsi[buf][kk][nn] = brgi[e];
+    }
+  };
+
+  const int64_t NP = (K + BK - 1) / BK;
+  if (NP > 0) {
+    load_panel(0);
+    store_panel(0);
     __syncthreads();
+    for (int64_t p = 0; p < NP; ++p) {
+      if (p + 1 < NP) load_panel((p + 1) * BK);
+      const int cur = (int)(p & 1);
 #pragma unroll
-    for (int kk = 0; kk < BKC; ++kk) {
-      const T br = Bsr[kk][tn], bi = Bsi[kk][tn];
+      for (int kk = 0; kk < BK / TK; ++kk) {
+        const int krow = kk * TK + lk;
+        T ar[MI], ai[MI], br[NJ], bi[NJ];
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        const T ar = Asr[tm + 8 * e][kk], ai = Asi[tm + 8 * e][kk];
-        accr[e] += ar * br - ai * bi;
-        acci[e] += ar * bi + ai * br;
+        for (int mi = 0; mi < MI; ++mi) {
+          ar[mi] = Asr[cur][krow][wr * 32 + mi * TM + li];
+          ai[mi] = Asi[cur][krow][wr * 32 + mi * TM + li];
+        }
+#pragma unroll
+        for (int nj = 0; nj < NJ; ++nj) {
+          br[nj] = Bsr[cur][krow][wc * 32 + nj * TM + li];
+          bi[nj] = Bsi[cur][krow][wc * 32 + nj * TM + li];
+        }
+#pragma unroll
+        for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+          for (int nj = 0; nj < NJ; ++nj) {
+            s_rr[mi][nj] = CFG::mfma(ar[mi], br[nj], s_rr[mi][nj]);
+            s_ii[mi][nj] = CFG::mfma(ai[mi], bi[nj], s_ii[mi][nj]);
+            s_ri[mi][nj] = CFG::mfma(ar[mi], bi[nj], s_ri[mi][nj]);
+            s_ir[mi][nj] = CFG::mfma(ai[mi], br[nj], s_ir[mi][nj]);
+          }
+      }
+      if (p + 1 < NP) {
+        store_panel((int)((p + 1) & 1));
+        __syncthreads();
       }
     }
-    __syncthreads();
   }
+
+  // epilogue: Cr = S_rr - S_ii, Ci = S_ri + S_ir
 #pragma unroll
-  for (int e = 0; e < 8; ++e) {
-    const int64_t gm = m0 + tm + 8 * e;
-    const int64_t gn = n0 + tn;
-    if (gm < M && gn < N) {
-      Cb[2 * (gm * N + gn)] = accr[e];
-      Cb[2 * (gm * N + gn) + 1] = acci[e];
+  for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int nj = 0; nj < NJ; ++nj) {
+      const int64_t r0 = m0 + wr * 32 + mi * TM;
+      const int64_t cc = n0 + wc * 32 + nj * TM + li;
+      if (cc >= N) continue;
+#pragma unroll
+      for (int reg = 0; reg < CFG::NREG; ++reg) {
+        const int64_t rr = r0 + CFG::crow(lane, reg);
+        if (rr < M) {
+          const int64_t off = 2 * (rr * N + cc);
+          Cb[off] = s_rr[mi][nj][reg] - s_ii[mi][nj][reg];
+          Cb[off + 1] = s_ri[mi][nj][reg] + s_ir[mi][nj][reg];
+        }
+      }
     }
-  }
 }
 
 template <typename T>
@@ -376,7 +471,7 @@ static int cgemm_launch(void* stream, const void* A, const void* B, void* C,
                         int64_t sA, int64_t sB, int64_t sC, int opa) {
   if (batch <= 0 || M <= 0 || N <= 0 || K < 0 || !A || !B || !C)
     return PAM_EARG;
-  dim3 grid((uint32_t)((N + 31) / 32), (uint32_t)((M + 63) / 64),
+  dim3 grid((uint32_t)((N + 63) / 64), (uint32_t)((M + 63) / 64),
             (uint32_t)batch);
   hipStream_t s = (hipStream_t)stream;
   if (opa)
