@@ -226,3 +226,51 @@ def test_ista_half_thresholding():
     got = host(xs.asarray())
     assert np.allclose(got, xt, atol=0.2)
     assert cost[-1] <= cost[0]
+
+
+def test_cgls_hstack_broadcastdata():
+    """Mirror of ref tests/test_solver.py:203-244 (world-1): CGLS over an
+    MPIHStack — SCATTER model, BROADCAST data — vs the dense serial
+    recurrence."""
+    import oracle
+    rng = np.random.default_rng(42)
+    ny, nx = 36, 24
+    A = rng.standard_normal((ny, nx)) + np.eye(ny, nx) * 3
+    op = pm.MPIHStack([pm.DenseLocal(dev(A))])
+    x = pm.DistributedArray((nx,))
+    xg = rng.standard_normal(nx)
+    x[:] = dev(xg)
+    y = op.matvec(x)
+    assert y.partition is pm.Partition.BROADCAST
+    x0 = pm.DistributedArray((nx,))
+    x0[:] = 0.0
+    # tol=0 keeps both sides at exactly niter iterations (a tol crossing
+    # is a borderline fp event that would desynchronize the comparison)
+    xinv = pm.cgls(op, y, x0, niter=nx, tol=0.0)[0]
+    assert isinstance(xinv, pm.DistributedArray)
+    yg = A @ xg
+    xref = oracle.dense_cgls(A, yg, np.zeros(nx), niter=nx, tol=0.0)
+    # nx full CGLS iterations amplify backend fp-order differences;
+    # measured divergence ~4e-6 relative
+    assert_allclose(host(xinv.asarray()), xref, rtol=1e-5, atol=1e-7)
+
+
+def test_cgls_vstack_broadcastmodel():
+    """Mirror of ref tests/test_solver.py:251-300 (world-1): CGLS over an
+    MPIVStack with a BROADCAST model."""
+    import oracle
+    rng = np.random.default_rng(43)
+    nx = 20
+    A = rng.standard_normal((nx, nx))
+    S = A.T @ A + 1e-5 * np.eye(nx)  # SPD-ish, the reference's recipe
+    op = pm.MPIVStack([pm.DenseLocal(dev(S))])
+    x = pm.DistributedArray((nx,), partition=pm.Partition.BROADCAST)
+    xg = rng.standard_normal(nx)
+    x[:] = dev(xg)
+    y = op.matvec(x)
+    x0 = pm.DistributedArray((nx,), partition=pm.Partition.BROADCAST)
+    x0[:] = 0.0
+    xinv = pm.cgls(op, y, x0, niter=3 * nx, tol=0.0)[0]
+    yg = S @ xg
+    xref = oracle.dense_cgls(S, yg, np.zeros(nx), niter=3 * nx, tol=0.0)
+    assert_allclose(host(xinv.asarray()), xref, rtol=1e-6, atol=1e-8)
