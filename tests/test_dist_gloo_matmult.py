@@ -258,3 +258,49 @@ def body_halo_2d_tuple(c):
                                   "body_halo_2d_tuple"])
 def test_gloo_world4_halo(body):
     _spawn(body)
+
+
+def body_fftnd_world4(c):
+    """MPIFFTND across 4 ranks: the multi-transpose path (axes[-1] == 0
+    forces the input realignment, ref FFTND.py:190-196) and a 3-D real
+    case, vs the serial oracle."""
+    import numpy as np
+    import torch
+    import oracle
+    import pylops_mpi_amd as pm
+    rng = np.random.default_rng(91)
+    for par in (
+        dict(dims=(12, 9), axes=(1, 0), real=False, norm="none",
+             dtype=np.complex128, imag=1j),
+        dict(dims=(11, 8, 6), axes=(0, 1, 2), real=True, norm="1/n",
+             dtype=np.float64, imag=0),
+        dict(dims=(8, 11, 6), axes=(2, 1, 0), real=True, norm="none",
+             dtype=np.float64, imag=0),
+    ):
+        op = pm.MPIFFTND(dims=par["dims"], axes=par["axes"],
+                         norm=par["norm"], real=par["real"],
+                         dtype=par["dtype"], base_comm=c)
+        n = int(np.prod(par["dims"]))
+        xg = rng.standard_normal(n)
+        if par["imag"]:
+            xg = xg + 1j * rng.standard_normal(n)
+        xg = xg.astype(par["dtype"])
+        x = pm.DistributedArray.to_dist(torch.from_numpy(xg), c)
+        y = op.matvec(x)
+        y_ref = oracle.serial_fftnd_mv(xg, par["dims"], par["axes"],
+                                       norm=par["norm"], real=par["real"])
+        np.testing.assert_allclose(y.asarray().numpy(), y_ref,
+                                   rtol=1e-10, atol=1e-11)
+        yg = (rng.standard_normal(op.shape[0])
+              + 1j * rng.standard_normal(op.shape[0]))
+        yd = pm.DistributedArray.to_dist(torch.from_numpy(yg), c)
+        z = op.rmatvec(yd)
+        z_ref = oracle.serial_fftnd_rmv(yg, par["dims"], par["axes"],
+                                        norm=par["norm"], real=par["real"])
+        np.testing.assert_allclose(z.asarray().numpy(), z_ref,
+                                   rtol=1e-10, atol=1e-11)
+
+
+@pytest.mark.parametrize("body", ["body_fftnd_world4"])
+def test_gloo_world4_fftnd(body):
+    _spawn(body)

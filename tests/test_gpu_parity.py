@@ -456,3 +456,17 @@ def test_composite_operator_algebra_vs_dense():
         u = pm.DistributedArray.to_dist(dev(rng.standard_normal(n)))
         v = pm.DistributedArray.to_dist(dev(rng.standard_normal(n)))
         assert pm.dottest(o, u, v, rtol=1e-10), name
+
+
+def test_norm_axis_gpu_vs_numpy():
+    """norm(ord, axis) on device tensors == np.linalg.norm (the
+    reference's own pin, ref tests/test_distributedarray.py:215-222)."""
+    rng = np.random.default_rng(33)
+    g = rng.standard_normal((7, 5, 4))
+    x = pm.DistributedArray.to_dist(dev(g))
+    for ordv in (1, 2, None, np.inf, 0):
+        for ax in (0, 1, 2):
+            got = x.norm(ord=ordv, axis=ax).cpu().numpy()
+            want = np.linalg.norm(g, ord=ordv, axis=ax)
+            assert_allclose(got, want, rtol=1e-13,
+                            err_msg=f"ord={ordv} axis={ax}")
