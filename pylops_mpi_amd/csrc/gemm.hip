@@ -36,10 +36,11 @@ template <> struct GemmCfg<float> {
   static constexpr int TK = 2;    // K per MFMA
   static constexpr int BM = 128, BN = 128, BK = 32;
   static constexpr int MI = 2, NJ = 2;  // MFMA tiles per wave (rows, cols)
-  // NBUF=2 was A/B-measured for f32: 135.4 -> 123.0 TF at 8192^3 —
-  // doubling the 66 KB LDS footprint halves occupancy (4 -> 2 WG/CU)
-  // and costs more than the hidden staging latency.  f64's smaller
-  // panels (25 -> 50 KB) keep 3 WG/CU and gained +14% (57.2 -> 65.1).
+  // Pipeline A/Bs at 8192^3 (all NEGATIVE for f32, kept single-buffered):
+  //   BK=32 NBUF=2: 135.4 -> 123.0 TF (66 KB LDS halves occupancy)
+  //   BK=16 NBUF=2: 135.4 -> 131.7 TF (4 WG/CU kept, but the shallower
+  //   panels double the barrier count).  f64's 128x64/BK16 panels DID
+  //   gain from NBUF=2 (57.2 -> 65.1, see GemmCfg<double>).
   static constexpr int NBUF = 1;        // LDS K-panel buffers (see kernel)
   using acc_t = f32x16;
   using vec_t = f32x4;            // 16-B staging vector
