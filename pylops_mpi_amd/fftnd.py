@@ -197,6 +197,18 @@ class MPIFFTND(_MPIBaseFFTND):
         pending = [int(a) for a in self.axes]
         t = arr.local_array
         world1 = arr.base_comm.size == 1
+        if world1 and self.real:
+            # one fused rocFFT real plan (canonical rfftn/irfftn order)
+            dims_ = [int(a) for a in self.axes]
+            if forward:
+                t = torch.fft.rfftn(t, dim=dims_, norm="backward")
+            else:
+                t = torch.fft.irfftn(t, s=[self.nffts[-1]], dim=[last],
+                                     norm="backward") \
+                    if len(dims_) == 1 else torch.fft.irfftn(
+                        t, s=[t.shape[a] for a in dims_[:-1]]
+                        + [self.nffts[-1]], dim=dims_, norm="backward")
+            return self._wrap(arr, t)
         while pending:
             # at world 1 every axis is local: one fused rocFFT plan
             cur = -1 if world1 else arr.axis
