@@ -306,7 +306,8 @@ class MPIFFTND(_MPIBaseFFTND):
         # ref FFTND.py:214-244
         arr = self._reshape_in(
             x, self.dims,
-            copy=bool(self.ifftshift_before.any()) or not self.clinear)
+            copy=bool(self.ifftshift_before.any())
+            or (not self.clinear and x.local_array.is_complex()))
         if self.ifftshift_before.any():
             arr = ifftshift_nd(
                 arr, axes=[int(a) for a in
