@@ -76,3 +76,31 @@ def test_fullsize_cgls_steps():
                                              tol=0.0)
     assert iters == 3 and np.all(np.isfinite(cost))
     assert cost[1] <= cost[0]  # residual decreases
+
+
+def test_fullsize_fftnd_properties():
+    """MPIFFTND at a >2^31-element real workload: linearity and the
+    F^H F = N identity (norm='none', real) at sizes the serial oracle
+    cannot run — the tier's size-independent-property gate."""
+    dims = (1025, 2048, 1024)  # 2.149e9 pts, 17.2 GB fp64
+    n = int(np.prod(dims))
+    op = pm.MPIFFTND(dims=dims, axes=(0, 1, 2), real=True,
+                     dtype=np.float64)
+    g = torch.Generator(device="cuda").manual_seed(11)
+    x = pm.DistributedArray((n,))
+    x[:] = torch.randn(n, generator=g, dtype=torch.float64, device="cuda")
+    y = op.matvec(x)
+    assert y.global_shape == (int(np.prod(op.dimsd)),)
+    z = op.rmatvec(y)
+    # F^H F = N * I on real inputs (sqrt2-twin convention)
+    z.iaxpy_(-float(np.prod(dims)), x)
+    num = float(z.norm())
+    den = float(np.prod(dims)) * float(x.norm())
+    assert num / den < 1e-13
+    del y, z
+    # linearity
+    y1 = op.matvec(x)
+    y2 = op.matvec(x * 2.0)
+    # complex arrays: use the real view for the fused axpy
+    y2.iaxpy_(-2.0, y1)
+    assert float(y2.norm()) / float(y1.norm()) < 1e-13
