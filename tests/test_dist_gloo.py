@@ -171,7 +171,7 @@ def body_blockdiag(c):
     "body_sendrecv", "body_blockdiag", "body_fredholm", "body_vstack",
     "body_post_neighbors_overlap", "body_mask_subcomm", "body_nonstatconv",
     "body_proximal_call_reduction", "body_redistribute", "body_fftnd",
-    "body_norm_axis",
+    "body_norm_axis", "body_broadcast_setitem",
 ])
 def test_gloo_world2(body):
     _spawn(body)
@@ -430,3 +430,18 @@ def body_norm_axis(c):
             want = np.linalg.norm(g, ord=ordv, axis=ax)
             np.testing.assert_allclose(got, want, rtol=1e-13,
                                        err_msg=f"ord={ordv} axis={ax}")
+
+
+def body_broadcast_setitem(c):
+    """__setitem__ partition semantics across 2 ranks (ref
+    DistributedArray.py:217-252): BROADCAST re-broadcasts rank 0's
+    assignment; UNSAFE_BROADCAST keeps each rank's local value."""
+    import numpy as np
+    from pylops_mpi_amd import DistributedArray, Partition
+    b = DistributedArray((6,), c, Partition.BROADCAST)
+    b[:] = torch.full((6,), float(c.rank + 1), dtype=torch.float64)
+    np.testing.assert_array_equal(b.local_array.numpy(), np.ones(6))
+    u = DistributedArray((6,), c, Partition.UNSAFE_BROADCAST)
+    u[:] = torch.full((6,), float(c.rank + 1), dtype=torch.float64)
+    np.testing.assert_array_equal(u.local_array.numpy(),
+                                  np.full(6, c.rank + 1.0))
