@@ -185,22 +185,34 @@ __global__ void __launch_bounds__(GBLK) gemm_kernel(
     for (int64_t p = 0; p < NP; ++p) {
       if (p + 1 < NP) load_panel((p + 1) * BK);
       const int cur = (int)(p % NBUF);
-      // ---- MFMA inner loop on the current panel
+      // ---- MFMA inner loop on the current panel; LDS fragments are
+      // prefetched one k-step ahead so their latency hides under the
+      // MFMAs (the naive form serializes ds_read -> lgkmcnt(0) -> MFMA
+      // per step — seen in the ISA dump)
+      T a[2][MI], b[2][NJ];
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi)
+        a[0][mi] = As[cur][lk][wr * (BM / 2) + mi * TM + li];
+#pragma unroll
+      for (int nj = 0; nj < NJ; ++nj)
+        b[0][nj] = Bs[cur][lk][wc * (BN / 2) + nj * TM + li];
 #pragma unroll
       for (int kk = 0; kk < BK / TK; ++kk) {
-        const int krow = kk * TK + lk;
-        T a[MI], b[NJ];
+        const int cf = kk & 1, nf = (kk + 1) & 1;
+        if (kk + 1 < BK / TK) {
+          const int krow = (kk + 1) * TK + lk;
 #pragma unroll
-        for (int mi = 0; mi < MI; ++mi)
-          a[mi] = As[cur][krow][wr * (BM / 2) + mi * TM + li];
+          for (int mi = 0; mi < MI; ++mi)
+            a[nf][mi] = As[cur][krow][wr * (BM / 2) + mi * TM + li];
 #pragma unroll
-        for (int nj = 0; nj < NJ; ++nj)
-          b[nj] = Bs[cur][krow][wc * (BN / 2) + nj * TM + li];
+          for (int nj = 0; nj < NJ; ++nj)
+            b[nf][nj] = Bs[cur][krow][wc * (BN / 2) + nj * TM + li];
+        }
 #pragma unroll
         for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
           for (int nj = 0; nj < NJ; ++nj)
-            acc[mi][nj] = CFG::mfma(a[mi], b[nj], acc[mi][nj]);
+            acc[mi][nj] = CFG::mfma(a[cf][mi], b[cf][nj], acc[mi][nj]);
       }
       if (p + 1 < NP) {
         if (NBUF == 1) __syncthreads();  // readers done before overwrite
@@ -415,28 +427,43 @@ __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
     for (int64_t p = 0; p < NP; ++p) {
       if (p + 1 < NP) load_panel((p + 1) * BK);
       const int cur = (int)(p & 1);
+      // fragments prefetched one k-step ahead (same scheme as
+      // gemm_kernel)
+      T ar[2][MI], ai[2][MI], br[2][NJ], bi[2][NJ];
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi) {
+        ar[0][mi] = Asr[cur][lk][wr * 32 + mi * TM + li];
+        ai[0][mi] = Asi[cur][lk][wr * 32 + mi * TM + li];
+      }
+#pragma unroll
+      for (int nj = 0; nj < NJ; ++nj) {
+        br[0][nj] = Bsr[cur][lk][wc * 32 + nj * TM + li];
+        bi[0][nj] = Bsi[cur][lk][wc * 32 + nj * TM + li];
+      }
 #pragma unroll
       for (int kk = 0; kk < BK / TK; ++kk) {
-        const int krow = kk * TK + lk;
-        T ar[MI], ai[MI], br[NJ], bi[NJ];
+        const int cf = kk & 1, nf = (kk + 1) & 1;
+        if (kk + 1 < BK / TK) {
+          const int krow = (kk + 1) * TK + lk;
 #pragma unroll
-        for (int mi = 0; mi < MI; ++mi) {
-          ar[mi] = Asr[cur][krow][wr * 32 + mi * TM + li];
-          ai[mi] = Asi[cur][krow][wr * 32 + mi * TM + li];
-        }
+          for (int mi = 0; mi < MI; ++mi) {
+            ar[nf][mi] = Asr[cur][krow][wr * 32 + mi * TM + li];
+            ai[nf][mi] = Asi[cur][krow][wr * 32 + mi * TM + li];
+          }
 #pragma unroll
-        for (int nj = 0; nj < NJ; ++nj) {
-          br[nj] = Bsr[cur][krow][wc * 32 + nj * TM + li];
-          bi[nj] = Bsi[cur][krow][wc * 32 + nj * TM + li];
+          for (int nj = 0; nj < NJ; ++nj) {
+            br[nf][nj] = Bsr[cur][krow][wc * 32 + nj * TM + li];
+            bi[nf][nj] = Bsi[cur][krow][wc * 32 + nj * TM + li];
+          }
         }
 #pragma unroll
         for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
           for (int nj = 0; nj < NJ; ++nj) {
-            s_rr[mi][nj] = CFG::mfma(ar[mi], br[nj], s_rr[mi][nj]);
-            s_ii[mi][nj] = CFG::mfma(ai[mi], bi[nj], s_ii[mi][nj]);
-            s_ri[mi][nj] = CFG::mfma(ar[mi], bi[nj], s_ri[mi][nj]);
-            s_ir[mi][nj] = CFG::mfma(ai[mi], br[nj], s_ir[mi][nj]);
+            s_rr[mi][nj] = CFG::mfma(ar[cf][mi], br[cf][nj], s_rr[mi][nj]);
+            s_ii[mi][nj] = CFG::mfma(ai[cf][mi], bi[cf][nj], s_ii[mi][nj]);
+            s_ri[mi][nj] = CFG::mfma(ar[cf][mi], bi[cf][nj], s_ri[mi][nj]);
+            s_ir[mi][nj] = CFG::mfma(ai[cf][mi], br[cf][nj], s_ir[mi][nj]);
           }
       }
       if (p + 1 < NP) {
