@@ -60,10 +60,47 @@ def read_traffic_calibration(dims, n_gpus):
     return None
 
 
+def _fd_pair_threaded(x3, y3, z3, nthreads):
+    """One centered3 matvec+rmatvec pair, row-sharded across host threads
+    (NumPy releases the GIL on slice arithmetic).  Same slice formulas as
+    oracle/serial.py (matvec y[i]=0.5(x[i+1]-x[i-1]), adjoint
+    y[j]=0.5 x[j-1] - 0.5 x[j+1] with the edge ranges of the dense
+    transpose)."""
+    import concurrent.futures as cf
+    n0 = x3.shape[0]
+    bounds = [(n0 * t // nthreads, n0 * (t + 1) // nthreads)
+              for t in range(nthreads)]
+
+    def mv(a, b):
+        lo, hi = max(a, 1), min(b, n0 - 1)
+        if hi > lo:
+            y3[lo:hi] = 0.5 * (x3[lo + 1: hi + 1] - x3[lo - 1: hi - 1])
+        if a == 0:
+            y3[0] = 0.0
+        if b == n0:
+            y3[n0 - 1] = 0.0
+
+    def rmv(a, b):
+        for j in range(a, b):
+            acc = None
+            if 2 <= j <= n0 - 1:
+                acc = 0.5 * y3[j - 1]
+            if j <= n0 - 3:
+                acc = (-0.5 * y3[j + 1]) if acc is None \
+                    else acc - 0.5 * y3[j + 1]
+            z3[j] = 0.0 if acc is None else acc
+
+    with cf.ThreadPoolExecutor(nthreads) as ex:
+        list(ex.map(lambda ab: mv(*ab), bounds))
+        list(ex.map(lambda ab: rmv(*ab), bounds))
+
+
 def cpu_baseline(dims):
     """The oracle (NumPy restatement of the reference path) timed on the
     host cores — kind 'port' (the reference's own mpi4py+NumPy path cannot
-    run here: no MPI/pylops in the image)."""
+    run here: no MPI/pylops in the image).  Reports the
+    all-core row-sharded figure as `value` (SURVEY §8d asks for both);
+    the 1-thread figure is in `sample`."""
     import oracle
     n = int(np.prod(dims))
     rng = np.random.default_rng(42)
@@ -77,19 +114,31 @@ def cpu_baseline(dims):
     for _ in range(pairs):
         y = op.matvec(x)
         _ = op.rmatvec(y)
-    dt = time.perf_counter() - t0
+    dt1 = time.perf_counter() - t0
     try:
         cores = len(os.sched_getaffinity(0))
     except AttributeError:
         cores = os.cpu_count()
+    # all-core: row-sharded threads over the same slice arithmetic
+    nthreads = min(int(cores), 64)
+    x3 = xg.reshape(dims[0], -1)
+    y3 = np.empty_like(x3)
+    z3 = np.empty_like(x3)
+    _fd_pair_threaded(x3, y3, z3, nthreads)  # warmup
+    t0 = time.perf_counter()
+    for _ in range(pairs):
+        _fd_pair_threaded(x3, y3, z3, nthreads)
+    dtT = time.perf_counter() - t0
     return {
-        "value": pairs / dt,
+        "value": pairs / dtT,
         "unit": "pairs/s",
-        "cores": 1,  # NumPy slice arithmetic is single-threaded
+        "cores": nthreads,
         "kind": "port",
         "sample": (f"{pairs} matvec+rmatvec pairs on the full "
-                   f"{'x'.join(map(str, dims))} fp64 workload, NumPy oracle, "
-                   f"1 thread (host has {cores} cores)"),
+                   f"{'x'.join(map(str, dims))} fp64 workload, NumPy "
+                   f"oracle row-sharded over {nthreads} threads "
+                   f"(host has {cores} cores); 1-thread figure: "
+                   f"{pairs / dt1:.3f} pairs/s"),
     }
 
 
