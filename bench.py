@@ -145,7 +145,7 @@ def cpu_baseline(dims):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--steps", type=int, default=50)  # >=50
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
