@@ -304,19 +304,22 @@ extern "C" int pam_gemm(void* stream, const void* A, const void* B, void* C,
 // numbers are in DESIGN.md) — the batch dimension fills the chip either
 // way, but only MFMA reaches the matrix-core rate.
 // ---------------------------------------------------------------------------
-template <typename T, bool CT>
+template <typename T, bool CT, int BK = 16>
 __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
     const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
     int64_t M, int64_t N, int64_t K, int64_t strideA, int64_t strideB,
     int64_t strideC) {
   using CFG = GemmCfg<T>;
   constexpr int TM = CFG::TM, TK = CFG::TK;
-  constexpr int BM = 64, BN = 64, BK = 16;  // BK=32 A/B'd: 49.5 -> 37.2 TF
+  constexpr int BM = 64, BN = 64;  // BK=32 A/B'd: 49.5 -> 37.2 TF; BK is
+  // a template param: 16 (pipelined) in general, 64 single-panel when
+  // K <= 64 (the Fredholm rmatvec's tall-skinny GT panels)
+  constexpr int NBUFC = (BK >= 64) ? 1 : 2;
   constexpr int MI = 32 / TM, NJ = 32 / TM;  // MFMA tiles per wave
   using acc_t = typename CFG::acc_t;
 
-  __shared__ T Asr[2][BK][BM + 1], Asi[2][BK][BM + 1];  // op(A), k-major
-  __shared__ T Bsr[2][BK][BN + 1], Bsi[2][BK][BN + 1];
+  __shared__ T Asr[NBUFC][BK][BM + 1], Asi[NBUFC][BK][BM + 1];  // k-major
+  __shared__ T Bsr[NBUFC][BK][BN + 1], Bsi[NBUFC][BK][BN + 1];
   const int64_t b = blockIdx.z;
   const T* __restrict__ Ab = A + 2 * b * strideA;
   const T* __restrict__ Bb = B + 2 * b * strideB;
@@ -425,8 +428,8 @@ __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
     store_panel(0);
     __syncthreads();
     for (int64_t p = 0; p < NP; ++p) {
-      if (p + 1 < NP) load_panel((p + 1) * BK);
-      const int cur = (int)(p & 1);
+      if (NBUFC > 1 && p + 1 < NP) load_panel((p + 1) * BK);
+      const int cur = (int)(p % NBUFC);
       // fragments prefetched one k-step ahead (same scheme as
       // gemm_kernel)
       T ar[2][MI], ai[2][MI], br[2][NJ], bi[2][NJ];
@@ -467,7 +470,11 @@ __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
           }
       }
       if (p + 1 < NP) {
-        store_panel((int)((p + 1) & 1));
+        if (NBUFC == 1) {
+          __syncthreads();
+          load_panel((p + 1) * BK);
+        }
+        store_panel((int)((p + 1) % NBUFC));
         __syncthreads();
       }
     }
@@ -502,6 +509,10 @@ static int cgemm_launch(void* stream, const void* A, const void* B, void* C,
   dim3 grid((uint32_t)((N + 63) / 64), (uint32_t)((M + 63) / 64),
             (uint32_t)batch);
   hipStream_t s = (hipStream_t)stream;
+  // A BK=64 single-panel variant for K <= 64 (one barrier, no pipeline)
+  // was A/B'd NEGATIVE at the cfg5 rmatvec shape (0.432 -> 0.518 ms):
+  // the 133 KB LDS footprint drops occupancy to 2 WG/CU, which costs
+  // more than the three extra barriers it saves.  BK=16 for all K.
   if (opa)
     hipLaunchKernelGGL((cgemm_batched_kernel<T, true>), grid, dim3(GBLK), 0,
                        s, (const T*)A, (const T*)B, (T*)C, M, N, K, sA, sB,
