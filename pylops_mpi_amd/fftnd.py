@@ -313,7 +313,10 @@ class MPIFFTND(_MPIBaseFFTND):
                 arr, axes=[int(a) for a in
                            self.axes[self.ifftshift_before]])
         if not self.clinear and arr.local_array.is_complex():
-            arr[:] = arr.local_array.real
+            # the reference casts the complex carrier to REAL storage
+            # before the real transform (FFTND.py:222-227 via the
+            # real-dtype PFFT input array)
+            arr = self._wrap(arr, arr.local_array.real.contiguous())
         y = self._dist_fft(arr, forward=True)
         if self.real:
             self._scale_real_fft(y, inverse=False)

@@ -150,3 +150,19 @@ def test_fftshift_helpers_world1():
     y2 = pm.ifftshift_nd(x2, axes=(0, 1))
     assert_allclose(y2.asarray().numpy().reshape(9, 6),
                     np.fft.ifftshift(g, axes=(0, 1)))
+
+
+def test_real_fft_complex_carrier():
+    """real=True with a COMPLEX-storage input (a solver carrying a real
+    model in complex storage): the reference casts to real before the
+    transform (ref FFTND.py:222-227); regression for the rfftn dtype
+    crash."""
+    rng = np.random.default_rng(9)
+    for dims, axes in (((8, 6), (0, 1)), ((5, 6, 4), (2, 0, 1))):
+        op = pm.MPIFFTND(dims=dims, axes=axes, real=True, dtype=np.float64)
+        n = int(np.prod(dims))
+        x = rng.standard_normal(n) + 1j * rng.standard_normal(n)
+        y = op.matvec(pm.DistributedArray.to_dist(torch.from_numpy(x)))
+        want = oracle.serial_fftnd_mv(x, dims, axes, real=True,
+                                      clinear=False)
+        assert_allclose(y.asarray().numpy(), want, rtol=1e-10, atol=1e-11)
