@@ -389,6 +389,9 @@ def body_fftnd(c):
              dtype=np.complex128, imag=1j),
         dict(dims=(5, 6, 4), axes=(2, 0, 1), real=True, norm="1/n",
              dtype=np.float64, imag=0),
+        # 0 not in axes: no pencil transpose at all
+        dict(dims=(6, 5, 4), axes=(1, 2), real=True, norm="none",
+             dtype=np.float64, imag=0),
     ):
         op = pm.MPIFFTND(dims=par["dims"], axes=par["axes"],
                          norm=par["norm"], real=par["real"],
