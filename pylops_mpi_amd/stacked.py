@@ -192,13 +192,45 @@ class MPIStackedLinearOperator:
 
     H = property(adjoint)
 
+    def transpose(self):
+        # ref StackedLinearOperator.py:406-426
+        return _StackedTransposed(self)
+
+    T = property(transpose)
+
+    def conj(self):
+        # ref :543-568
+        return _StackedConj(self)
+
     def __mul__(self, x):
+        # ref :223-262 dot dispatch
+        if isinstance(x, MPIStackedLinearOperator):
+            return _StackedProduct(self, x)
         if np.isscalar(x):
             return _StackedScaled(self, x)
         return self.matvec(x)
 
+    def __rmul__(self, x):
+        if np.isscalar(x):
+            return _StackedScaled(self, x)
+        return NotImplemented
+
     def __matmul__(self, x):
         return self.__mul__(x)
+
+    def __pow__(self, p):
+        # ref :515-540
+        return _StackedPower(self, p)
+
+    def __add__(self, x):
+        # ref :486-512
+        return _StackedSum(self, x)
+
+    def __neg__(self):
+        return _StackedScaled(self, -1)
+
+    def __sub__(self, x):
+        return self.__add__(-x)
 
 
 class _StackedAdjoint(MPIStackedLinearOperator):
@@ -216,6 +248,9 @@ class _StackedAdjoint(MPIStackedLinearOperator):
 
 class _StackedScaled(MPIStackedLinearOperator):
     def __init__(self, A, alpha):
+        # ref StackedLinearOperator.py:456-484
+        if not np.isscalar(alpha):
+            raise ValueError('scalar expected as alpha')
         self.A = A
         self.alpha = alpha
         super().__init__(shape=A.shape, dtype=A.dtype, base_comm=A.base_comm)
@@ -225,3 +260,102 @@ class _StackedScaled(MPIStackedLinearOperator):
 
     def _rmatvec(self, x):
         return self.A.rmatvec(x) * np.conj(self.alpha)
+
+
+class _StackedTransposed(MPIStackedLinearOperator):
+    """ref StackedLinearOperator.py:406-426 (Aᵀ = conj ∘ Aᴴ ∘ conj)."""
+
+    def __init__(self, A):
+        self.A = A
+        super().__init__(shape=(A.shape[1], A.shape[0]), dtype=A.dtype,
+                         base_comm=A.base_comm)
+
+    def _matvec(self, x):
+        return self.A.rmatvec(x.conj()).conj()
+
+    def _rmatvec(self, x):
+        return self.A.matvec(x.conj()).conj()
+
+
+class _StackedConj(MPIStackedLinearOperator):
+    """ref StackedLinearOperator.py:543-568."""
+
+    def __init__(self, A):
+        if not isinstance(A, MPIStackedLinearOperator):
+            raise TypeError('A must be a MPIStackedLinearOperator')
+        self.A = A
+        super().__init__(shape=A.shape, dtype=A.dtype, base_comm=A.base_comm)
+
+    def _matvec(self, x):
+        return self.A.matvec(x.conj()).conj()
+
+    def _rmatvec(self, x):
+        return self.A.rmatvec(x.conj()).conj()
+
+
+class _StackedProduct(MPIStackedLinearOperator):
+    """ref StackedLinearOperator.py:428-454."""
+
+    def __init__(self, A, B):
+        if not isinstance(A, MPIStackedLinearOperator) \
+                or not isinstance(B, MPIStackedLinearOperator):
+            raise ValueError(
+                'both operands have to be a MPIStackedLinearOperator')
+        if A.shape[1] != B.shape[0]:
+            raise ValueError('cannot multiply %r and %r: shape mismatch'
+                             % (A, B))
+        self.args = (A, B)
+        super().__init__(shape=(A.shape[0], B.shape[1]), dtype=A.dtype,
+                         base_comm=A.base_comm)
+
+    def _matvec(self, x):
+        return self.args[0].matvec(self.args[1].matvec(x))
+
+    def _rmatvec(self, x):
+        return self.args[1].rmatvec(self.args[0].rmatvec(x))
+
+
+class _StackedSum(MPIStackedLinearOperator):
+    """ref StackedLinearOperator.py:486-512."""
+
+    def __init__(self, A, B):
+        if not isinstance(A, MPIStackedLinearOperator) \
+                or not isinstance(B, MPIStackedLinearOperator):
+            raise ValueError(
+                'both operands have to be a MPIStackedLinearOperator')
+        if A.shape != B.shape:
+            raise ValueError("cannot add %r and %r: shape mismatch"
+                             % (A, B))
+        self.args = (A, B)
+        super().__init__(shape=A.shape, dtype=A.dtype, base_comm=A.base_comm)
+
+    def _matvec(self, x):
+        return self.args[0].matvec(x) + self.args[1].matvec(x)
+
+    def _rmatvec(self, x):
+        return self.args[0].rmatvec(x) + self.args[1].rmatvec(x)
+
+
+class _StackedPower(MPIStackedLinearOperator):
+    """ref StackedLinearOperator.py:515-540."""
+
+    def __init__(self, A, p):
+        if A.shape[0] != A.shape[1]:
+            raise ValueError("square MPIStackedLinearOperator expected, "
+                             "got %r" % A)
+        if not isinstance(p, (int, np.integer)) or p < 0:
+            raise ValueError("non-negative integer expected as p")
+        self.args = (A, int(p))
+        super().__init__(shape=A.shape, dtype=A.dtype, base_comm=A.base_comm)
+
+    def _power(self, fun, x):
+        res = x.copy()
+        for _ in range(self.args[1]):
+            res = fun(res)
+        return res
+
+    def _matvec(self, x):
+        return self._power(self.args[0].matvec, x)
+
+    def _rmatvec(self, x):
+        return self._power(self.args[0].rmatvec, x)

@@ -274,3 +274,49 @@ def test_cgls_vstack_broadcastmodel():
     yg = S @ xg
     xref = oracle.dense_cgls(S, yg, np.zeros(nx), niter=3 * nx, tol=0.0)
     assert_allclose(host(xinv.asarray()), xref, rtol=1e-6, atol=1e-8)
+
+
+def test_stacked_operator_algebra_vs_dense():
+    """Mirror of ref tests/test_stackedlinearop.py:40-331 (transpose/
+    scaled/conj/power/sum/product on MPIStackedLinearOperator) against
+    dense block-diagonal matrices (world 1)."""
+    rng = np.random.default_rng(21)
+    n1, n2 = 12, 9
+    A1 = rng.standard_normal((n1, n1))
+    A2 = rng.standard_normal((n2, n2))
+    S = pm.MPIStackedBlockDiag([pm.MPIBlockDiag([pm.DenseLocal(dev(A1))]),
+                                pm.MPIBlockDiag([pm.DenseLocal(dev(A2))])])
+    D = np.zeros((n1 + n2, n1 + n2))
+    D[:n1, :n1], D[n1:, n1:] = A1, A2
+
+    def mk(v):
+        return pm.StackedDistributedArray(
+            [pm.DistributedArray.to_dist(dev(v[:n1])),
+             pm.DistributedArray.to_dist(dev(v[n1:]))])
+
+    def get(sd):
+        return sd.asarray().cpu().numpy()
+
+    x = rng.standard_normal(n1 + n2)
+    cases = [
+        ("scaled", 2.5 * S, 2.5 * D),
+        ("neg", -S, -D),
+        ("adjoint", S.H, D.T),
+        ("transpose", S.T, D.T),
+        ("conj", S.conj(), D),
+        ("power", S ** 2, D @ D),
+        ("sum", S + S, D + D),
+        ("product", S * S, D @ D),
+    ]
+    for name, o, Dm in cases:
+        assert_allclose(get(o.matvec(mk(x))), Dm @ x, rtol=1e-11,
+                        atol=1e-11, err_msg=f"{name} fwd")
+        assert_allclose(get(o.rmatvec(mk(x))), Dm.T @ x, rtol=1e-11,
+                        atol=1e-11, err_msg=f"{name} adj")
+    # validation errors, ref :486-512,515-540
+    import pytest as _pt
+    with _pt.raises(ValueError, match="shape mismatch"):
+        _ = S + pm.MPIStackedBlockDiag(
+            [pm.MPIBlockDiag([pm.DenseLocal(dev(A1))])])
+    with _pt.raises(ValueError, match="non-negative integer"):
+        _ = S ** -1
