@@ -311,11 +311,14 @@ __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
     int64_t strideC) {
   using CFG = GemmCfg<T>;
   constexpr int TM = CFG::TM, TK = CFG::TK;
-  constexpr int BM = 64, BN = 64;  // BK=32 A/B'd: 49.5 -> 37.2 TF; BK is
-  // a template param: 16 (pipelined) in general, 64 single-panel when
-  // K <= 64 (the Fredholm rmatvec's tall-skinny GT panels)
+  // 64x64 block tile.  A/B'd alternatives, all NEGATIVE at cfg5:
+  //   BK=32 (49.5 -> 37.2 TF), BK=64 single-panel (0.432 -> 0.518 ms
+  //   rmatvec), f32 BN=128 wide tile (69.2 -> 66.2 TF: halving the WG
+  //   count costs more latency hiding than the doubled fragment reuse
+  //   gains).  BK stays a template param: 16 everywhere.
+  constexpr int BM = 64, BN = 64;
   constexpr int NBUFC = (BK >= 64) ? 1 : 2;
-  constexpr int MI = 32 / TM, NJ = 32 / TM;  // MFMA tiles per wave
+  constexpr int MI = (BM / 2) / TM, NJ = (BN / 2) / TM;
   using acc_t = typename CFG::acc_t;
 
   __shared__ T Asr[NBUFC][BK][BM + 1], Asi[NBUFC][BK][BM + 1];  // k-major
@@ -435,13 +438,13 @@ __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
       T ar[2][MI], ai[2][MI], br[2][NJ], bi[2][NJ];
 #pragma unroll
       for (int mi = 0; mi < MI; ++mi) {
-        ar[0][mi] = Asr[cur][lk][wr * 32 + mi * TM + li];
-        ai[0][mi] = Asi[cur][lk][wr * 32 + mi * TM + li];
+        ar[0][mi] = Asr[cur][lk][wr * (BM / 2) + mi * TM + li];
+        ai[0][mi] = Asi[cur][lk][wr * (BM / 2) + mi * TM + li];
       }
 #pragma unroll
       for (int nj = 0; nj < NJ; ++nj) {
-        br[0][nj] = Bsr[cur][lk][wc * 32 + nj * TM + li];
-        bi[0][nj] = Bsi[cur][lk][wc * 32 + nj * TM + li];
+        br[0][nj] = Bsr[cur][lk][wc * (BN / 2) + nj * TM + li];
+        bi[0][nj] = Bsi[cur][lk][wc * (BN / 2) + nj * TM + li];
       }
 #pragma unroll
       for (int kk = 0; kk < BK / TK; ++kk) {
@@ -450,13 +453,13 @@ __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
           const int krow = (kk + 1) * TK + lk;
 #pragma unroll
           for (int mi = 0; mi < MI; ++mi) {
-            ar[nf][mi] = Asr[cur][krow][wr * 32 + mi * TM + li];
-            ai[nf][mi] = Asi[cur][krow][wr * 32 + mi * TM + li];
+            ar[nf][mi] = Asr[cur][krow][wr * (BM / 2) + mi * TM + li];
+            ai[nf][mi] = Asi[cur][krow][wr * (BM / 2) + mi * TM + li];
           }
 #pragma unroll
           for (int nj = 0; nj < NJ; ++nj) {
-            br[nf][nj] = Bsr[cur][krow][wc * 32 + nj * TM + li];
-            bi[nf][nj] = Bsi[cur][krow][wc * 32 + nj * TM + li];
+            br[nf][nj] = Bsr[cur][krow][wc * (BN / 2) + nj * TM + li];
+            bi[nf][nj] = Bsi[cur][krow][wc * (BN / 2) + nj * TM + li];
           }
         }
 #pragma unroll
@@ -485,8 +488,8 @@ __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
   for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
     for (int nj = 0; nj < NJ; ++nj) {
-      const int64_t r0 = m0 + wr * 32 + mi * TM;
-      const int64_t cc = n0 + wc * 32 + nj * TM + li;
+      const int64_t r0 = m0 + wr * (BM / 2) + mi * TM;
+      const int64_t cc = n0 + wc * (BN / 2) + nj * TM + li;
       if (cc >= N) continue;
 #pragma unroll
       for (int reg = 0; reg < CFG::NREG; ++reg) {
