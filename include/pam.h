@@ -13,7 +13,8 @@
  *    stream; the library never synchronizes).
  *  - all functions return 0 on success, a hipError_t (>0) on launch
  *    failure, or a PAM_E* code (<0) on argument errors.
- *  - `dtype`: 0 = float64, 1 = float32.
+ *  - `dtype`: 0 = float64, 1 = float32, 2 = complex128, 3 = complex64
+ *    (complex: interleaved re,im pairs of the base type).
  *  - scalar outputs (`out`) are single-element float64 device buffers.
  */
 #ifndef PAM_H
