@@ -911,9 +911,22 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
     if ((ov == 1 || ov == 2 || ov == 4) && m % ov == 0) V = ov;
   }
   const int64_t mv = m / V;
-  int gy = (int)(nrows < 512 ? nrows : 512);
+  // Grid shape (PAM_FD_GY / PAM_FD_CAP override for A/Bs): one block-row
+  // per local row (gy = nrows) with ~16k blocks total measured best —
+  // bench A/B on one box, 3 reps: gy=512/cap=4096 286.7, gy=1024 293.3,
+  // gy=nrows/cap=16384 295.8 pairs/s (+3.2%); cap 32768/65536 regress.
+  static int gycap = [] {
+    const char* e = getenv("PAM_FD_GY");
+    return e ? atoi(e) : 65535;
+  }();
+  static int totcap = [] {
+    const char* e = getenv("PAM_FD_CAP");
+    return e ? atoi(e) : 16384;
+  }();
+  int gy = (int)(nrows < gycap ? nrows : gycap);
+  if (gy < 1) gy = 1;
   int64_t gx64 = (mv + BLK - 1) / BLK;
-  int64_t cap = 4096 / gy;
+  int64_t cap = totcap / gy;
   if (cap < 1) cap = 1;
   if (gx64 > cap) gx64 = cap;
   dim3 grid((uint32_t)gx64, (uint32_t)gy);
