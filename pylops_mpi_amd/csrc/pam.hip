@@ -48,8 +48,19 @@ extern "C" int64_t pam_reduce_ws_elems(void) { return NPARTIAL; }
 static inline int check(hipError_t e) { return (int)e; }
 
 static inline int64_t grid_1d(int64_t work) {
+  // One block per BLK work items — NO grid-stride looping by default.
+  // A/B at the bench size (fp64 axpy, single box): cap=4096 (the old
+  // default) 4.52 TB/s, 32768 4.98, 65536 5.14, uncapped 5.78 TB/s
+  // (+28%) — the stride loop was the bottleneck, not launch width.
+  // (The deterministic dot/norm reductions are unaffected: their
+  // stage-1 grid is the fixed NPARTIAL tree, not grid_1d.)
+  // PAM_EW_CAP restores a cap for A/Bs.
+  static int64_t cap = [] {
+    const char* e = getenv("PAM_EW_CAP");
+    return (int64_t)(e ? atoll(e) : 0x7FFFFFFF);
+  }();
   int64_t g = (work + BLK - 1) / BLK;
-  if (g > 4096) g = 4096;
+  if (g > cap) g = cap;
   if (g < 1) g = 1;
   return g;
 }
