@@ -923,16 +923,20 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
   }
   const int64_t mv = m / V;
   // Grid shape (PAM_FD_GY / PAM_FD_CAP override for A/Bs): one block-row
-  // per local row (gy = nrows) with ~16k blocks total measured best —
-  // bench A/B on one box, 3 reps: gy=512/cap=4096 286.7, gy=1024 293.3,
-  // gy=nrows/cap=16384 295.8 pairs/s (+3.2%); cap 32768/65536 regress.
+  // per local row (gy = nrows), column blocks sized so each block runs
+  // ~4 vector iterations (total blocks ~= work/(BLK*4)).  Bench A/Bs on
+  // one box, 2-3 reps each (pairs/s at 2048x2048x128 fp64):
+  //   old default gy<=512/cap=4096: 286.7
+  //   gy=nrows cap=16384: 295.8   cap=131072: 316.6
+  //   cap=262144: 332.0 (the peak — frac 0.719)   cap=524288: 319.6
+  //   cap=1048576 (no loop at all): 216.6 (hard regression)
   static int gycap = [] {
     const char* e = getenv("PAM_FD_GY");
     return e ? atoi(e) : 65535;
   }();
   static int totcap = [] {
     const char* e = getenv("PAM_FD_CAP");
-    return e ? atoi(e) : 16384;
+    return e ? atoi(e) : 262144;
   }();
   int gy = (int)(nrows < gycap ? nrows : gycap);
   if (gy < 1) gy = 1;

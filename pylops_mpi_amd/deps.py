@@ -9,6 +9,12 @@ pylops_mpi/utils/deps.py:58-66 (NCCL_PYLOPS_MPI / PYLOPS_MPI_CUDA_AWARE):
   PAM_FD_NT=1             nt cache hint on the stencil's output stores
                           (measured no-change on the production kernel —
                           documented negative, csrc/pam.hip)
+  PAM_FD_GY / PAM_FD_CAP  stencil launch-shape A/B knobs (defaults: one
+                          block-row per row, ~4 vector iterations per
+                          block — the measured optimum, csrc/pam.hip)
+  PAM_EW_CAP              1-D elementwise grid cap (default: none — one
+                          block per 256 vector items; the old 4096-block
+                          grid-stride loop cost 28% of axpy bandwidth)
   PAM_DISABLE_DEVSCALARS=1  run CG/CGLS with host-side scalars (one
                           blocking readback per dot) instead of the
                           single-sync device-scalar iteration (debug aid;
