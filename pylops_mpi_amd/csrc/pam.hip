@@ -876,6 +876,67 @@ __global__ void __launch_bounds__(BLK) fd_kernel(Rows<T> R, T* __restrict__ y,
   }
 }
 
+// Rolling-window stencil: each block owns one column-vector per thread
+// and WALKS a row range, carrying the 2W+1 input rows its stencil spans
+// in registers — every x element is loaded from HBM exactly once per
+// block (16 B/pt real traffic, independent of row size), where the
+// row-parallel fd_kernel relies on L2 absorbing the neighbour-row
+// re-reads (which fails once a row plane exceeds the per-XCD L2: the
+// (512,4096,256) N=8 per-rank shape measured 4.9 TB/s row-parallel).
+// Guards: load indices are clamped to the rows that exist (halo planes
+// included); term/edge masks ignore the clamped garbage, exactly as the
+// interval masks do in fd_kernel.
+template <typename T, int OP, int V>
+__global__ void __launch_bounds__(BLK) fd_roll_kernel(
+    Rows<T> R, T* __restrict__ y, int64_t row0, int64_t N, T c, int edge,
+    int64_t rbegin, int64_t rend) {
+  constexpr int W = FDDef<OP>::W;
+  constexpr int NROLL = 2 * W + 1;
+  const int64_t m = R.m, mv = m / V;
+  const int64_t jv = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (jv >= mv) return;
+  const int64_t j = jv * V;
+  // this block's row chunk
+  const int64_t nr = rend - rbegin;
+  const int64_t c0 = rbegin + (nr * blockIdx.y) / gridDim.y;
+  const int64_t c1 = rbegin + (nr * (blockIdx.y + 1)) / gridDim.y;
+  if (c1 <= c0) return;
+  const int64_t lo = R.gf ? -(int64_t)R.w : 0;
+  const int64_t hi = R.nloc - 1 + (R.gb ? R.w : 0);
+  auto clamp_row = [&](int64_t i) {
+    return i < lo ? lo : (i > hi ? hi : i);
+  };
+  T buf[NROLL][V];  // rows i-W .. i+W of this thread's columns
+#pragma unroll
+  for (int k = 0; k < NROLL; ++k)
+    loadv<T, V>(R.row(clamp_row(c0 - W + k)) + j, buf[k]);
+  for (int64_t i = c0; i < c1; ++i) {
+    const int64_t g = row0 + i;
+    T acc[V];
+#pragma unroll
+    for (int k = 0; k < V; ++k) acc[k] = (T)0;
+#pragma unroll
+    for (int t = 0; t < FDDef<OP>::NT; ++t) {
+      const Term tm = FDDef<OP>::TERMS[t];
+      if (g >= tm.lo && g <= N - 1 - tm.hi) {
+#pragma unroll
+        for (int k = 0; k < V; ++k)
+          acc[k] += (T)tm.coeff * buf[tm.off + W][k];
+      }
+    }
+    if (edge) fd_edge<T, OP, V>(R, i, j, g, N, acc);
+#pragma unroll
+    for (int k = 0; k < V; ++k) acc[k] *= c;
+    storev<T, V>(y + i * m + j, acc);
+    // roll the window one row forward
+#pragma unroll
+    for (int k = 0; k < NROLL - 1; ++k)
+#pragma unroll
+      for (int v = 0; v < V; ++v) buf[k][v] = buf[k + 1][v];
+    loadv<T, V>(R.row(clamp_row(i + 1 + W)) + j, buf[NROLL - 1]);
+  }
+}
+
 static int fd_vec_override() {
   static int v = [] {
     const char* e = getenv("PAM_FD_VEC");
@@ -946,6 +1007,39 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
   if (gx64 > cap) gx64 = cap;
   dim3 grid((uint32_t)gx64, (uint32_t)gy);
   hipStream_t s = (hipStream_t)stream;
+  // rolling-window path (fd_roll_kernel) — A/B'd NEGATIVE as default:
+  // bench shape 331 -> 306 pairs/s, and the long-row (512,4096,256)
+  // shape it was built for is unchanged (4.9 TB/s both ways: that
+  // shape's wall is not neighbour-row re-read traffic).  Kept behind
+  // PAM_FD_ROLL=1 with PAM_FD_ROLL_TGT for round-2 investigation.
+  static int rollov = [] {
+    const char* e = getenv("PAM_FD_ROLL");
+    return e ? atoi(e) : 0;
+  }();
+  static int rolltgt = [] {
+    const char* e = getenv("PAM_FD_ROLL_TGT");
+    return e ? atoi(e) : 4096;
+  }();
+  if (rollov && V > 1) {
+    const int64_t mv2 = m / V;
+    int64_t gx = (mv2 + BLK - 1) / BLK;
+    int64_t gyr = rolltgt / (gx ? gx : 1);
+    const int64_t minchunk = 16;  // amortize the 2W+1-row preload
+    int64_t maxgy = (nrows + minchunk - 1) / minchunk;
+    if (gyr > maxgy) gyr = maxgy;
+    if (gyr < 1) gyr = 1;
+    if (gyr > 65535) gyr = 65535;
+    dim3 gridr((uint32_t)gx, (uint32_t)gyr);
+    if (V == 4)
+      hipLaunchKernelGGL((fd_roll_kernel<T, OP, 4>), gridr, dim3(BLK), 0, s,
+                         R, (T*)y, row0, nglob, (T)coeff, edge, rbegin,
+                         rend);
+    else
+      hipLaunchKernelGGL((fd_roll_kernel<T, OP, 2>), gridr, dim3(BLK), 0, s,
+                         R, (T*)y, row0, nglob, (T)coeff, edge, rbegin,
+                         rend);
+    return check(hipGetLastError());
+  }
   const bool nt = fd_nt_override() != 0;
   if (nt) {
     if (V == 4)
