@@ -20,8 +20,9 @@ MI355X-first differences (both bit-identical to the reference recurrence):
   (ref cls_basic.py:433) requires — instead of one blocking readback per
   dot/norm (5 per CGLS iteration).  Measured r01 (scripts/
   gpu_cgls_probe.py): at the bench config the iteration is GPU-bound
-  (~20 ms of kernels) and the paths are within a few percent
-  (20.3 vs 21.1 ms/iter); at latency-bound sizes (256x256x64 and below)
+  (~17 ms of kernels after the round-final launch-shape retunes) and
+  the paths are within ~5% (17.9 vs 18.7 ms/iter); at latency-bound
+  sizes (256x256x64 and below)
   the device path is ~35% faster/iter (0.23 vs 0.31 ms), and at N>1 it
   keeps collective latency off the host critical path.  The device path
   covers real SCATTER mask-free CUDA arrays (the north
