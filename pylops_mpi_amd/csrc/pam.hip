@@ -876,64 +876,94 @@ __global__ void __launch_bounds__(BLK) fd_kernel(Rows<T> R, T* __restrict__ y,
   }
 }
 
-// Rolling-window stencil: each block owns one column-vector per thread
-// and WALKS a row range, carrying the 2W+1 input rows its stencil spans
-// in registers — every x element is loaded from HBM exactly once per
-// block (16 B/pt real traffic, independent of row size), where the
-// row-parallel fd_kernel relies on L2 absorbing the neighbour-row
-// re-reads (which fails once a row plane exceeds the per-XCD L2: the
-// (512,4096,256) N=8 per-rank shape measured 4.9 TB/s row-parallel).
-// Guards: load indices are clamped to the rows that exist (halo planes
-// included); term/edge masks ignore the clamped garbage, exactly as the
-// interval masks do in fd_kernel.
-template <typename T, int OP, int V>
+// Rolling-window stencil (EXPERIMENTAL, PAM_FD_ROLL=1; r01 status
+// below): each block walks a row range carrying the 2W+1 input rows in
+// registers — every x element loaded from HBM exactly once, where the
+// row-parallel fd_kernel relies on L2 absorbing neighbour-row re-reads
+// (which fades at long rows: the (512,4096,256) N=8 per-rank shape
+// measures 4.9 TB/s row-parallel vs 5.7 at the bench shape; a pad/
+// non-pow2 probe — scripts/probe_rowstride.hip — ruled out row-stride
+// channel aliasing, leaving re-read absorption as the cause).
+// r01 measurements of THIS kernel: single chain/thread 4.8-5.2 TB/s
+// (memory-latency-bound: ONE dependent load per row step); CV=4
+// independent chains collapse to ~2.0 TB/s because indexing the rolling
+// register window by the constexpr table's row offset (buf[q][tm.off+W])
+// does not constant-fold through the Term struct and the windows spill
+// to scratch.  Round-2 fix: specialize the window access per op (switch
+// on constexpr offsets) so the window stays in registers, then re-A/B.
+// Guards: load indices are clamped to existing rows (halo planes
+// included); term/edge masks ignore the clamped garbage.
+template <typename T, int OP, int V, int CV>
 __global__ void __launch_bounds__(BLK) fd_roll_kernel(
     Rows<T> R, T* __restrict__ y, int64_t row0, int64_t N, T c, int edge,
     int64_t rbegin, int64_t rend) {
   constexpr int W = FDDef<OP>::W;
   constexpr int NROLL = 2 * W + 1;
   const int64_t m = R.m, mv = m / V;
-  const int64_t jv = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (jv >= mv) return;
-  const int64_t j = jv * V;
-  // this block's row chunk
+  // CV independent column-vectors per thread (separate rolling windows)
+  // => CV outstanding loads per row step instead of 1: the single-chain
+  // form was memory-latency-bound (one dependent load per row).  Chains
+  // are BLOCK-interleaved (chain q of lane t at column block*BLK*CV +
+  // q*BLK + t) so each chain's wave accesses stay fully coalesced.
+  const int64_t base = (int64_t)blockIdx.x * blockDim.x * CV + threadIdx.x;
   const int64_t nr = rend - rbegin;
   const int64_t c0 = rbegin + (nr * blockIdx.y) / gridDim.y;
   const int64_t c1 = rbegin + (nr * (blockIdx.y + 1)) / gridDim.y;
-  if (c1 <= c0) return;
+  if (c1 <= c0 || base >= mv) return;
   const int64_t lo = R.gf ? -(int64_t)R.w : 0;
   const int64_t hi = R.nloc - 1 + (R.gb ? R.w : 0);
   auto clamp_row = [&](int64_t i) {
     return i < lo ? lo : (i > hi ? hi : i);
   };
-  T buf[NROLL][V];  // rows i-W .. i+W of this thread's columns
+  // constant-indexed per-chain state (a runtime-compacted chain array
+  // made the windows spill to scratch: 5.2 -> 1.9 TB/s)
+  int64_t js[CV];
+  bool vq[CV];
 #pragma unroll
-  for (int k = 0; k < NROLL; ++k)
-    loadv<T, V>(R.row(clamp_row(c0 - W + k)) + j, buf[k]);
+  for (int q = 0; q < CV; ++q) {
+    const int64_t jq = base + (int64_t)q * blockDim.x;
+    vq[q] = jq < mv;
+    js[q] = (vq[q] ? jq : mv - 1) * V;  // clamped: loads stay in range
+  }
+  T buf[CV][NROLL][V];  // rows i-W .. i+W of each owned column vector
+#pragma unroll
+  for (int q = 0; q < CV; ++q)
+#pragma unroll
+    for (int k = 0; k < NROLL; ++k)
+      loadv<T, V>(R.row(clamp_row(c0 - W + k)) + js[q], buf[q][k]);
   for (int64_t i = c0; i < c1; ++i) {
     const int64_t g = row0 + i;
-    T acc[V];
+    // issue ALL next-row loads first (CV independent chains)
+    T nxt[CV][V];
+    const int64_t rn = clamp_row(i + 1 + W);
 #pragma unroll
-    for (int k = 0; k < V; ++k) acc[k] = (T)0;
+    for (int q = 0; q < CV; ++q)
+      loadv<T, V>(R.row(rn) + js[q], nxt[q]);
 #pragma unroll
-    for (int t = 0; t < FDDef<OP>::NT; ++t) {
-      const Term tm = FDDef<OP>::TERMS[t];
-      if (g >= tm.lo && g <= N - 1 - tm.hi) {
+    for (int q = 0; q < CV; ++q) {
+      T acc[V];
 #pragma unroll
-        for (int k = 0; k < V; ++k)
-          acc[k] += (T)tm.coeff * buf[tm.off + W][k];
+      for (int k = 0; k < V; ++k) acc[k] = (T)0;
+#pragma unroll
+      for (int t = 0; t < FDDef<OP>::NT; ++t) {
+        const Term tm = FDDef<OP>::TERMS[t];
+        if (g >= tm.lo && g <= N - 1 - tm.hi) {
+#pragma unroll
+          for (int k = 0; k < V; ++k)
+            acc[k] += (T)tm.coeff * buf[q][tm.off + W][k];
+        }
       }
+      if (edge) fd_edge<T, OP, V>(R, i, js[q], g, N, acc);
+#pragma unroll
+      for (int k = 0; k < V; ++k) acc[k] *= c;
+      if (vq[q]) storev<T, V>(y + i * m + js[q], acc);
+#pragma unroll
+      for (int k = 0; k < NROLL - 1; ++k)
+#pragma unroll
+        for (int v = 0; v < V; ++v) buf[q][k][v] = buf[q][k + 1][v];
+#pragma unroll
+      for (int v = 0; v < V; ++v) buf[q][NROLL - 1][v] = nxt[q][v];
     }
-    if (edge) fd_edge<T, OP, V>(R, i, j, g, N, acc);
-#pragma unroll
-    for (int k = 0; k < V; ++k) acc[k] *= c;
-    storev<T, V>(y + i * m + j, acc);
-    // roll the window one row forward
-#pragma unroll
-    for (int k = 0; k < NROLL - 1; ++k)
-#pragma unroll
-      for (int v = 0; v < V; ++v) buf[k][v] = buf[k + 1][v];
-    loadv<T, V>(R.row(clamp_row(i + 1 + W)) + j, buf[NROLL - 1]);
   }
 }
 
@@ -1021,8 +1051,9 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
     return e ? atoi(e) : 4096;
   }();
   if (rollov && V > 1) {
+    constexpr int CV = 4;  // independent rolling chains per thread
     const int64_t mv2 = m / V;
-    int64_t gx = (mv2 + BLK - 1) / BLK;
+    int64_t gx = (mv2 + (int64_t)BLK * CV - 1) / ((int64_t)BLK * CV);
     int64_t gyr = rolltgt / (gx ? gx : 1);
     const int64_t minchunk = 16;  // amortize the 2W+1-row preload
     int64_t maxgy = (nrows + minchunk - 1) / minchunk;
@@ -1031,13 +1062,13 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
     if (gyr > 65535) gyr = 65535;
     dim3 gridr((uint32_t)gx, (uint32_t)gyr);
     if (V == 4)
-      hipLaunchKernelGGL((fd_roll_kernel<T, OP, 4>), gridr, dim3(BLK), 0, s,
-                         R, (T*)y, row0, nglob, (T)coeff, edge, rbegin,
-                         rend);
+      hipLaunchKernelGGL((fd_roll_kernel<T, OP, 4, CV>), gridr, dim3(BLK),
+                         0, s, R, (T*)y, row0, nglob, (T)coeff, edge,
+                         rbegin, rend);
     else
-      hipLaunchKernelGGL((fd_roll_kernel<T, OP, 2>), gridr, dim3(BLK), 0, s,
-                         R, (T*)y, row0, nglob, (T)coeff, edge, rbegin,
-                         rend);
+      hipLaunchKernelGGL((fd_roll_kernel<T, OP, 2, CV>), gridr, dim3(BLK),
+                         0, s, R, (T*)y, row0, nglob, (T)coeff, edge,
+                         rbegin, rend);
     return check(hipGetLastError());
   }
   const bool nt = fd_nt_override() != 0;
