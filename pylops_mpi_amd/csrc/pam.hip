@@ -31,6 +31,7 @@
 #include <math.h>
 #include <stdint.h>
 #include <stdlib.h>
+#include <utility>
 
 #include "../../include/pam.h"
 #include "fd_defs.h"
@@ -876,6 +877,31 @@ __global__ void __launch_bounds__(BLK) fd_kernel(Rows<T> R, T* __restrict__ y,
   }
 }
 
+
+// constexpr-folded term accumulation for the rolling window: `tm` is a
+// constexpr LOCAL here, so buf[tm.off + W] is a CONSTANT register index
+// (the plain `for (t) { Term tm = TERMS[t]; buf[tm.off+W] }` form keeps
+// the window in scratch — measured 2.0 TB/s vs in-register)
+template <typename T, int OP, int V, int W, int t>
+__device__ __forceinline__ void roll_term(const T (&win)[2 * W + 1][V],
+                                          T (&acc)[V], int64_t g,
+                                          int64_t N) {
+  constexpr Term tm = FDDef<OP>::TERMS[t];
+  if (g >= tm.lo && g <= N - 1 - tm.hi) {
+#pragma unroll
+    for (int k = 0; k < V; ++k)
+      acc[k] += (T)tm.coeff * win[tm.off + W][k];
+  }
+}
+
+template <typename T, int OP, int V, int W, int... Ts>
+__device__ __forceinline__ void roll_terms(const T (&win)[2 * W + 1][V],
+                                           T (&acc)[V], int64_t g,
+                                           int64_t N,
+                                           std::integer_sequence<int, Ts...>) {
+  (roll_term<T, OP, V, W, Ts>(win, acc, g, N), ...);
+}
+
 // Rolling-window stencil (EXPERIMENTAL, PAM_FD_ROLL=1; r01 status
 // below): each block walks a row range carrying the 2W+1 input rows in
 // registers — every x element loaded from HBM exactly once, where the
@@ -884,13 +910,16 @@ __global__ void __launch_bounds__(BLK) fd_kernel(Rows<T> R, T* __restrict__ y,
 // measures 4.9 TB/s row-parallel vs 5.7 at the bench shape; a pad/
 // non-pow2 probe — scripts/probe_rowstride.hip — ruled out row-stride
 // channel aliasing, leaving re-read absorption as the cause).
-// r01 measurements of THIS kernel: single chain/thread 4.8-5.2 TB/s
-// (memory-latency-bound: ONE dependent load per row step); CV=4
-// independent chains collapse to ~2.0 TB/s because indexing the rolling
-// register window by the constexpr table's row offset (buf[q][tm.off+W])
-// does not constant-fold through the Term struct and the windows spill
-// to scratch.  Round-2 fix: specialize the window access per op (switch
-// on constexpr offsets) so the window stays in registers, then re-A/B.
+// r01 measurements of THIS kernel: single chain/thread 4.8-5.2 TB/s;
+// CV=4 chains with a runtime-indexed window spilled to scratch
+// (~2.0 TB/s); with the constexpr-folded roll_terms below the window
+// stays in registers and CV=4 reaches 4.9 (long rows) / 5.2 (bench
+// shape) TB/s — STILL below the row-parallel kernel (4.9 / 5.7).  Key
+// datum: rolling reads each x element exactly once, so its 4.9 TB/s on
+// the long-row shape is a TRUE 16 B/pt streaming rate — the long-row
+// gap is therefore NOT neighbour-row re-read traffic; something about
+// the 1-load+1-store-per-row-step pattern (vs axpy's 5.78 on the same
+// mix) is the round-2 question.
 // Guards: load indices are clamped to existing rows (halo planes
 // included); term/edge masks ignore the clamped garbage.
 template <typename T, int OP, int V, int CV>
@@ -944,15 +973,9 @@ __global__ void __launch_bounds__(BLK) fd_roll_kernel(
       T acc[V];
 #pragma unroll
       for (int k = 0; k < V; ++k) acc[k] = (T)0;
-#pragma unroll
-      for (int t = 0; t < FDDef<OP>::NT; ++t) {
-        const Term tm = FDDef<OP>::TERMS[t];
-        if (g >= tm.lo && g <= N - 1 - tm.hi) {
-#pragma unroll
-          for (int k = 0; k < V; ++k)
-            acc[k] += (T)tm.coeff * buf[q][tm.off + W][k];
-        }
-      }
+      roll_terms<T, OP, V, W>(
+          buf[q], acc, g, N,
+          std::make_integer_sequence<int, FDDef<OP>::NT>{});
       if (edge) fd_edge<T, OP, V>(R, i, js[q], g, N, acc);
 #pragma unroll
       for (int k = 0; k < V; ++k) acc[k] *= c;
