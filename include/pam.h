@@ -224,15 +224,23 @@ int pam_transpose(void* stream, const void* A, void* At, int64_t nr,
                   int64_t nc, int dtype);
 
 /* Batched complex GEMM: for b in [0, batch):
- *   C_b = op(A_b) @ B_b   with op = N (opa=0) or conjugate-transpose
- *   (opa=1; then A_b is [K, M]).
+ *   C_b (+)= op(A_b) @ B_b  with op = N (opa=0) or conjugate-transpose
+ *   (opa=1; then A_b is [K, M]); accumulate != 0 adds into C.
  * A_b = A + b*strideA etc. (strides in ELEMENTS = complex pairs).
  * The Fredholm1 batched integral kernel (ref signalprocessing/
- * Fredholm1.py:123 `ncp.matmul(G, x)` and :149-156 adjoint). */
+ * Fredholm1.py:123 `ncp.matmul(G, x)` and :149-156 adjoint), and —
+ * with batch=1 — the complex MatrixMult panel product
+ * (ref MatrixMult.py:341-427,610-765 complex dtypes). */
 int pam_cgemm_batched(void* stream, const void* A, const void* B, void* C,
                       int64_t batch, int64_t M, int64_t N, int64_t K,
                       int64_t strideA, int64_t strideB, int64_t strideC,
-                      int opa, int dtype);
+                      int opa, int accumulate, int dtype);
+
+/* Complex (conj-)transpose on interleaved (re,im) pairs: At = A^T
+ * (conj=0) or A^H (conj=1) — the complex MatrixMult adjoint panels
+ * (ref MatrixMult.py:416,737 `A.T.conj()`). */
+int pam_ctranspose(void* stream, const void* A, void* At, int64_t nr,
+                   int64_t nc, int conj, int dtype);
 
 #ifdef __cplusplus
 }

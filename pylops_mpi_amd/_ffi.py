@@ -71,7 +71,10 @@ _SIGS = {
                            ctypes.c_void_p, ctypes.c_int64, ctypes.c_int64,
                            ctypes.c_int64, ctypes.c_int64, ctypes.c_int64,
                            ctypes.c_int64, ctypes.c_int64, ctypes.c_int,
-                           ctypes.c_int], ctypes.c_int),
+                           ctypes.c_int, ctypes.c_int], ctypes.c_int),
+    "pam_ctranspose": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                        ctypes.c_int64, ctypes.c_int64, ctypes.c_int,
+                        ctypes.c_int], ctypes.c_int),
     "pam_norm_local": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
                         ctypes.c_int, ctypes.c_double, ctypes.c_void_p,
                         ctypes.c_void_p, ctypes.c_int], ctypes.c_int),

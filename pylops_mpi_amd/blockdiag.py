@@ -13,7 +13,7 @@ import numpy as np
 import torch
 
 from .comm import PamComm, get_default_comm
-from .distributedarray import DistributedArray, Partition
+from .distributedarray import DistributedArray, Partition, as_torch_dtype
 from .linearoperator import MPILinearOperator
 from .localops import LocalOperator
 from .rebalance import rebalance_1d
@@ -60,9 +60,7 @@ class MPIBlockDiag(MPILinearOperator):
             seg = local[int(offs[iop]): int(offs[iop + 1])]
             pieces.append(op.matvec(seg) if forward else op.rmatvec(seg))
         out = torch.cat(pieces) if len(pieces) != 1 else pieces[0].reshape(-1)
-        out = out.to(
-            {np.dtype(np.float64): torch.float64,
-             np.dtype(np.float32): torch.float32}[np.dtype(self.dtype)])
+        out = out.to(as_torch_dtype(np.dtype(self.dtype)))
         gshape = self.shape[0] if forward else self.shape[1]
         return DistributedArray(
             int(gshape), comm, Partition.SCATTER, 0, local_array=out,

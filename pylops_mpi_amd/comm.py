@@ -68,24 +68,31 @@ class PamComm:
                  = None) -> "PamComm":
         """The reference's ``comm.Split(color, key)``
         (ref MatrixMult.py:305-306): ``colors``/``keys`` are indexed by
-        GLOBAL rank and must be identical on every rank (they are
-        computed from the deterministic grid layout).  Every process
-        creates every group (torch.distributed requirement) and returns
-        its own."""
+        THIS communicator's group rank (0..self.size) and must be
+        identical on every member (they are computed from the
+        deterministic grid layout).  Valid on sub-communicators too:
+        member group ranks are translated to global ranks through
+        ``self.ranks``, and groups are created with
+        ``use_local_synchronization`` so only the members of each new
+        group enter the call — ranks outside ``self`` (e.g. the inactive
+        ranks of ``active_grid_comm``) need not participate."""
         if not self._use_dist:
             return PamComm(0, 1, self.device, use_dist=False)
+        if len(colors) != self.size:
+            raise ValueError(
+                f"colors must have one entry per group rank: "
+                f"{len(colors)} != {self.size}")
         if keys is None:
             keys = list(range(len(colors)))
-        mine = None
-        my_ranks = None
-        for color in sorted(set(colors)):
-            members = sorted((r for r in range(len(colors))
-                              if colors[r] == color),
-                             key=lambda r: (keys[r], r))
-            g = dist.new_group(ranks=members)
-            if colors[dist.get_rank()] == color:
-                mine, my_ranks = g, members
-        gr = my_ranks.index(dist.get_rank())
+        my_color = colors[self.rank]
+        members = sorted((r for r in range(len(colors))
+                          if colors[r] == my_color),
+                         key=lambda r: (keys[r], r))
+        my_ranks = [self.ranks[r] for r in members]
+        mine = dist.new_group(ranks=my_ranks,
+                              use_local_synchronization=True) \
+            if len(my_ranks) > 1 else None
+        gr = members.index(self.rank)
         return PamComm(gr, len(my_ranks), self.device,
                        use_dist=len(my_ranks) > 1, group=mine,
                        ranks=my_ranks, gloo_group=self._gloo_group)
@@ -93,9 +100,13 @@ class PamComm:
     # ------------------------------------------------------- collectives
     def allreduce_(self, t: torch.Tensor, op: str = "sum") -> torch.Tensor:
         """In-place allreduce of a tensor (scalar dots/norms,
-        ref Distributed.py:35-73)."""
+        ref Distributed.py:35-73).  Complex sums travel as the (re,im)
+        real view — the reference's NCCL scheme (ref utils/_nccl.py:23-35
+        complex-as-2-floats), and gloo has no complex types."""
         if self._use_dist:
-            dist.all_reduce(t, op=getattr(dist.ReduceOp, _REDUCE_OPS[op]),
+            rt = torch.view_as_real(t) if (t.is_complex() and op == "sum") \
+                else t
+            dist.all_reduce(rt, op=getattr(dist.ReduceOp, _REDUCE_OPS[op]),
                             group=self._group)
         return t
 

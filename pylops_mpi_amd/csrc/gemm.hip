@@ -304,7 +304,7 @@ extern "C" int pam_gemm(void* stream, const void* A, const void* B, void* C,
 // numbers are in DESIGN.md) — the batch dimension fills the chip either
 // way, but only MFMA reaches the matrix-core rate.
 // ---------------------------------------------------------------------------
-template <typename T, bool CT, int BK = 16>
+template <typename T, bool CT, bool ACC = false, int BK = 16>
 __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
     const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
     int64_t M, int64_t N, int64_t K, int64_t strideA, int64_t strideB,
@@ -496,8 +496,13 @@ __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
         const int64_t rr = r0 + CFG::crow(lane, reg);
         if (rr < M) {
           const int64_t off = 2 * (rr * N + cc);
-          Cb[off] = s_rr[mi][nj][reg] - s_ii[mi][nj][reg];
-          Cb[off + 1] = s_ri[mi][nj][reg] + s_ir[mi][nj][reg];
+          if constexpr (ACC) {
+            Cb[off] += s_rr[mi][nj][reg] - s_ii[mi][nj][reg];
+            Cb[off + 1] += s_ri[mi][nj][reg] + s_ir[mi][nj][reg];
+          } else {
+            Cb[off] = s_rr[mi][nj][reg] - s_ii[mi][nj][reg];
+            Cb[off + 1] = s_ri[mi][nj][reg] + s_ir[mi][nj][reg];
+          }
         }
       }
     }
@@ -506,7 +511,8 @@ __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
 template <typename T>
 static int cgemm_launch(void* stream, const void* A, const void* B, void* C,
                         int64_t batch, int64_t M, int64_t N, int64_t K,
-                        int64_t sA, int64_t sB, int64_t sC, int opa) {
+                        int64_t sA, int64_t sB, int64_t sC, int opa,
+                        int acc) {
   if (batch <= 0 || M <= 0 || N <= 0 || K < 0 || !A || !B || !C)
     return PAM_EARG;
   dim3 grid((uint32_t)((N + 63) / 64), (uint32_t)((M + 63) / 64),
@@ -516,27 +522,97 @@ static int cgemm_launch(void* stream, const void* A, const void* B, void* C,
   // was A/B'd NEGATIVE at the cfg5 rmatvec shape (0.432 -> 0.518 ms):
   // the 133 KB LDS footprint drops occupancy to 2 WG/CU, which costs
   // more than the three extra barriers it saves.  BK=16 for all K.
-  if (opa)
-    hipLaunchKernelGGL((cgemm_batched_kernel<T, true>), grid, dim3(GBLK), 0,
-                       s, (const T*)A, (const T*)B, (T*)C, M, N, K, sA, sB,
-                       sC);
+  if (opa && acc)
+    hipLaunchKernelGGL((cgemm_batched_kernel<T, true, true>), grid,
+                       dim3(GBLK), 0, s, (const T*)A, (const T*)B, (T*)C, M,
+                       N, K, sA, sB, sC);
+  else if (opa)
+    hipLaunchKernelGGL((cgemm_batched_kernel<T, true, false>), grid,
+                       dim3(GBLK), 0, s, (const T*)A, (const T*)B, (T*)C, M,
+                       N, K, sA, sB, sC);
+  else if (acc)
+    hipLaunchKernelGGL((cgemm_batched_kernel<T, false, true>), grid,
+                       dim3(GBLK), 0, s, (const T*)A, (const T*)B, (T*)C, M,
+                       N, K, sA, sB, sC);
   else
-    hipLaunchKernelGGL((cgemm_batched_kernel<T, false>), grid, dim3(GBLK), 0,
-                       s, (const T*)A, (const T*)B, (T*)C, M, N, K, sA, sB,
-                       sC);
+    hipLaunchKernelGGL((cgemm_batched_kernel<T, false, false>), grid,
+                       dim3(GBLK), 0, s, (const T*)A, (const T*)B, (T*)C, M,
+                       N, K, sA, sB, sC);
   return gcheck(hipGetLastError());
 }
 
 extern "C" int pam_cgemm_batched(void* stream, const void* A, const void* B,
                                  void* C, int64_t batch, int64_t M, int64_t N,
                                  int64_t K, int64_t strideA, int64_t strideB,
-                                 int64_t strideC, int opa, int dtype) {
+                                 int64_t strideC, int opa, int accumulate,
+                                 int dtype) {
   if (dtype == PAM_C128)
     return cgemm_launch<double>(stream, A, B, C, batch, M, N, K, strideA,
-                                strideB, strideC, opa);
+                                strideB, strideC, opa, accumulate);
   if (dtype == PAM_C64)
     return cgemm_launch<float>(stream, A, B, C, batch, M, N, K, strideA,
-                               strideB, strideC, opa);
+                               strideB, strideC, opa, accumulate);
+  return PAM_EDTYPE;
+}
+
+// ---------------------------------------------------------------------------
+// complex (conj-)transpose: At[c][r] = (conj?)(A[r][c]) on interleaved
+// (re,im) pairs — materializes A^H for the complex MatrixMult adjoint
+// panels (ref MatrixMult.py:416,737 "A.T.conj()").  Same 32x32 LDS tile
+// as transpose_kernel, elements are (re,im) pairs.
+// ---------------------------------------------------------------------------
+template <typename T, bool CONJ>
+__global__ void __launch_bounds__(GBLK) ctranspose_kernel(
+    const T* __restrict__ A, T* __restrict__ At, int64_t nr, int64_t nc) {
+  __shared__ T tile[32][33][2];
+  const int64_t r0 = (int64_t)blockIdx.y * 32;
+  const int64_t c0 = (int64_t)blockIdx.x * 32;
+  const int tr = threadIdx.x / 32;
+  const int tc = threadIdx.x % 32;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int64_t r = r0 + tr + 8 * i;
+    if (r < nr && c0 + tc < nc) {
+      const int64_t off = 2 * (r * nc + c0 + tc);
+      tile[tr + 8 * i][tc][0] = A[off];
+      tile[tr + 8 * i][tc][1] = CONJ ? -A[off + 1] : A[off + 1];
+    }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int64_t c = c0 + tr + 8 * i;
+    if (c < nc && r0 + tc < nr) {
+      const int64_t off = 2 * (c * nr + r0 + tc);
+      At[off] = tile[tc][tr + 8 * i][0];
+      At[off + 1] = tile[tc][tr + 8 * i][1];
+    }
+  }
+}
+
+extern "C" int pam_ctranspose(void* stream, const void* A, void* At,
+                              int64_t nr, int64_t nc, int conj, int dtype) {
+  if (nr <= 0 || nc <= 0 || !A || !At) return PAM_EARG;
+  dim3 grid((uint32_t)((nc + 31) / 32), (uint32_t)((nr + 31) / 32));
+  hipStream_t s = (hipStream_t)stream;
+  if (dtype == PAM_C128) {
+    if (conj)
+      hipLaunchKernelGGL((ctranspose_kernel<double, true>), grid, dim3(GBLK),
+                         0, s, (const double*)A, (double*)At, nr, nc);
+    else
+      hipLaunchKernelGGL((ctranspose_kernel<double, false>), grid, dim3(GBLK),
+                         0, s, (const double*)A, (double*)At, nr, nc);
+    return gcheck(hipGetLastError());
+  }
+  if (dtype == PAM_C64) {
+    if (conj)
+      hipLaunchKernelGGL((ctranspose_kernel<float, true>), grid, dim3(GBLK),
+                         0, s, (const float*)A, (float*)At, nr, nc);
+    else
+      hipLaunchKernelGGL((ctranspose_kernel<float, false>), grid, dim3(GBLK),
+                         0, s, (const float*)A, (float*)At, nr, nc);
+    return gcheck(hipGetLastError());
+  }
   return PAM_EDTYPE;
 }
 

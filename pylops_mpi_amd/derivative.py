@@ -76,9 +76,6 @@ class _FDBase(MPILinearOperator):
         if P == 1 or w == 0:
             return None, None
         nloc = planes.shape[0]
-        if nloc < w and 0 < r < P - 1:
-            raise ValueError(
-                f"Local Shape at rank={r} along axis=0 should be > {w}")
         send_prev = planes[:w].contiguous() if r > 0 else None
         send_next = planes[-w:].contiguous() if r < P - 1 else None
         gshape = (w,) + tuple(planes.shape[1:])
@@ -106,6 +103,17 @@ class _FDBase(MPILinearOperator):
         planes = flat.view(nloc, m) if m > 1 else flat.view(nloc, 1)
         x._require_compute()
         w = int(_ffi.lib().pam_fd_halo_width(op))
+        if comm.size > 1 and w > 0:
+            # every rank with a neighbour sends w planes (rank 0 to the
+            # next, rank P-1 to the previous), so EVERY rank's block must
+            # hold >= w planes; shapes is deterministic on all ranks, so
+            # this raises consistently everywhere instead of one rank
+            # posting a mismatched irecv (ref :996-1002 sender guard)
+            for rr in range(comm.size):
+                if shapes[rr][0] < w:
+                    raise ValueError(
+                        f"Local Shape at rank={rr} along axis=0 should "
+                        f"be > {w}")
         y_out = torch.empty_like(planes)
         y = y_out
         if planes.is_complex():
@@ -137,9 +145,6 @@ class _FDBase(MPILinearOperator):
             # halo-independent interior rows concurrently, then the
             # boundary rows once the planes have arrived
             r, P = comm.rank, comm.size
-            if nloc < w and 0 < r < P - 1:
-                raise ValueError(
-                    f"Local Shape at rank={r} along axis=0 should be > {w}")
             gshape = (w,) + tuple(planes.shape[1:])
             gf = torch.empty(gshape, dtype=planes.dtype,
                              device=planes.device) if r > 0 else None

@@ -386,13 +386,17 @@ class DistributedArray:
         nloc = t.shape[0]
         if width == 0 or P == 1:
             return None, None
-        if width > nloc:
-            # ref :996-1002 sender-side guard
-            raise ValueError(
-                f"Local Shape at rank={r} along axis=0 should be > {width}: "
-                f"dim(0) {nloc} < {width}; to achieve this use "
-                f"NUM_PROCESSES <= "
-                f"{max(1, self._global_shape[0] // width)}")
+        # ref :996-1002 sender-side guard, made deterministic: every rank
+        # with a neighbour sends ``width`` planes, and all ranks know all
+        # local shapes, so raise identically everywhere instead of one
+        # rank posting a mismatched irecv
+        for rr in range(P):
+            if self._all_local_shapes[rr][0] < width:
+                raise ValueError(
+                    f"Local Shape at rank={rr} along axis=0 should be > "
+                    f"{width}: dim(0) {self._all_local_shapes[rr][0]} < "
+                    f"{width}; to achieve this use NUM_PROCESSES <= "
+                    f"{max(1, self._global_shape[0] // width)}")
         send_prev = t[:width].contiguous() if r > 0 else None
         send_next = t[-width:].contiguous() if r < P - 1 else None
         gf = torch.empty_like(t[:width]) if r > 0 else None
@@ -415,10 +419,13 @@ class DistributedArray:
                                  dtype=t.dtype, device=t.device)
             else:
                 gf = None
-            if cells_front > t.shape[0] and r < P - 1:
-                raise ValueError(
-                    f"Local Shape at rank={r} along axis={self._axis} "
-                    f"should be > {cells_front}")
+            # senders are ranks < P-1; validate ALL of them on every rank
+            # so the raise is deterministic (ref :996-1002)
+            for rr in range(P - 1):
+                if cells_front > self._all_local_shapes[rr][self._axis]:
+                    raise ValueError(
+                        f"Local Shape at rank={rr} along axis={self._axis} "
+                        f"should be > {cells_front}")
             self._base_comm.sendrecv_neighbors(
                 None, t[-cells_front:].contiguous() if r < P - 1 else None,
                 gf, None)
@@ -430,10 +437,12 @@ class DistributedArray:
                                  dtype=t.dtype, device=t.device)
             else:
                 gb = None
-            if cells_back > t.shape[0] and r > 0:
-                raise ValueError(
-                    f"Local Shape at rank={r} along axis={self._axis} "
-                    f"should be > {cells_back}")
+            # senders are ranks > 0; validate ALL of them on every rank
+            for rr in range(1, P):
+                if cells_back > self._all_local_shapes[rr][self._axis]:
+                    raise ValueError(
+                        f"Local Shape at rank={rr} along axis={self._axis} "
+                        f"should be > {cells_back}")
             self._base_comm.sendrecv_neighbors(
                 t[:cells_back].contiguous() if r > 0 else None, None,
                 None, gb)
@@ -477,7 +486,9 @@ class DistributedArray:
         return self
 
     def sub(self, other: "DistributedArray") -> "DistributedArray":
-        # bitwise equal to ref __sub__ = add(-other) (ref :624-625)
+        # bitwise equal to ref __sub__ = add(-other) (ref :624-625),
+        # including its mask validation (add() raises on mismatch)
+        self._check_mask(other)
         self._check_partition_shape(other)
         self._require_compute()
         out = self._like()
@@ -527,6 +538,7 @@ class DistributedArray:
         """self += alpha * x (REAL alpha), fused — componentwise on the
         real view for complex arrays (CG/CGLS scalars are real,
         ref cls_basic.py:389-395)."""
+        self._check_mask(x)
         self._check_partition_shape(x)
         self._require_compute()
         a, n, dt = self._ew()
@@ -538,6 +550,7 @@ class DistributedArray:
 
     def xpby_(self, x: "DistributedArray", beta: float) -> "DistributedArray":
         """self = x + beta * self (REAL beta), fused (CGLS c = r + b*c)."""
+        self._check_mask(x)
         self._check_partition_shape(x)
         self._require_compute()
         a, n, dt = self._ew()

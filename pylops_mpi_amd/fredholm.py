@@ -67,7 +67,7 @@ class MPIFredholm1(MPILinearOperator):
         if A.is_complex():
             _ffi.checked(_ffi.lib().pam_cgemm_batched(
                 _stream(A), A.data_ptr(), X.data_ptr(), Y.data_ptr(), batch,
-                M, N, K, A.shape[1] * A.shape[2], K * N, M * N, opa,
+                M, N, K, A.shape[1] * A.shape[2], K * N, M * N, opa, 0,
                 _ffi.dtype_code(A.dtype)), "cgemm_batched")
         else:
             # real dtypes: per-slice MFMA panels (small batch loop)

@@ -11,9 +11,11 @@ Drop-in for /root/reference/pylops_mpi/basicoperators/MatrixMult.py:
 
 MI355X-first differences:
   * local panels run on hand-written MFMA kernels (pam_gemm:
-    v_mfma_f32_32x32x2_f32 exact-f32 / v_mfma_f64_16x16x4_f64) instead of
-    CuPy matmul; A^T panels are materialized by an LDS-tiled transpose
-    kernel (pam_transpose; `saveAt=True` stores it once, ref :317-318);
+    v_mfma_f32_32x32x2_f32 exact-f32 / v_mfma_f64_16x16x4_f64; complex
+    dtypes on the 4-chain MFMA pam_cgemm_batched with batch=1) instead
+    of CuPy matmul; A^T/A^H panels are materialized by LDS-tiled
+    transpose kernels (pam_transpose / pam_ctranspose; `saveAt=True`
+    stores it once, ref :317-318);
   * row/col communicators are RCCL process groups (comm.split_by); the
     SUMMA adjoint's tag-routed A^T exchange (ref :745-760) is posted as
     one batched isend/irecv round per step — RCCL has no tags, and each
@@ -127,22 +129,40 @@ class _MatMultBase(MPILinearOperator):
             raise RuntimeError(
                 "pam: compute ops require a CUDA (MI355X) device tensor — "
                 "there is no CPU compute path")
-        _ffi.checked(_ffi.lib().pam_gemm(
-            _stream(A), A.contiguous().data_ptr(), B.contiguous().data_ptr(),
-            C.data_ptr(), M, N, K, K, N, N, 1 if accumulate else 0,
-            _ffi.dtype_code(A.dtype)), "gemm")
+        if A.is_complex():
+            # complex panels: the batched MFMA cgemm with batch=1
+            # (ref MatrixMult.py supports complex dtypes throughout)
+            _ffi.checked(_ffi.lib().pam_cgemm_batched(
+                _stream(A), A.contiguous().data_ptr(),
+                B.contiguous().data_ptr(), C.data_ptr(), 1, M, N, K,
+                0, 0, 0, 0, 1 if accumulate else 0,
+                _ffi.dtype_code(A.dtype)), "cgemm")
+        else:
+            _ffi.checked(_ffi.lib().pam_gemm(
+                _stream(A), A.contiguous().data_ptr(),
+                B.contiguous().data_ptr(), C.data_ptr(), M, N, K, K, N, N,
+                1 if accumulate else 0, _ffi.dtype_code(A.dtype)), "gemm")
         return C
 
     def _local_transpose(self, A: torch.Tensor) -> torch.Tensor:
+        """A^T for real dtypes, A^H for complex (ref :317-318,416,737
+        ``A.T.conj()`` — conj is a no-op on reals)."""
         if A.device.type != "cuda":
             raise RuntimeError(
                 "pam: compute ops require a CUDA (MI355X) device tensor — "
                 "there is no CPU compute path")
         At = torch.empty((A.shape[1], A.shape[0]), dtype=A.dtype,
                          device=A.device)
-        _ffi.checked(_ffi.lib().pam_transpose(
-            _stream(A), A.contiguous().data_ptr(), At.data_ptr(),
-            A.shape[0], A.shape[1], _ffi.dtype_code(A.dtype)), "transpose")
+        if A.is_complex():
+            _ffi.checked(_ffi.lib().pam_ctranspose(
+                _stream(A), A.contiguous().data_ptr(), At.data_ptr(),
+                A.shape[0], A.shape[1], 1, _ffi.dtype_code(A.dtype)),
+                "ctranspose")
+        else:
+            _ffi.checked(_ffi.lib().pam_transpose(
+                _stream(A), A.contiguous().data_ptr(), At.data_ptr(),
+                A.shape[0], A.shape[1], _ffi.dtype_code(A.dtype)),
+                "transpose")
         return At
 
     def _AH(self) -> torch.Tensor:

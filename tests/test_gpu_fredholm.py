@@ -84,7 +84,7 @@ def test_cgemm_batched(dtype, tol, B, M, K, N):
     Ad, Xd = dev(A).contiguous(), dev(X).contiguous()
     _ffi.checked(_ffi.lib().pam_cgemm_batched(
         s, Ad.data_ptr(), Xd.data_ptr(), Y.data_ptr(), B, M, N, K,
-        M * K, K * N, M * N, 0, _ffi.dtype_code(Ad.dtype)), "cgemm")
+        M * K, K * N, M * N, 0, 0, _ffi.dtype_code(Ad.dtype)), "cgemm")
     assert_allclose(host(Y), A @ X, rtol=tol, atol=tol)
     # conj-transpose op: Y2 = A^H @ X2, A stored [K=M_stored rows...]
     X2 = crand(rng, (B, M, N), dtype)
@@ -92,7 +92,7 @@ def test_cgemm_batched(dtype, tol, B, M, K, N):
     X2d = dev(X2).contiguous()
     _ffi.checked(_ffi.lib().pam_cgemm_batched(
         s, Ad.data_ptr(), X2d.data_ptr(), Y2.data_ptr(), B, K, N, M,
-        M * K, M * N, K * N, 1, _ffi.dtype_code(Ad.dtype)), "cgemm")
+        M * K, M * N, K * N, 1, 0, _ffi.dtype_code(Ad.dtype)), "cgemm")
     assert_allclose(host(Y2), A.conj().transpose(0, 2, 1) @ X2,
                     rtol=tol, atol=tol)
 

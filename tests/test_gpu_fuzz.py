@@ -95,7 +95,7 @@ def test_fuzz_cgemm_odd_shapes(seed):
     s = torch.cuda.current_stream().cuda_stream
     _ffi.checked(_ffi.lib().pam_cgemm_batched(
         s, g.data_ptr(), x.data_ptr(), y.data_ptr(), nb, M, N, K,
-        M * K, K * N, M * N, 0, _ffi.dtype_code(torch.complex64)),
+        M * K, K * N, M * N, 0, 0, _ffi.dtype_code(torch.complex64)),
         "cgemm")
     assert_allclose(host(y), np.matmul(G, X), rtol=1e-3,
                     atol=1e-3 * np.sqrt(K), err_msg=f"b{nb} {M}x{K}x{N}")
