@@ -77,7 +77,10 @@ class PamComm:
         group enter the call — ranks outside ``self`` (e.g. the inactive
         ranks of ``active_grid_comm``) need not participate."""
         if not self._use_dist:
-            return PamComm(0, 1, self.device, use_dist=False)
+            # preserve the global-rank identity of this process
+            return PamComm(0, 1, self.device, use_dist=False,
+                           ranks=[self.ranks[self.rank]],
+                           gloo_group=self._gloo_group)
         if len(colors) != self.size:
             raise ValueError(
                 f"colors must have one entry per group rank: "

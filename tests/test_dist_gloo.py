@@ -172,6 +172,7 @@ def body_blockdiag(c):
     "body_post_neighbors_overlap", "body_mask_subcomm", "body_nonstatconv",
     "body_proximal_call_reduction", "body_redistribute", "body_fftnd",
     "body_norm_axis", "body_broadcast_setitem",
+    "body_halo_guard_deterministic", "body_subcomm_split",
 ])
 def test_gloo_world2(body):
     _spawn(body)
@@ -448,3 +449,39 @@ def body_broadcast_setitem(c):
     u[:] = torch.full((6,), float(c.rank + 1), dtype=torch.float64)
     np.testing.assert_array_equal(u.local_array.numpy(),
                                   np.full(6, c.rank + 1.0))
+
+
+def body_halo_guard_deterministic(c):
+    """Undersized halo sources raise ValueError on EVERY rank (r01
+    advice: the old per-rank guard exempted edge ranks, which would post
+    a mismatched irecv and hang instead of raising)."""
+    import pylops_mpi_amd as pm
+    shapes = [(1, 4), (2, 4)]
+    d = pm.DistributedArray(
+        (3, 4), c, pm.Partition.SCATTER, 0,
+        local_array=torch.zeros(shapes[c.rank], dtype=torch.float64),
+        local_shapes=shapes, dtype=np.float64)
+    # rank 0 (an edge rank, and a sender for cells_front) holds only 1
+    # plane < 2 requested -> both ranks must raise, deterministically
+    with pytest.raises(ValueError, match="Local Shape at rank=0"):
+        d.add_ghost_cells(cells_front=2)
+    with pytest.raises(ValueError, match="Local Shape at rank=0"):
+        d.halo_exchange(2)
+
+
+def body_subcomm_split(c):
+    """split_by on a SUBcommunicator whose members are not the identity
+    prefix (r01 advice, medium): group-rank colors translate through
+    comm.ranks, and ranks outside the subcomm need not participate."""
+    sub = c.split_by([0, 1])     # rank 0 alone; rank 1 alone
+    assert sub.size == 1 and sub.ranks == [c.rank]
+    sub2 = c.split_by([5, 5])    # both in one group
+    assert sub2.size == 2 and sub2.ranks == [0, 1]
+    # nested split on the subcomm: rank 0 does NOT call it (the old
+    # world-collective new_group would deadlock here)
+    if c.rank == 1:
+        nested = sub.split_by([0])
+        assert nested.size == 1 and nested.ranks == [1]
+    t = torch.tensor([float(c.rank + 1)])
+    sub2.allreduce_(t, "sum")
+    assert t.item() == 3.0

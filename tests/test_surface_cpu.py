@@ -199,3 +199,17 @@ def test_reference_public_surface_complete():
     have = set(dir(pm)) | set(dir(prox)) | set(dir(po))
     missing = sorted(n for n in names if n not in have)
     assert missing == [], f"reference public names missing: {missing}"
+
+
+def test_mask_mismatch_raises_on_all_math():
+    # ref DistributedArray.py:581-585: every binary math op validates the
+    # mask (the reference's __sub__ = add(-other) path does too).  r01
+    # advice: sub/iaxpy_/xpby_ must validate like add/iadd.
+    a = pm.DistributedArray((4,), mask=[0])
+    b = pm.DistributedArray((4,), mask=[1])
+    a[:] = 1.0
+    b[:] = 2.0
+    for call in (lambda: a.add(b), lambda: a.sub(b),
+                 lambda: a.iaxpy_(1.0, b), lambda: a.xpby_(b, 2.0)):
+        with pytest.raises(ValueError, match="Mask"):
+            call()
