@@ -1,0 +1,413 @@
+"""Shared case table for the REFERENCE-executed golden fixtures.
+
+Two symmetric computations over the SAME deterministic inputs:
+
+  * :func:`compute_reference` — runs /root/reference/pylops_mpi in this
+    container (P ranks as threads, oracle/_refshim mpi4py + pylops
+    stubs) and collects the distributed outputs;
+  * :func:`compute_oracle` — runs the repo's rank-simulating oracle on
+    the identical cases.
+
+``tests/golden/generate_golden_ref.py`` saves the reference outputs to
+``golden_ref.npz`` (committed — the GPU box has no /root/reference);
+``tests/test_golden_ref.py`` asserts oracle == fixtures everywhere, and
+``tests/test_ref_parity.py`` asserts fixtures == a live reference run
+when /root/reference is present (fixture staleness guard).
+
+Inputs follow the reference's own test recipe (seed-42
+``normal(rank, 10, local_shape)`` per rank,
+ref tests/test_derivative.py:25,207).
+"""
+import os
+import sys
+
+import numpy as np
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_ROOT = os.path.dirname(os.path.dirname(_HERE))
+if _ROOT not in sys.path:
+    sys.path.insert(0, _ROOT)
+
+import oracle  # noqa: E402
+from oracle import matmult as om  # noqa: E402
+
+GOLDEN_PATH = os.path.join(_HERE, "golden_ref.npz")
+
+PS = (1, 2, 3, 4)
+DIMS = [(32,), (17, 5), (16, 4, 3)]
+SAMPLING = 1.5
+FD1_CASES = [("forward", 3, False), ("backward", 3, False),
+             ("centered", 3, False), ("centered", 3, True),
+             ("centered", 5, False), ("centered", 5, True)]
+FD2_CASES = [("forward", False), ("backward", False),
+             ("centered", False), ("centered", True)]
+MATH_N = (13, 4)
+NORM_ORDS = [("2", None), ("0", 0), ("inf", np.inf), ("ninf", -np.inf),
+             ("1p5", 1.5)]
+CGLS_DIMS = (17, 5)
+CGLS_NITER = 10
+CGLS_DAMPS = [("d0", 0.0), ("d05", 0.5)]
+MM_SHAPES = (7, 5, 9)       # N, K, M (uneven vs any P in PS)
+MM_DTYPES = ["float64", "complex128"]
+FRED_SHAPE = (21, 4, 6, 5)  # nsl, nx, ny, nz (the reference test's)
+FRED_DTYPES = ["float64", "complex128"]
+
+
+def tagd(dims):
+    return "x".join(map(str, dims))
+
+
+def make_global_x(n, P, seed_shift=0):
+    """seed-42 normal(rank,10) per rank, concatenated (ref test recipe)."""
+    parts = []
+    for r in range(P):
+        np.random.seed(42 + seed_shift)
+        parts.append(np.random.normal(
+            r, 10, oracle.local_split((n,), P, r)))
+    return np.concatenate(parts)
+
+
+def blockdiag_mats(P):
+    """Per-rank dense serial blocks (examples/plot_cgls.py:30-33 style)."""
+    rng = np.random.default_rng(31)
+    shapes = [[(4, 6)], [(3, 2), (5, 5)], [(2, 3)], [(4, 4), (1, 2)]]
+    return [[rng.standard_normal(s) for s in shapes[r]] for r in range(P)]
+
+
+def fred_G(dtype):
+    nsl, nx, ny, _ = FRED_SHAPE
+    G = (np.outer(np.ones(nsl), np.arange(nx * ny))
+         .reshape(nsl, nx, ny).astype(dtype))
+    if np.dtype(dtype).kind == "c":
+        G = G + 1j * (G[:, ::-1, ::-1] / 7.0)
+    return G
+
+
+def fred_x(dtype, forward):
+    nsl, nx, ny, nz = FRED_SHAPE
+    rng = np.random.default_rng(17)
+    shape = (nsl, ny, nz) if forward else (nsl, nx, nz)
+    x = rng.standard_normal(shape).astype(dtype)
+    if np.dtype(dtype).kind == "c":
+        x = x + 1j * rng.standard_normal(shape).astype(dtype)
+    return x
+
+
+def fred_split(P):
+    nsl = FRED_SHAPE[0]
+    return [oracle.local_split((nsl,), P, r)[0] for r in range(P)]
+
+
+def plane_counts(dims, P):
+    return [int(np.prod(oracle.local_split(dims, P, r)))
+            for r in range(P)]
+
+
+def _oracle_dist_planes(xg, dims, P):
+    from oracle.ranksim import Partition, SimArray
+    counts = plane_counts(dims, P)
+    offs = np.cumsum([0] + counts)
+    return SimArray([xg[offs[r]: offs[r + 1]] for r in range(P)],
+                    (int(xg.size),), 0, Partition.SCATTER)
+
+
+# ===================================================== oracle side
+def compute_oracle():
+    out = {}
+    for P in PS:
+        # FD operators
+        for dims in DIMS:
+            n = int(np.prod(dims))
+            xg = make_global_x(n, P)
+            yg = make_global_x(n, P, seed_shift=1)
+            for kind, order, edge in FD1_CASES:
+                key = f"fd1_P{P}_{kind}{order}{'e' if edge else 'n'}_" \
+                      f"{tagd(dims)}"
+                op = oracle.SimFirstDerivative(dims, SAMPLING, kind, edge,
+                                               order)
+                out[key + "_mv"] = op.matvec(oracle.to_dist(xg, P)).asarray()
+                out[key + "_rmv"] = op.rmatvec(oracle.to_dist(yg, P)).asarray()
+            for kind, edge in FD2_CASES:
+                key = f"fd2_P{P}_{kind}{'e' if edge else 'n'}_{tagd(dims)}"
+                op = oracle.SimSecondDerivative(dims, SAMPLING, kind, edge)
+                out[key + "_mv"] = op.matvec(oracle.to_dist(xg, P)).asarray()
+                out[key + "_rmv"] = op.rmatvec(oracle.to_dist(yg, P)).asarray()
+        # array math
+        n = int(np.prod(MATH_N))
+        xg = make_global_x(n, P)
+        yg = make_global_x(n, P, seed_shift=1)
+        dx = oracle.to_dist(xg, P)
+        dy = oracle.to_dist(yg, P)
+        out[f"math_P{P}_dot"] = np.asarray(dx.dot(dy))
+        dxa = oracle.to_dist(np.abs(xg), P)
+        for name, o in NORM_ORDS:
+            # |x| input for the fractional norm: the reference's
+            # float_power(negative, 1.5) is NaN (ref :786) — pin values
+            d = dxa if name == "1p5" else dx
+            out[f"math_P{P}_norm{name}"] = np.asarray(d.norm(o))
+        out[f"math_P{P}_addmul"] = ((dx + dy) * dx).asarray()
+        d2 = oracle.to_dist(xg.reshape(MATH_N), P)
+        for w in (1, 2):
+            gh = d2.add_ghost_cells(cells_front=w, cells_back=w)
+            out[f"math_P{P}_ghost{w}"] = np.concatenate(
+                [g.ravel() for g in gh])
+        # CGLS on FD1 centered3.  x0 must be plane-aligned: the
+        # reference's reshaped wrapper leaves operator OUTPUTS on the
+        # plane split (ref decorators.py:79-82), and CGLS subtracts
+        # damped_x (x0's split) from rmatvec outputs (ref
+        # cls_basic.py:348) — a default-split x0 raises "Local Array
+        # Shape Mismatch" in the reference itself when dims[0] % P != 0.
+        nc = int(np.prod(CGLS_DIMS))
+        xg = make_global_x(nc, P)
+        op = oracle.SimFirstDerivative(CGLS_DIMS, SAMPLING, "centered",
+                                       False, 3)
+        y = op.matvec(oracle.to_dist(xg, P))
+        x0 = _oracle_dist_planes(np.zeros(nc), CGLS_DIMS, P)
+        for dn, damp in CGLS_DAMPS:
+            xs, cost = oracle.sim_cgls(op, y, x0, CGLS_NITER, damp=damp,
+                                       tol=0.0)
+            out[f"cgls_P{P}_{dn}_x"] = xs.asarray()
+            out[f"cgls_P{P}_{dn}_cost"] = np.asarray(cost)
+        # BlockDiag
+        mats = blockdiag_mats(P)
+        bop = oracle.SimBlockDiag(mats)
+        nr, nc2 = bop.shape
+        xg = make_global_x(nc2, P)
+        yg = make_global_x(nr, P, seed_shift=1)
+        out[f"bd_P{P}_mv"] = bop.matvec(oracle.to_dist(xg, P)).asarray()
+        out[f"bd_P{P}_rmv"] = bop.rmatvec(oracle.to_dist(yg, P)).asarray()
+        # Fredholm1
+        for dt in FRED_DTYPES:
+            G = fred_G(dt)
+            nsls = fred_split(P)
+            blocks, off = [], 0
+            for r in range(P):
+                blocks.append(G[off: off + nsls[r]])
+                off += nsls[r]
+            for sg in (False, True):
+                sop = oracle.SimFredholm1(blocks, nz=FRED_SHAPE[3],
+                                          saveGt=sg)
+                key = f"fred_P{P}_{np.dtype(dt).char}_{'s' if sg else 'n'}"
+                xb = fred_x(dt, True).ravel()
+                yb = fred_x(dt, False).ravel()
+                from oracle.ranksim import Partition, SimArray
+                xd = SimArray([xb.copy() for _ in range(P)], xb.shape,
+                              partition=Partition.BROADCAST)
+                yd = SimArray([yb.copy() for _ in range(P)], yb.shape,
+                              partition=Partition.BROADCAST)
+                out[key + "_mv"] = sop.matvec(xd).asarray()
+                out[key + "_rmv"] = sop.rmatvec(yd).asarray()
+    # MatrixMult (square P only, like the reference)
+    N, K, M = MM_SHAPES
+    for P in (1, 4):
+        rng = np.random.default_rng(77)
+        for dt in MM_DTYPES:
+            A = rng.standard_normal((N, K)).astype(dt)
+            X = rng.standard_normal((K, M)).astype(dt)
+            Y = rng.standard_normal((N, M)).astype(dt)
+            if np.dtype(dt).kind == "c":
+                A = A + 1j * rng.standard_normal((N, K))
+                X = X + 1j * rng.standard_normal((K, M))
+                Y = Y + 1j * rng.standard_normal((N, M))
+            for kind in ("block", "summa"):
+                key = f"mm_{kind}_P{P}_{np.dtype(dt).char}"
+                if kind == "block":
+                    mv = om.block_expected_mv(A, X, P)
+                    rmv = om.block_expected_rmv(A, Y, P)
+                else:
+                    mv = om.summa_expected_mv(A, X, P)
+                    rmv = om.summa_expected_rmv(A, Y, P)
+                out[key + "_mv"] = np.concatenate(
+                    [v.ravel() for v in mv])
+                out[key + "_rmv"] = np.concatenate(
+                    [v.ravel() for v in rmv])
+    return out
+
+
+# ================================================== reference side
+def compute_reference():
+    from oracle.refrun import run_reference
+
+    out = {}
+    for P in PS:
+        outs = run_reference(P, _ref_rank_fn(P))
+        for key, val in outs[0].items():
+            if key.startswith("__perrank__"):
+                # rank-resolved pieces (ghost cells) come back per rank
+                real = key[len("__perrank__"):]
+                out[real] = np.concatenate(
+                    [np.asarray(o[key]).ravel() for o in outs])
+            else:
+                out[key] = np.asarray(val)
+    for P in (1, 4):
+        outs = run_reference(P, _ref_mm_fn(P))
+        for key, val in outs[0].items():
+            out[key] = np.asarray(val)
+    return out
+
+
+def _ref_rank_fn(P):
+    # precompute EVERY input in the spawning thread: make_global_x uses
+    # the global numpy RNG (the reference test recipe's np.random.seed),
+    # which is not thread-safe across the P rank threads
+    fd_inputs = {}
+    for dims in DIMS:
+        n = int(np.prod(dims))
+        fd_inputs[dims] = (make_global_x(n, P),
+                           make_global_x(n, P, seed_shift=1))
+    nmath = int(np.prod(MATH_N))
+    math_x = make_global_x(nmath, P)
+    math_y = make_global_x(nmath, P, seed_shift=1)
+    ncgls = int(np.prod(CGLS_DIMS))
+    cgls_x = make_global_x(ncgls, P)
+    bd_mats = blockdiag_mats(P)
+    import oracle as _oracle_pkg
+    nr_bd = int(sum(sum(A.shape[0] for A in ms) for ms in bd_mats))
+    nc_bd = int(sum(sum(A.shape[1] for A in ms) for ms in bd_mats))
+    bd_x = make_global_x(nc_bd, P)
+    bd_y = make_global_x(nr_bd, P, seed_shift=1)
+
+    def fn(rank):
+        from pylops_mpi import (DistributedArray, MPIBlockDiag,
+                                MPIFirstDerivative, MPISecondDerivative,
+                                Partition, cgls)
+        from pylops_mpi.signalprocessing import MPIFredholm1
+        import pylops
+
+        def dist_from_global(xg, dims=None):
+            n = int(xg.size)
+            d = DistributedArray(global_shape=n, dtype=xg.dtype)
+            counts = [oracle.local_split((n,), P, r)[0] for r in range(P)]
+            off = int(np.sum(counts[:rank], initial=0))
+            d[:] = xg[off: off + d.local_shape[0]]
+            return d
+
+        res = {}
+        for dims in DIMS:
+            n = int(np.prod(dims))
+            xg, yg = fd_inputs[dims]
+            for kind, order, edge in FD1_CASES:
+                key = f"fd1_P{P}_{kind}{order}{'e' if edge else 'n'}_" \
+                      f"{tagd(dims)}"
+                op = MPIFirstDerivative(dims, sampling=SAMPLING, kind=kind,
+                                        edge=edge, order=order)
+                res[key + "_mv"] = op.matvec(dist_from_global(xg)).asarray()
+                res[key + "_rmv"] = op.rmatvec(
+                    dist_from_global(yg)).asarray()
+            for kind, edge in FD2_CASES:
+                key = f"fd2_P{P}_{kind}{'e' if edge else 'n'}_{tagd(dims)}"
+                op = MPISecondDerivative(dims, sampling=SAMPLING, kind=kind,
+                                         edge=edge)
+                res[key + "_mv"] = op.matvec(dist_from_global(xg)).asarray()
+                res[key + "_rmv"] = op.rmatvec(
+                    dist_from_global(yg)).asarray()
+        n = nmath
+        xg, yg = math_x, math_y
+        dx, dy = dist_from_global(xg), dist_from_global(yg)
+        res[f"math_P{P}_dot"] = np.asarray(dx.dot(dy))
+        dxa = dist_from_global(np.abs(xg))
+        for name, o in NORM_ORDS:
+            d = dxa if name == "1p5" else dx
+            res[f"math_P{P}_norm{name}"] = np.asarray(
+                d.norm() if o is None else d.norm(o))
+        res[f"math_P{P}_addmul"] = ((dx + dy) * dx).asarray()
+        d2 = DistributedArray(global_shape=MATH_N, dtype=np.float64)
+        counts = [oracle.local_split(MATH_N, P, r)[0] for r in range(P)]
+        off = int(np.sum(counts[:rank], initial=0))
+        d2[:] = xg.reshape(MATH_N)[off: off + d2.local_shape[0]]
+        for w in (1, 2):
+            res[f"__perrank__math_P{P}_ghost{w}"] = d2.add_ghost_cells(
+                cells_front=w, cells_back=w)
+        # CGLS
+        nc = ncgls
+        xg = cgls_x
+        op = MPIFirstDerivative(CGLS_DIMS, sampling=SAMPLING,
+                                kind="centered", edge=False, order=3)
+        y = op.matvec(dist_from_global(xg))
+        pcounts = plane_counts(CGLS_DIMS, P)
+        for dn, damp in CGLS_DAMPS:
+            # plane-aligned x0 (see the oracle-side comment)
+            x0 = DistributedArray(
+                global_shape=nc, local_shapes=[(c,) for c in pcounts],
+                dtype=np.float64)
+            x0[:] = 0.0
+            xs, istop, iit, r1, r2, cost = cgls(
+                op, y, x0, niter=CGLS_NITER, damp=damp, tol=0.0,
+                show=False)
+            res[f"cgls_P{P}_{dn}_x"] = xs.asarray()
+            res[f"cgls_P{P}_{dn}_cost"] = np.asarray(cost)
+        # BlockDiag with serial dense blocks (pylops stub MatrixMult)
+        ops = [pylops.MatrixMult(A) for A in bd_mats[rank]]
+        bop = MPIBlockDiag(ops=ops)
+        assert bop.shape == (nr_bd, nc_bd)
+        xg, yg = bd_x, bd_y
+        res[f"bd_P{P}_mv"] = bop.matvec(dist_from_global(xg)).asarray()
+        res[f"bd_P{P}_rmv"] = bop.rmatvec(dist_from_global(yg)).asarray()
+        # Fredholm1
+        for dt in FRED_DTYPES:
+            G = fred_G(dt)
+            nsls = fred_split(P)
+            off = int(np.sum(nsls[:rank], initial=0))
+            Gl = G[off: off + nsls[rank]]
+            for sg in (False, True):
+                fop = MPIFredholm1(Gl, nz=FRED_SHAPE[3], saveGt=sg,
+                                   usematmul=True, dtype=dt)
+                key = f"fred_P{P}_{np.dtype(dt).char}_{'s' if sg else 'n'}"
+                xb = fred_x(dt, True).ravel()
+                yb = fred_x(dt, False).ravel()
+                xd = DistributedArray(global_shape=xb.size,
+                                      partition=Partition.BROADCAST,
+                                      dtype=dt)
+                xd[:] = xb
+                yd = DistributedArray(global_shape=yb.size,
+                                      partition=Partition.BROADCAST,
+                                      dtype=dt)
+                yd[:] = yb
+                res[key + "_mv"] = fop.matvec(xd).asarray()
+                res[key + "_rmv"] = fop.rmatvec(yd).asarray()
+        return res
+    return fn
+
+
+def _ref_mm_fn(P):
+    def fn(rank):
+        from pylops_mpi import DistributedArray, Partition
+        from pylops_mpi.basicoperators.MatrixMult import MPIMatrixMult
+
+        N, K, M = MM_SHAPES
+        rng = np.random.default_rng(77)
+        res = {}
+        for dt in MM_DTYPES:
+            A = rng.standard_normal((N, K)).astype(dt)
+            X = rng.standard_normal((K, M)).astype(dt)
+            Y = rng.standard_normal((N, M)).astype(dt)
+            if np.dtype(dt).kind == "c":
+                A = A + 1j * rng.standard_normal((N, K))
+                X = X + 1j * rng.standard_normal((K, M))
+                Y = Y + 1j * rng.standard_normal((N, M))
+            for kind in ("block", "summa"):
+                key = f"mm_{kind}_P{P}_{np.dtype(dt).char}"
+                if kind == "block":
+                    pp = om._isqrt(P)
+                    inputs = om.block_inputs(A, X, P)
+                    ylocals = [Y[:, om.split_slice(M, pp, q // pp)].ravel()
+                               for q in range(P)]
+                else:
+                    inputs = om.summa_inputs(A, X, P)
+                    ylocals = [om.summa_tile(Y, P, q).ravel()
+                               for q in range(P)]
+                op = MPIMatrixMult(inputs[rank][0], M, kind=kind, dtype=dt)
+                counts = [v.size for _, v in inputs]
+                xd = DistributedArray(
+                    global_shape=int(sum(counts)),
+                    local_shapes=[(int(v),) for v in counts], dtype=dt)
+                xd[:] = inputs[rank][1].ravel()
+                res[key + "_mv"] = op.matvec(xd).asarray()
+                ycounts = [v.size for v in ylocals]
+                yd = DistributedArray(
+                    global_shape=int(sum(ycounts)),
+                    local_shapes=[(int(v),) for v in ycounts], dtype=dt)
+                yd[:] = ylocals[rank]
+                res[key + "_rmv"] = op.rmatvec(yd).asarray()
+        return res
+    return fn
