@@ -113,6 +113,18 @@ class PamComm:
                             group=self._group)
         return t
 
+    def reduce_(self, t: torch.Tensor, root: int = 0,
+                op: str = "sum") -> torch.Tensor:
+        """Reduce to group rank ``root`` (the rectangular-SUMMA adjoint's
+        panel-sum; valid on t only at the root afterwards)."""
+        if self._use_dist:
+            rt = torch.view_as_real(t) if (t.is_complex() and op == "sum") \
+                else t
+            dist.reduce(rt, dst=self.ranks[root],
+                        op=getattr(dist.ReduceOp, _REDUCE_OPS[op]),
+                        group=self._group)
+        return t
+
     def broadcast_(self, t: torch.Tensor, root: int = 0) -> torch.Tensor:
         """Broadcast from group rank ``root`` (ref Distributed.py:195-225)."""
         if self._use_dist:

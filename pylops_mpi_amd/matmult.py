@@ -23,9 +23,20 @@ MI355X-first differences:
     preserved by construction;
   * metadata (shape gathers) travels on the gloo control plane of the
     WORLD communicator (all memberships are deterministic).
-Like the reference (ref :295-298,564-566), both kinds require a square
-number of ranks; use active_grid_comm to carve an active square grid out
-of a non-square world (e.g. 2x2 out of 8).
+Like the reference (ref :295-298,564-566), both kinds default to a
+square number of ranks, and active_grid_comm carves an active square
+grid out of a non-square world (e.g. 2x2 out of 8).  BEYOND the
+reference, the summa kind also accepts an explicit rectangular process
+grid (``grid=(Pr, Pc)`` with Pr*Pc == comm size): BASELINE cfg4 names a
+2x4 grid on 8 GPUs, which the reference cannot run (its SUMMA is
+square-only, so active_grid_comm would idle 4 of 8 GPUs).  The
+rectangular path is classic panel SUMMA: K is padded to a multiple of
+L = lcm(Pr, Pc) and walked in L panels; each step row-broadcasts an A
+panel, col-broadcasts an X panel and accumulates on the MFMA panel
+GEMM; the adjoint reuses the same A-panel broadcast and turns the
+tag-routed A^T exchange into a col-comm reduce-to-owner (RCCL-friendly
+— no tags needed).  Results match dense A@X / A^H y exactly like the
+square path (world-8 gloo tests, tests/test_dist_gloo_rect.py).
 """
 import math
 from typing import Optional, Tuple
@@ -103,6 +114,32 @@ def block_gather(x: DistributedArray, orig_shape: Tuple[int, int],
             C[rs, cs] = blks[rank].reshape(rs.stop - rs.start,
                                            cs.stop - cs.start)
     return C
+
+
+def summa_grid_splits(N: int, K: int, M: int, Pr: int, Pc: int):
+    """Per-rank (row_slice, col_slice) for A [N,K] and X [K,M] on a
+    rectangular Pr x Pc grid (rank = row*Pc + col).
+
+    The K direction is padded to a multiple of L = lcm(Pr, Pc) so that
+    both A's column split (by Pc) and X's row split (by Pr) land on
+    panel boundaries — the rectangular analogue of the reference's
+    square padding (ref :589-601)."""
+    L = math.lcm(Pr, Pc)
+    K_pad = math.ceil(K / L) * L
+    bn = math.ceil(N / Pr)
+    bkA = K_pad // Pc
+    bkX = K_pad // Pr
+    bm = math.ceil(M / Pc)
+    a_slices, x_slices, y_slices = [], [], []
+    for q in range(Pr * Pc):
+        qr, qc = divmod(q, Pc)
+        a_slices.append((slice(qr * bn, min((qr + 1) * bn, N)),
+                         slice(qc * bkA, min((qc + 1) * bkA, K))))
+        x_slices.append((slice(qr * bkX, min((qr + 1) * bkX, K)),
+                         slice(qc * bm, min((qc + 1) * bm, M))))
+        y_slices.append((slice(qr * bn, min((qr + 1) * bn, N)),
+                         slice(qc * bm, min((qc + 1) * bm, M))))
+    return a_slices, x_slices, y_slices
 
 
 class _MatMultBase(MPILinearOperator):
@@ -248,13 +285,22 @@ class _MPIBlockMatrixMult(_MatMultBase):
 
 
 class _MPISummaMatrixMult(_MatMultBase):
-    """2-D SUMMA (ref :430-765)."""
+    """2-D SUMMA (ref :430-765); ``grid=(Pr, Pc)`` extends it to
+    rectangular process grids (BASELINE cfg4's 2x4 on 8 ranks — beyond
+    the reference's square-only restriction, ref :564-566)."""
 
     def __init__(self, A: torch.Tensor, M: int, saveAt: bool = False,
-                 base_comm: Optional[PamComm] = None, dtype="float64"):
+                 base_comm: Optional[PamComm] = None, dtype="float64",
+                 grid: Optional[Tuple[int, int]] = None):
         comm = base_comm if base_comm is not None else get_default_comm()
         rank, size = comm.rank, comm.size
         self.base_comm_grid = comm
+        if grid is not None and tuple(grid)[0] != tuple(grid)[1]:
+            self._init_rect(A, M, comm, dtype, tuple(grid))
+            return
+        if grid is not None and grid[0] * grid[1] != size:
+            raise ValueError(f"grid {grid} does not tile {size} ranks")
+        self._rect = False
         P = self._make_grid(comm)
         self._row_id, self._col_id = divmod(rank, P)
         self._row_comm = comm.split_by([r // P for r in range(size)],
@@ -285,6 +331,156 @@ class _MPISummaMatrixMult(_MatMultBase):
             self.At = self._local_transpose(self.A)
         super().__init__(dims=(self.K, self.M), dimsd=(self.N, self.M),
                          dtype=np.dtype(dtype), base_comm=comm)
+
+    # ------------------------------------------------ rectangular grid
+    def _init_rect(self, A: torch.Tensor, M: int, comm: PamComm, dtype,
+                   grid: Tuple[int, int]):
+        Pr, Pc = grid
+        if Pr * Pc != comm.size:
+            raise ValueError(f"grid {grid} does not tile {comm.size} ranks")
+        self._rect = True
+        self._grid = (Pr, Pc)
+        rank, size = comm.rank, comm.size
+        self._row_id, self._col_id = divmod(rank, Pc)
+        self._row_comm = comm.split_by([r // Pc for r in range(size)],
+                                       [r % Pc for r in range(size)])
+        self._col_comm = comm.split_by([r % Pc for r in range(size)],
+                                       [r // Pc for r in range(size)])
+        self.A = A.to(as_torch_dtype(np.dtype(dtype)))
+        all_rows = comm.allgather_obj(int(A.shape[0]))
+        all_cols = comm.allgather_obj(int(A.shape[1]))
+        col_members = [r for r in range(size) if r % Pc == self._col_id]
+        row_members = [r for r in range(size) if r // Pc == self._row_id]
+        self.N = int(sum(all_rows[r] for r in col_members))
+        self.K = int(sum(all_cols[r] for r in row_members))
+        self.M = M
+        L = math.lcm(Pr, Pc)
+        self._L = L
+        self._N_padded = math.ceil(self.N / Pr) * Pr
+        self._K_padded = math.ceil(self.K / L) * L
+        self._M_padded = math.ceil(self.M / Pc) * Pc
+        bn = self._N_padded // Pr
+        bkA = self._K_padded // Pc
+        # remainder-only-on-last-block rule (inherited from the
+        # reference's square padding, ref :589-601)
+        if self.N < (Pr - 1) * bn or self.K < (Pc - 1) * bkA \
+                or self.K < (Pr - 1) * (self._K_padded // Pr) \
+                or self.M < (Pc - 1) * (self._M_padded // Pc):
+            raise ValueError(
+                f"shape ({self.N},{self.K},{self.M}) too small for grid "
+                f"{grid}: non-final blocks would be ragged")
+        pr_ = (bn - int(A.shape[0])) if self._row_id == Pr - 1 else 0
+        pc_ = (bkA - int(A.shape[1])) if self._col_id == Pc - 1 else 0
+        if pr_ > 0 or pc_ > 0:
+            Ap = torch.zeros((int(A.shape[0]) + pr_, int(A.shape[1]) + pc_),
+                             dtype=self.A.dtype, device=self.A.device)
+            Ap[: A.shape[0], : A.shape[1]] = self.A
+            self.A = Ap
+        # saveAt is a no-op on the rect path: the adjoint transposes the
+        # BROADCAST-RECEIVED panels, not this rank's own A
+        MPILinearOperator.__init__(
+            self, dims=(self.K, self.M), dimsd=(self.N, self.M),
+            dtype=np.dtype(dtype), base_comm=comm)
+
+    def _rect_sizes(self):
+        Pr, Pc = self._grid
+        bn = self._N_padded // Pr
+        bkX = self._K_padded // Pr
+        bm = self._M_padded // Pc
+        local_n = bn if self._row_id != Pr - 1 else self.N - (Pr - 1) * bn
+        local_k = max(0, bkX if self._row_id != Pr - 1
+                      else self.K - (Pr - 1) * bkX)
+        local_m = bm if self._col_id != Pc - 1 else self.M - (Pc - 1) * bm
+        return bn, bkX, bm, local_n, local_k, local_m
+
+    def _rect_counts(self, kind: str):
+        Pr, Pc = self._grid
+        bn = self._N_padded // Pr
+        bkX = self._K_padded // Pr
+        bm = self._M_padded // Pc
+        out = []
+        for q in range(Pr * Pc):
+            qr, qc = divmod(q, Pc)
+            ln = bn if qr != Pr - 1 else self.N - (Pr - 1) * bn
+            lk = max(0, bkX if qr != Pr - 1 else self.K - (Pr - 1) * bkX)
+            lm = bm if qc != Pc - 1 else self.M - (Pc - 1) * bm
+            out.append(int((ln if kind == "n" else lk) * lm))
+        return out
+
+    def _rect_panels(self, t: int):
+        """Step t owners/offsets: A panel lives at grid col cA (offset
+        pA within its block), X panel at grid row rX (offset pX)."""
+        Pr, Pc = self._grid
+        L = self._L
+        cA, pA = divmod(t, L // Pc)
+        rX, pX = divmod(t, L // Pr)
+        return cA, pA, rX, pX
+
+    def _matvec_rect(self, x: DistributedArray) -> DistributedArray:
+        self._check_scatter(x)
+        Pr, Pc = self._grid
+        L = self._L
+        bkp = self._K_padded // L
+        bn, bkX, bm, local_n, local_k, local_m = self._rect_sizes()
+        y = DistributedArray(
+            int(self.N * self.M), x.base_comm, Partition.SCATTER, 0,
+            local_shapes=[(c,) for c in self._rect_counts("n")],
+            dtype=self.dtype)
+        x_block = self._pad_block(
+            x.local_array.reshape(local_k, local_m).to(self.A.dtype),
+            bkX, bm)
+        Y_local = torch.zeros((bn, bm), dtype=self.A.dtype,
+                              device=self.A.device)
+        for t in range(L):
+            cA, pA, rX, pX = self._rect_panels(t)
+            if self._col_id == cA:
+                Aslice = self.A[:, pA * bkp: (pA + 1) * bkp].contiguous()
+            else:
+                Aslice = torch.empty((bn, bkp), dtype=self.A.dtype,
+                                     device=self.A.device)
+            self._row_comm.broadcast_(Aslice, root=cA)
+            if self._row_id == rX:
+                Xslice = x_block[pX * bkp: (pX + 1) * bkp, :].contiguous()
+            else:
+                Xslice = torch.empty((bkp, bm), dtype=self.A.dtype,
+                                     device=self.A.device)
+            self._col_comm.broadcast_(Xslice, root=rX)
+            self._local_gemm(Aslice, Xslice, Y_local, accumulate=True)
+        y[:] = Y_local[:local_n, :local_m].reshape(-1)
+        return y
+
+    def _rmatvec_rect(self, x: DistributedArray) -> DistributedArray:
+        self._check_scatter(x)
+        Pr, Pc = self._grid
+        L = self._L
+        bkp = self._K_padded // L
+        bn, bkX, bm, local_n, local_k, local_m = self._rect_sizes()
+        y = DistributedArray(
+            int(self.K * self.M), x.base_comm, Partition.SCATTER, 0,
+            local_shapes=[(c,) for c in self._rect_counts("k")],
+            dtype=self.dtype)
+        x_block = self._pad_block(
+            x.local_array.reshape(local_n, local_m).to(self.A.dtype),
+            bn, bm)
+        Z = torch.zeros((bkX, bm), dtype=self.A.dtype,
+                        device=self.A.device)
+        for t in range(L):
+            cA, pA, rX, pX = self._rect_panels(t)
+            if self._col_id == cA:
+                Aslice = self.A[:, pA * bkp: (pA + 1) * bkp].contiguous()
+            else:
+                Aslice = torch.empty((bn, bkp), dtype=self.A.dtype,
+                                     device=self.A.device)
+            self._row_comm.broadcast_(Aslice, root=cA)
+            # G = Aslice^H @ x_block, summed over the grid column to the
+            # X-panel owner (replaces the square path's tag-routed A^T
+            # exchange, ref :745-760 — RCCL has no tags)
+            G = self._local_gemm(self._local_transpose(Aslice), x_block)
+            self._col_comm.reduce_(G, root=rX)
+            if self._row_id == rX:
+                Z[pX * bkp: (pX + 1) * bkp, :] = G
+        y[:] = Z[:local_k, :local_m].reshape(-1)
+        return y
 
     def _tile_sizes(self):
         P = self._P_prime
@@ -321,6 +517,8 @@ class _MPISummaMatrixMult(_MatMultBase):
 
     def _matvec(self, x: DistributedArray) -> DistributedArray:
         # ref :610-672
+        if self._rect:
+            return self._matvec_rect(x)
         self._check_scatter(x)
         P = self._P_prime
         bn, bk, bm, local_n, local_k, local_m = self._tile_sizes()
@@ -345,6 +543,8 @@ class _MPISummaMatrixMult(_MatMultBase):
 
     def _rmatvec(self, x: DistributedArray) -> DistributedArray:
         # ref :674-765
+        if self._rect:
+            return self._rmatvec_rect(x)
         self._check_scatter(x)
         P = self._P_prime
         bn, bk, bm, local_n, local_k, local_m = self._tile_sizes()
@@ -386,10 +586,17 @@ class _MPISummaMatrixMult(_MatMultBase):
 
 def MPIMatrixMult(A: torch.Tensor, M: int, saveAt: bool = False,
                   base_comm: Optional[PamComm] = None,
-                  kind: str = "summa", dtype="float64"):
-    """Factory, ref :768-872."""
+                  kind: str = "summa", dtype="float64",
+                  grid: Optional[Tuple[int, int]] = None):
+    """Factory, ref :768-872.  ``grid=(Pr, Pc)`` (summa only) selects a
+    rectangular process grid — the cfg4 2x4-on-8-GPUs layout the
+    reference's square-only SUMMA cannot express."""
     if kind == "summa":
-        return _MPISummaMatrixMult(A, M, saveAt, base_comm, dtype)
+        return _MPISummaMatrixMult(A, M, saveAt, base_comm, dtype,
+                                   grid=grid)
     elif kind == "block":
+        if grid is not None:
+            raise NotImplementedError(
+                "rectangular grids are a summa-kind extension")
         return _MPIBlockMatrixMult(A, M, saveAt, base_comm, dtype)
     raise NotImplementedError("kind must be summa or block")
