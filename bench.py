@@ -36,11 +36,56 @@ HBM_PEAK_GBS = 8000.0  # gfx950 spec peak (MI355X_MICROARCH.md chip table)
 BYTES_PER_PT = 16.0    # centered3 fp64: 8 read + 8 write, algorithmic
 
 
-def workload_dims(n_gpus: int):
+def workload_dims(n_gpus: int, smoke: bool = False):
+    if smoke:
+        # --smoke-gloo: same structure, CPU-sized (orchestration test
+        # only — never a measured configuration)
+        return (64 * n_gpus, 32, 16), "smoke-gloo (not a benchmark)"
     if n_gpus == 1:
         return (2048, 2048, 128), "MPIFirstDerivative cgls config (BASELINE configs[1])"
     return (512 * n_gpus, 4096, 256), \
         f"north-star weak scaling ({n_gpus}/8 of BASELINE 8-GPU config)"
+
+
+def _smoke_fd_pair(op, x, comm):
+    """CPU stand-in for one matvec+rmatvec pair, used ONLY under
+    --smoke-gloo to exercise bench.py's full orchestration (rank env,
+    comms, plane-aligned layout, halo exchange, reductions, JSON) with
+    no GPU.  Torch slice restatement of the centered3 formulas
+    (ref FirstDerivative.py:201-246); the real bench path never runs
+    it — the product kernels stay fail-loud GPU-only."""
+    shapes, counts = op._plane_counts()
+    r, P = comm.rank, comm.size
+    nloc = shapes[r][0]
+    m = int(np.prod(op.dims[1:], initial=1))
+    row0 = int(np.sum([s[0] for s in shapes[:r]], initial=0))
+    N0 = op.dims[0]
+
+    def pair(planes):
+        gf, gb = None, None
+        if P > 1:
+            gf = torch.empty((1, m), dtype=planes.dtype)
+            gb = torch.empty((1, m), dtype=planes.dtype)
+            comm.sendrecv_neighbors(
+                planes[:1].contiguous() if r > 0 else None,
+                planes[-1:].contiguous() if r < P - 1 else None,
+                gf if r > 0 else None, gb if r < P - 1 else None)
+        top = gf if r > 0 else torch.zeros((1, m), dtype=planes.dtype)
+        bot = gb if r < P - 1 else torch.zeros((1, m), dtype=planes.dtype)
+        g = torch.cat([top, planes, bot])
+        y = 0.5 * (g[2:] - g[:-2])
+        if row0 == 0:
+            y[0] = 0.0
+        if row0 + nloc == N0:
+            y[-1] = 0.0
+        return y
+
+    planes = x.local_array.view(nloc, m)
+    y = pair(planes)
+    z = pair(y)   # adjoint of centered3 = -centered3 up to edges; the
+    # smoke only exercises data movement, not numerics (parity tests
+    # pin numerics)
+    return z
 
 
 def read_traffic_calibration(dims, n_gpus):
@@ -142,12 +187,46 @@ def cpu_baseline(dims):
     }
 
 
+def _smoke_main(args, op, x, comm, n_gpus, dims, wname, rank):
+    """--smoke-gloo body: the same step loop / barrier+sync protocol /
+    max-over-ranks reduction / JSON emission as the real bench, sized
+    for CPU (`bench.py --gpus 8` must run under
+    gloo end-to-end before any 8-GPU lease exists)."""
+    for _ in range(args.warmup):
+        _smoke_fd_pair(op, x, comm)
+    comm.barrier()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        _smoke_fd_pair(op, x, comm)
+    comm.barrier()
+    t1 = time.perf_counter()
+    elapsed = torch.tensor([t1 - t0], dtype=torch.float64)
+    comm.allreduce_(elapsed, "max")
+    t = float(elapsed.item())
+    if rank == 0:
+        print(json.dumps({
+            "metric": "MPIFirstDerivative fp64 matvec+rmatvec pairs/sec",
+            "value": args.steps / t, "unit": "pairs/s", "n_gpus": n_gpus,
+            "steps": args.steps, "warmup": args.warmup,
+            "ms_per_step": 1e3 * t / args.steps, "higher_is_better": True,
+            "scaling": "weak", "vs_baseline": None, "dtype": "f64",
+            "data": "synthetic", "smoke": True,
+            "config": {"workload": wname, "dims": list(dims),
+                       "partition": "scatter-axis0", "kind": "centered",
+                       "order": 3},
+            "roofline": None, "cpu_baseline": None,
+        }), flush=True)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=50)  # SURVEY 8d: >=50
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--smoke-gloo", action="store_true",
+                    help="CPU orchestration smoke (tiny dims, torch "
+                         "stand-in kernel; NOT a benchmark)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -155,10 +234,11 @@ def main():
     comm = init_default_comm()
     rank = comm.rank
     assert comm.size == n_gpus, f"launch {n_gpus} ranks (got {comm.size})"
-    assert torch.cuda.is_available(), "bench needs MI355X GPUs"
+    if not args.smoke_gloo:
+        assert torch.cuda.is_available(), "bench needs MI355X GPUs"
     device = comm.device
 
-    dims, wname = workload_dims(n_gpus)
+    dims, wname = workload_dims(n_gpus, smoke=args.smoke_gloo)
     n = int(np.prod(dims))
     m = dims[1] * dims[2]
 
@@ -173,6 +253,10 @@ def main():
                             local_array=local,
                             local_shapes=[(c,) for c in counts],
                             dtype=np.float64)
+
+    if args.smoke_gloo:
+        _smoke_main(args, op, x, comm, n_gpus, dims, wname, rank)
+        return
 
     for _ in range(args.warmup):
         y = op.matvec(x)
@@ -197,13 +281,17 @@ def main():
     comm.allreduce_(elapsed, "max")
     t = float(elapsed.item())
 
-    # roofline of the dominant kernel (fd centered3 matvec, op code 4)
+    # roofline of the dominant kernel (fd centered3 matvec, op code 4).
+    # One event pair brackets one whole matvec: a single launch at
+    # world 1; interior launch + halo wait + boundary launches under the
+    # N>1 overlap path (so the N>1 figure conservatively includes any
+    # exposed halo time).  Aggregate per STEP (sum / steps).
     mv_events = deriv.KERNEL_EVENTS.get(4, [])
     kern_ms = [e0.elapsed_time(e1) for e0, e1 in mv_events]
     local_pts = counts[rank]
-    avg_ms = float(np.mean(kern_ms)) if kern_ms else None
+    step_ms = float(np.sum(kern_ms)) / args.steps if kern_ms else None
     alg_bytes = BYTES_PER_PT * local_pts
-    achieved = alg_bytes / (avg_ms * 1e-3) / 1e9 if avg_ms else None
+    achieved = alg_bytes / (step_ms * 1e-3) / 1e9 if step_ms else None
     traffic = read_traffic_calibration(dims, n_gpus)
 
     pairs_per_s = args.steps / t
