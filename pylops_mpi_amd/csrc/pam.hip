@@ -824,17 +824,25 @@ extern "C" int64_t pam_fd_halo_width(int op) {
   }
 }
 
-template <typename T, int OP, int V, bool NTS = false>
+template <typename T, int OP, int V, bool NTS = false, bool RSWAP = false>
 __global__ void __launch_bounds__(BLK) fd_kernel(Rows<T> R, T* __restrict__ y,
                                                  int64_t row0, int64_t N, T c,
                                                  int edge, int64_t rbegin,
                                                  int64_t rend) {
+  // RSWAP (PAM_FD_RSWAP=1 A/B): row index from blockIdx.x instead of .y,
+  // so XCD round-robin dispatch (linear id mod 8) lands CONSECUTIVE ROWS
+  // on consecutive XCDs instead of consecutive column chunks — probes
+  // whether the long-row deficit is an XCD/L2 block-mapping effect.
+  const int64_t brow = RSWAP ? blockIdx.x : blockIdx.y;
+  const int64_t grows = RSWAP ? gridDim.x : gridDim.y;
+  const int64_t bcol = RSWAP ? blockIdx.y : blockIdx.x;
+  const int64_t gcols = RSWAP ? gridDim.y : gridDim.x;
   const int64_t m = R.m, mv = m / V;
-  const int64_t cstride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = rbegin + blockIdx.y; i < rend; i += gridDim.y) {
+  const int64_t cstride = (int64_t)gcols * blockDim.x;
+  for (int64_t i = rbegin + brow; i < rend; i += grows) {
     const int64_t g = row0 + i;
     T* __restrict__ yrow = y + i * m;
-    for (int64_t jv = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; jv < mv;
+    for (int64_t jv = (int64_t)bcol * blockDim.x + threadIdx.x; jv < mv;
          jv += cstride) {
       const int64_t j = jv * V;
       T acc[V];
@@ -857,7 +865,7 @@ __global__ void __launch_bounds__(BLK) fd_kernel(Rows<T> R, T* __restrict__ y,
     }
     // scalar tail columns (m not divisible by V)
     if constexpr (V > 1) {
-      for (int64_t j = mv * V + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+      for (int64_t j = mv * V + (int64_t)bcol * blockDim.x + threadIdx.x;
            j < m; j += cstride) {
         T acc[1] = {(T)0};
 #pragma unroll
@@ -1092,6 +1100,28 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
       hipLaunchKernelGGL((fd_roll_kernel<T, OP, 2, CV>), gridr, dim3(BLK),
                          0, s, R, (T*)y, row0, nglob, (T)coeff, edge,
                          rbegin, rend);
+    return check(hipGetLastError());
+  }
+  // PAM_FD_RSWAP=1: row index from blockIdx.x (XCD-mapping A/B; see
+  // fd_kernel docstring).  Grid dims swap with it.
+  static int rswap = [] {
+    const char* e = getenv("PAM_FD_RSWAP");
+    return e ? atoi(e) : 0;
+  }();
+  if (rswap) {
+    dim3 grid_s((uint32_t)gy, (uint32_t)gx64);
+    if (V == 4)
+      hipLaunchKernelGGL((fd_kernel<T, OP, 4, false, true>), grid_s,
+                         dim3(BLK), 0, s, R, (T*)y, row0, nglob, (T)coeff,
+                         edge, rbegin, rend);
+    else if (V == 2 && sizeof(T) == 8)
+      hipLaunchKernelGGL((fd_kernel<T, OP, 2, false, true>), grid_s,
+                         dim3(BLK), 0, s, R, (T*)y, row0, nglob, (T)coeff,
+                         edge, rbegin, rend);
+    else
+      hipLaunchKernelGGL((fd_kernel<T, OP, 1, false, true>), grid_s,
+                         dim3(BLK), 0, s, R, (T*)y, row0, nglob, (T)coeff,
+                         edge, rbegin, rend);
     return check(hipGetLastError());
   }
   const bool nt = fd_nt_override() != 0;

@@ -118,3 +118,72 @@ def test_matmult_saveat_and_dottest():
     v = pm.DistributedArray((N * M,))
     v[:] = dev(rng.standard_normal(N * M))
     assert pm.dottest(op, u, v, rtol=1e-10)
+
+
+@pytest.mark.parametrize("kind", ["block", "summa"])
+@pytest.mark.parametrize("dtype", ["complex64", "complex128"])
+def test_matmult_grid1_complex(kind, dtype):
+    """Complex MatrixMult (r01 advice): batch-1 MFMA cgemm panels +
+    conj-transpose materialization, vs dense A@x / A^H y (the reference
+    supports complex throughout, ref MatrixMult.py:346-352,416,737)."""
+    N, K, M = 67, 45, 33
+    rng = np.random.default_rng(7)
+    A = (rng.standard_normal((N, K))
+         + 1j * rng.standard_normal((N, K))).astype(dtype)
+    X = (rng.standard_normal((K, M))
+         + 1j * rng.standard_normal((K, M))).astype(dtype)
+    Y = (rng.standard_normal((N, M))
+         + 1j * rng.standard_normal((N, M))).astype(dtype)
+    for saveAt in (False, True):
+        op = pm.MPIMatrixMult(dev(A), M, kind=kind, dtype=dtype,
+                              saveAt=saveAt)
+        assert op.N == N and op.K == K
+        x = pm.DistributedArray((K * M,), dtype=np.dtype(dtype))
+        x[:] = dev(X.ravel())
+        got = op.matvec(x)
+        tol = 1e-12 if dtype == "complex128" else 2e-4
+        assert_allclose(got.local_array.cpu().numpy(),
+                        (A @ X).ravel(), rtol=tol, atol=tol * 10)
+        yv = pm.DistributedArray((N * M,), dtype=np.dtype(dtype))
+        yv[:] = dev(Y.ravel())
+        gotr = op.rmatvec(yv)
+        assert_allclose(gotr.local_array.cpu().numpy(),
+                        (A.conj().T @ Y).ravel(), rtol=tol, atol=tol * 10)
+
+
+@pytest.mark.parametrize("dtype", ["complex64", "complex128"])
+def test_ctranspose(dtype):
+    rng = np.random.default_rng(8)
+    for nr, nc in [(32, 32), (33, 65), (1, 7), (100, 3)]:
+        A = (rng.standard_normal((nr, nc))
+             + 1j * rng.standard_normal((nr, nc))).astype(dtype)
+        Ad = dev(A).contiguous()
+        At = torch.empty((nc, nr), dtype=Ad.dtype, device="cuda:0")
+        s = torch.cuda.current_stream().cuda_stream
+        for conj in (0, 1):
+            _ffi.checked(_ffi.lib().pam_ctranspose(
+                s, Ad.data_ptr(), At.data_ptr(), nr, nc, conj,
+                _ffi.dtype_code(Ad.dtype)), "ct")
+            want = A.conj().T if conj else A.T
+            assert np.array_equal(At.cpu().numpy(), want)
+
+
+@pytest.mark.parametrize("dtype", ["complex64", "complex128"])
+def test_cgemm_accumulate(dtype):
+    """C += op(A) @ B (the accumulate flag added for complex SUMMA)."""
+    rng = np.random.default_rng(9)
+    B_, M, K, N = 3, 64, 48, 40
+    A = (rng.standard_normal((B_, M, K))
+         + 1j * rng.standard_normal((B_, M, K))).astype(dtype)
+    X = (rng.standard_normal((B_, K, N))
+         + 1j * rng.standard_normal((B_, K, N))).astype(dtype)
+    C0 = (rng.standard_normal((B_, M, N))
+          + 1j * rng.standard_normal((B_, M, N))).astype(dtype)
+    Cd = dev(C0).contiguous()
+    Ad, Xd = dev(A).contiguous(), dev(X).contiguous()
+    s = torch.cuda.current_stream().cuda_stream
+    _ffi.checked(_ffi.lib().pam_cgemm_batched(
+        s, Ad.data_ptr(), Xd.data_ptr(), Cd.data_ptr(), B_, M, N, K,
+        M * K, K * N, M * N, 0, 1, _ffi.dtype_code(Ad.dtype)), "cgemm")
+    tol = 1e-12 if dtype == "complex128" else 2e-4
+    assert_allclose(Cd.cpu().numpy(), C0 + A @ X, rtol=tol, atol=tol * 10)
