@@ -131,6 +131,29 @@ class PamComm:
             dist.broadcast(t, src=self.ranks[root], group=self._group)
         return t
 
+    def broadcast_async(self, t: torch.Tensor, root: int = 0):
+        """Asynchronous broadcast: returns the Work handle (or None when
+        single-rank).  The SUMMA k-loop posts step k+1's tile broadcasts
+        while step k's panel GEMM runs (RCCL uses its own stream, so the
+        transfer overlaps compute; work.wait() is stream-ordered)."""
+        if not self._use_dist:
+            return None
+        return dist.broadcast(t, src=self.ranks[root], group=self._group,
+                              async_op=True)
+
+    def exchange_async(self, sends: List, recvs: List):
+        """Batched point-to-point round posted WITHOUT waiting; returns
+        the Work handles (see exchange() for the pairing rules)."""
+        if not self._use_dist:
+            for (st, _), (rt, _) in zip(sends, recvs):
+                rt.copy_(st)
+            return []
+        ops = [dist.P2POp(dist.irecv, t, self.ranks[src])
+               for t, src in recvs]
+        ops += [dist.P2POp(dist.isend, t, self.ranks[dst])
+                for t, dst in sends]
+        return dist.batch_isend_irecv(ops) if ops else []
+
     def allgather_obj(self, obj) -> List:
         """Object allgather on the control plane (ref Distributed.py:113-154
         object-mode branch)."""

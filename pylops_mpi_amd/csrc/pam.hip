@@ -998,6 +998,111 @@ __global__ void __launch_bounds__(BLK) fd_roll_kernel(
   }
 }
 
+template <int... Us, typename F>
+__device__ __forceinline__ void fd_static_for(
+    std::integer_sequence<int, Us...>, F&& f) {
+  (f(std::integral_constant<int, Us>{}), ...);
+}
+
+// roll2 term accumulation: window slot is ((off + W + ROT) mod NROLL) —
+// all constexpr, so the window lives in registers with NO per-row shift
+template <typename T, int OP, int V, int W, int ROT, int t>
+__device__ __forceinline__ void roll2_term(const T (&win)[2 * W + 1][V],
+                                           T (&acc)[V], int64_t g,
+                                           int64_t N) {
+  constexpr Term tm = FDDef<OP>::TERMS[t];
+  constexpr int NR = 2 * W + 1;
+  constexpr int slot = ((tm.off + W + ROT) % NR + NR) % NR;
+  if (g >= tm.lo && g <= N - 1 - tm.hi) {
+#pragma unroll
+    for (int k = 0; k < V; ++k)
+      acc[k] += (T)tm.coeff * win[slot][k];
+  }
+}
+
+template <typename T, int OP, int V, int W, int ROT, int... Ts>
+__device__ __forceinline__ void roll2_terms(
+    const T (&win)[2 * W + 1][V], T (&acc)[V], int64_t g, int64_t N,
+    std::integer_sequence<int, Ts...>) {
+  (roll2_term<T, OP, V, W, ROT, Ts>(win, acc, g, N), ...);
+}
+
+// Statically-rotated rolling stencil (PAM_FD_ROLL=2, r02): like
+// fd_roll_kernel (1 load + 1 store per point — the PMC-measured fix for
+// the long-row shape, where the row-parallel kernel's neighbour re-read
+// really does go to HBM: FETCH 8.53 GB vs 4.29 algorithmic at
+// 512x4096x256, r02 session 1), but the row loop is unrolled by the
+// window depth so the slot indices rotate at COMPILE time — no
+// register-to-register window shift per row (fd_roll_kernel spends
+// 2W*V*CV moves per row step; this kernel spends zero).  Loads for row
+// i+W+1 issue before row i's arithmetic, giving CV independent loads
+// in flight per thread.
+template <typename T, int OP, int V, int CV>
+__global__ void __launch_bounds__(BLK) fd_roll2_kernel(
+    Rows<T> R, T* __restrict__ y, int64_t row0, int64_t N, T c, int edge,
+    int64_t rbegin, int64_t rend) {
+  constexpr int W = FDDef<OP>::W;
+  constexpr int NROLL = 2 * W + 1;
+  const int64_t m = R.m, mv = m / V;
+  const int64_t base = (int64_t)blockIdx.x * blockDim.x * CV + threadIdx.x;
+  const int64_t nr = rend - rbegin;
+  const int64_t c0 = rbegin + (nr * blockIdx.y) / gridDim.y;
+  const int64_t c1 = rbegin + (nr * (blockIdx.y + 1)) / gridDim.y;
+  if (c1 <= c0 || base >= mv) return;
+  const int64_t lo = R.gf ? -(int64_t)R.w : 0;
+  const int64_t hi = R.nloc - 1 + (R.gb ? R.w : 0);
+  auto clamp_row = [&](int64_t i) {
+    return i < lo ? lo : (i > hi ? hi : i);
+  };
+  int64_t js[CV];
+  bool vq[CV];
+#pragma unroll
+  for (int q = 0; q < CV; ++q) {
+    const int64_t jq = base + (int64_t)q * blockDim.x;
+    vq[q] = jq < mv;
+    js[q] = (vq[q] ? jq : mv - 1) * V;
+  }
+  T buf[CV][NROLL][V];
+#pragma unroll
+  for (int q = 0; q < CV; ++q)
+#pragma unroll
+    for (int k = 0; k < NROLL; ++k)
+      loadv<T, V>(R.row(clamp_row(c0 - W + k)) + js[q], buf[q][k]);
+  int64_t i = c0;
+  auto step = [&](auto rotc) {
+    constexpr int ROT = decltype(rotc)::value;
+    constexpr int oldest = ROT % NROLL;
+    const int64_t rn = clamp_row(i + W + 1);
+    T nxt[CV][V];
+#pragma unroll
+    for (int q = 0; q < CV; ++q)
+      loadv<T, V>(R.row(rn) + js[q], nxt[q]);
+    const int64_t g = row0 + i;
+#pragma unroll
+    for (int q = 0; q < CV; ++q) {
+      T acc[V];
+#pragma unroll
+      for (int k = 0; k < V; ++k) acc[k] = (T)0;
+      roll2_terms<T, OP, V, W, ROT>(
+          buf[q], acc, g, N,
+          std::make_integer_sequence<int, FDDef<OP>::NT>{});
+      if (edge) fd_edge<T, OP, V>(R, i, js[q], g, N, acc);
+#pragma unroll
+      for (int k = 0; k < V; ++k) acc[k] *= c;
+      if (vq[q]) storev<T, V>(y + i * m + js[q], acc);
+#pragma unroll
+      for (int v = 0; v < V; ++v) buf[q][oldest][v] = nxt[q][v];
+    }
+    ++i;
+  };
+  while (i + NROLL <= c1)
+    fd_static_for(std::make_integer_sequence<int, NROLL>{},
+                  [&](auto rc) { step(rc); });
+  fd_static_for(std::make_integer_sequence<int, NROLL>{}, [&](auto rc) {
+    if (i < c1) step(rc);
+  });
+}
+
 static int fd_vec_override() {
   static int v = [] {
     const char* e = getenv("PAM_FD_VEC");
@@ -1082,7 +1187,11 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
     return e ? atoi(e) : 4096;
   }();
   if (rollov && V > 1) {
-    constexpr int CV = 4;  // independent rolling chains per thread
+    static int cvov = [] {
+      const char* e = getenv("PAM_FD_ROLL_CV");
+      return e ? atoi(e) : 4;
+    }();
+    const int CV = (cvov == 2 || cvov == 8) ? cvov : 4;
     const int64_t mv2 = m / V;
     int64_t gx = (mv2 + (int64_t)BLK * CV - 1) / ((int64_t)BLK * CV);
     int64_t gyr = rolltgt / (gx ? gx : 1);
@@ -1092,14 +1201,25 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
     if (gyr < 1) gyr = 1;
     if (gyr > 65535) gyr = 65535;
     dim3 gridr((uint32_t)gx, (uint32_t)gyr);
-    if (V == 4)
-      hipLaunchKernelGGL((fd_roll_kernel<T, OP, 4, CV>), gridr, dim3(BLK),
-                         0, s, R, (T*)y, row0, nglob, (T)coeff, edge,
-                         rbegin, rend);
-    else
-      hipLaunchKernelGGL((fd_roll_kernel<T, OP, 2, CV>), gridr, dim3(BLK),
-                         0, s, R, (T*)y, row0, nglob, (T)coeff, edge,
-                         rbegin, rend);
+#define ROLL_LAUNCH(KER, VV, CC)                                              \
+  hipLaunchKernelGGL((KER<T, OP, VV, CC>), gridr, dim3(BLK), 0, s, R,         \
+                     (T*)y, row0, nglob, (T)coeff, edge, rbegin, rend)
+    if (rollov == 2) {        // statically-rotated variant
+      if (V == 4 && CV == 2) ROLL_LAUNCH(fd_roll2_kernel, 4, 2);
+      else if (V == 4 && CV == 8) ROLL_LAUNCH(fd_roll2_kernel, 4, 8);
+      else if (V == 4) ROLL_LAUNCH(fd_roll2_kernel, 4, 4);
+      else if (CV == 2) ROLL_LAUNCH(fd_roll2_kernel, 2, 2);
+      else if (CV == 8) ROLL_LAUNCH(fd_roll2_kernel, 2, 8);
+      else ROLL_LAUNCH(fd_roll2_kernel, 2, 4);
+    } else {
+      if (V == 4 && CV == 2) ROLL_LAUNCH(fd_roll_kernel, 4, 2);
+      else if (V == 4 && CV == 8) ROLL_LAUNCH(fd_roll_kernel, 4, 8);
+      else if (V == 4) ROLL_LAUNCH(fd_roll_kernel, 4, 4);
+      else if (CV == 2) ROLL_LAUNCH(fd_roll_kernel, 2, 2);
+      else if (CV == 8) ROLL_LAUNCH(fd_roll_kernel, 2, 8);
+      else ROLL_LAUNCH(fd_roll_kernel, 2, 4);
+    }
+#undef ROLL_LAUNCH
     return check(hipGetLastError());
   }
   // PAM_FD_RSWAP=1: row index from blockIdx.x (XCD-mapping A/B; see

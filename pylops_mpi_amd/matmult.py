@@ -431,21 +431,31 @@ class _MPISummaMatrixMult(_MatMultBase):
             bkX, bm)
         Y_local = torch.zeros((bn, bm), dtype=self.A.dtype,
                               device=self.A.device)
-        for t in range(L):
+
+        def post(t):
             cA, pA, rX, pX = self._rect_panels(t)
             if self._col_id == cA:
                 Aslice = self.A[:, pA * bkp: (pA + 1) * bkp].contiguous()
             else:
                 Aslice = torch.empty((bn, bkp), dtype=self.A.dtype,
                                      device=self.A.device)
-            self._row_comm.broadcast_(Aslice, root=cA)
+            wa = self._row_comm.broadcast_async(Aslice, root=cA)
             if self._row_id == rX:
                 Xslice = x_block[pX * bkp: (pX + 1) * bkp, :].contiguous()
             else:
                 Xslice = torch.empty((bkp, bm), dtype=self.A.dtype,
                                      device=self.A.device)
-            self._col_comm.broadcast_(Xslice, root=rX)
+            wx = self._col_comm.broadcast_async(Xslice, root=rX)
+            return Aslice, Xslice, [w for w in (wa, wx) if w is not None]
+
+        cur = post(0)
+        for t in range(L):
+            nxt = post(t + 1) if t + 1 < L else None
+            Aslice, Xslice, works = cur
+            for w in works:
+                w.wait()
             self._local_gemm(Aslice, Xslice, Y_local, accumulate=True)
+            cur = nxt
         y[:] = Y_local[:local_n, :local_m].reshape(-1)
         return y
 
@@ -464,14 +474,23 @@ class _MPISummaMatrixMult(_MatMultBase):
             bn, bm)
         Z = torch.zeros((bkX, bm), dtype=self.A.dtype,
                         device=self.A.device)
-        for t in range(L):
+
+        def post(t):
             cA, pA, rX, pX = self._rect_panels(t)
             if self._col_id == cA:
                 Aslice = self.A[:, pA * bkp: (pA + 1) * bkp].contiguous()
             else:
                 Aslice = torch.empty((bn, bkp), dtype=self.A.dtype,
                                      device=self.A.device)
-            self._row_comm.broadcast_(Aslice, root=cA)
+            wa = self._row_comm.broadcast_async(Aslice, root=cA)
+            return Aslice, rX, pX, ([wa] if wa is not None else [])
+
+        cur = post(0)
+        for t in range(L):
+            nxt = post(t + 1) if t + 1 < L else None
+            Aslice, rX, pX, works = cur
+            for w in works:
+                w.wait()
             # G = Aslice^H @ x_block, summed over the grid column to the
             # X-panel owner (replaces the square path's tag-routed A^T
             # exchange, ref :745-760 — RCCL has no tags)
@@ -479,6 +498,7 @@ class _MPISummaMatrixMult(_MatMultBase):
             self._col_comm.reduce_(G, root=rX)
             if self._row_id == rX:
                 Z[pX * bkp: (pX + 1) * bkp, :] = G
+            cur = nxt
         y[:] = Z[:local_k, :local_m].reshape(-1)
         return y
 
@@ -530,14 +550,27 @@ class _MPISummaMatrixMult(_MatMultBase):
             x.local_array.reshape(local_k, local_m).to(self.A.dtype), bk, bm)
         Y_local = torch.zeros((self.A.shape[0], bm), dtype=self.A.dtype,
                               device=self.A.device)
-        for k in range(P):
+
+        # bcasts pipelined across the k-loop: step k+1's tiles are in
+        # flight while step k's panel GEMM runs (the reference serializes
+        # bcast -> GEMM every step, ref :666-668)
+        def post(k):
             Atemp = self.A.contiguous() if self._col_id == k \
                 else torch.empty_like(self.A)
             Xtemp = x_block if self._row_id == k \
                 else torch.empty_like(x_block)
-            self._row_comm.broadcast_(Atemp, root=k)   # ref :666
-            self._col_comm.broadcast_(Xtemp, root=k)   # ref :667
+            wa = self._row_comm.broadcast_async(Atemp, root=k)  # ref :666
+            wx = self._col_comm.broadcast_async(Xtemp, root=k)  # ref :667
+            return Atemp, Xtemp, [w for w in (wa, wx) if w is not None]
+
+        cur = post(0)
+        for k in range(P):
+            nxt = post(k + 1) if k + 1 < P else None
+            Atemp, Xtemp, works = cur
+            for w in works:
+                w.wait()
             self._local_gemm(Atemp, Xtemp, Y_local, accumulate=True)
+            cur = nxt
         y[:] = Y_local[:local_n, :local_m].reshape(-1)
         return y
 
@@ -559,10 +592,13 @@ class _MPISummaMatrixMult(_MatMultBase):
                               device=self.A.device)
         comm = self.base_comm_grid
         me = comm.rank
-        for k in range(P):
+
+        # comm pipelined across the k-loop like the forward: step k+1's
+        # X bcast and A^H routing round are in flight under step k's GEMM
+        def post(k):
             Xtemp = x_block if self._row_id == k \
                 else torch.empty_like(x_block)
-            self._col_comm.broadcast_(Xtemp, root=k)
+            wx = self._col_comm.broadcast_async(Xtemp, root=k)
             # A^T routing (ref :745-760): rank (r,c) consumes the A^T of
             # grid rank (k, r); grid-row-k members send theirs to every
             # member of grid row <their col_id>.
@@ -578,8 +614,19 @@ class _MPISummaMatrixMult(_MatMultBase):
             else:
                 ATtemp = torch.empty_like(A_local)
                 recvs.append((ATtemp, srcA))
-            comm.exchange(sends, recvs)
+            works = comm.exchange_async(sends, recvs)
+            if wx is not None:
+                works = works + [wx]
+            return ATtemp, Xtemp, works
+
+        cur = post(0)
+        for k in range(P):
+            nxt = post(k + 1) if k + 1 < P else None
+            ATtemp, Xtemp, works = cur
+            for w in works:
+                w.wait()
             self._local_gemm(ATtemp, Xtemp, Y_local, accumulate=True)
+            cur = nxt
         y[:] = Y_local[:local_k, :local_m].reshape(-1)
         return y
 
