@@ -930,7 +930,7 @@ __device__ __forceinline__ void roll_terms(const T (&win)[2 * W + 1][V],
 // mix) is the round-2 question.
 // Guards: load indices are clamped to existing rows (halo planes
 // included); term/edge masks ignore the clamped garbage.
-template <typename T, int OP, int V, int CV>
+template <typename T, int OP, int V, int CV, bool NTS = false>
 __global__ void __launch_bounds__(BLK) fd_roll_kernel(
     Rows<T> R, T* __restrict__ y, int64_t row0, int64_t N, T c, int edge,
     int64_t rbegin, int64_t rend) {
@@ -987,7 +987,7 @@ __global__ void __launch_bounds__(BLK) fd_roll_kernel(
       if (edge) fd_edge<T, OP, V>(R, i, js[q], g, N, acc);
 #pragma unroll
       for (int k = 0; k < V; ++k) acc[k] *= c;
-      if (vq[q]) storev<T, V>(y + i * m + js[q], acc);
+      if (vq[q]) storev_p<T, V, NTS>(y + i * m + js[q], acc);
 #pragma unroll
       for (int k = 0; k < NROLL - 1; ++k)
 #pragma unroll
@@ -1037,7 +1037,7 @@ __device__ __forceinline__ void roll2_terms(
 // 2W*V*CV moves per row step; this kernel spends zero).  Loads for row
 // i+W+1 issue before row i's arithmetic, giving CV independent loads
 // in flight per thread.
-template <typename T, int OP, int V, int CV>
+template <typename T, int OP, int V, int CV, bool NTS = false>
 __global__ void __launch_bounds__(BLK) fd_roll2_kernel(
     Rows<T> R, T* __restrict__ y, int64_t row0, int64_t N, T c, int edge,
     int64_t rbegin, int64_t rend) {
@@ -1089,7 +1089,7 @@ __global__ void __launch_bounds__(BLK) fd_roll2_kernel(
       if (edge) fd_edge<T, OP, V>(R, i, js[q], g, N, acc);
 #pragma unroll
       for (int k = 0; k < V; ++k) acc[k] *= c;
-      if (vq[q]) storev<T, V>(y + i * m + js[q], acc);
+      if (vq[q]) storev_p<T, V, NTS>(y + i * m + js[q], acc);
 #pragma unroll
       for (int v = 0; v < V; ++v) buf[q][oldest][v] = nxt[q][v];
     }
@@ -1201,9 +1201,18 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
     if (gyr < 1) gyr = 1;
     if (gyr > 65535) gyr = 65535;
     dim3 gridr((uint32_t)gx, (uint32_t)gyr);
+    const bool rnt = fd_nt_override() != 0;
 #define ROLL_LAUNCH(KER, VV, CC)                                              \
-  hipLaunchKernelGGL((KER<T, OP, VV, CC>), gridr, dim3(BLK), 0, s, R,         \
-                     (T*)y, row0, nglob, (T)coeff, edge, rbegin, rend)
+  do {                                                                        \
+    if (rnt)                                                                  \
+      hipLaunchKernelGGL((KER<T, OP, VV, CC, true>), gridr, dim3(BLK), 0, s,  \
+                         R, (T*)y, row0, nglob, (T)coeff, edge, rbegin,       \
+                         rend);                                               \
+    else                                                                      \
+      hipLaunchKernelGGL((KER<T, OP, VV, CC, false>), gridr, dim3(BLK), 0, s, \
+                         R, (T*)y, row0, nglob, (T)coeff, edge, rbegin,       \
+                         rend);                                               \
+  } while (0)
     if (rollov == 2) {        // statically-rotated variant
       if (V == 4 && CV == 2) ROLL_LAUNCH(fd_roll2_kernel, 4, 2);
       else if (V == 4 && CV == 8) ROLL_LAUNCH(fd_roll2_kernel, 4, 8);

@@ -18,7 +18,7 @@ the semantically-required real-part extraction feeding rfft (the
 reference takes .real too) and the rocFFT transforms themselves.
 """
 import logging
-import math
+
 
 import numpy as np
 import torch
@@ -98,12 +98,21 @@ class _FusedMDC(MPILinearOperator):
                  shift: bool) -> torch.Tensor:
         t = x.local_array.reshape(self.nt, nmid * self.nv)
         if t.is_complex():
-            t = t.real
+            # contiguous real extraction: rfft on the stride-2 .real
+            # view forces a strided rocFFT plan (r01 DESIGN's named
+            # remaining cost); one coalesced copy is cheaper
+            t = t.real.contiguous()
         if shift:
             t = torch.fft.ifftshift(t, dim=0)
-        f = torch.fft.rfft(t, n=self.nt, dim=0, norm="ortho")
-        f[self._tw0: self._tw1] *= math.sqrt(2.0)
-        return f  # fresh storage (safe to slice/view)
+        # NOTE the sqrt(2) conjugate-twin scaling of the pylops real-FFT
+        # convention (fftlocal.py:47-48) is NOT applied here: inside
+        # this chain the forward's *sqrt(2) on bins [tw0, tw1) and the
+        # inverse's /sqrt(2) on the SAME bins surround the per-frequency
+        # block-diagonal Fredholm kernel, so they cancel algebraically
+        # (bins >= nfreq are masked to zero either way).  Two full
+        # passes over the frequency tensor disappear; standalone FFT
+        # operators (fftlocal.FFTLocal) keep the scaling.
+        return torch.fft.rfft(t, n=self.nt, dim=0, norm="ortho")
 
     def _inv_fft(self, fr: torch.Tensor, nmid: int,
                  shift: bool) -> torch.Tensor:
@@ -114,7 +123,7 @@ class _FusedMDC(MPILinearOperator):
             z = torch.zeros(self.nfft, nmid * self.nv, dtype=fr.dtype,
                             device=fr.device)
             z[: self.nfreq] = fr.reshape(self.nfreq, nmid * self.nv)
-        z[self._tw0: self._tw1] /= math.sqrt(2.0)
+        # twin-bin /sqrt(2) cancelled against the forward side (above)
         out = torch.fft.irfft(z, n=self.nt, dim=0, norm="ortho")
         if shift:
             out = torch.fft.fftshift(out, dim=0)

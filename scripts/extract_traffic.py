@@ -25,6 +25,8 @@ def parse(files):
             ccol = next((c for c in cols if "Counter_Name" in c), None)
             vcol = next((c for c in cols if "Counter_Value" in c), None)
             dcol = next((c for c in cols if "Dispatch" in c and "Id" in c), None)
+            if not (kcol and ccol and vcol and dcol):
+                continue  # agent_info / other sidecar CSVs
             rows = defaultdict(dict)
             for row in rd:
                 rows[row[dcol]][row[ccol]] = float(row[vcol])
