@@ -18,6 +18,7 @@
 
 #include <hip/hip_runtime.h>
 #include <stdint.h>
+#include <stdlib.h>
 
 #include "../../include/pam.h"
 
@@ -304,7 +305,8 @@ extern "C" int pam_gemm(void* stream, const void* A, const void* B, void* C,
 // numbers are in DESIGN.md) — the batch dimension fills the chip either
 // way, but only MFMA reaches the matrix-core rate.
 // ---------------------------------------------------------------------------
-template <typename T, bool CT, bool ACC = false, int BK = 16>
+template <typename T, bool CT, bool ACC = false, int BK = 16,
+          int BM = 64, int BN = 64>
 __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
     const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
     int64_t M, int64_t N, int64_t K, int64_t strideA, int64_t strideB,
@@ -316,7 +318,9 @@ __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
   //   rmatvec), f32 BN=128 wide tile (69.2 -> 66.2 TF: halving the WG
   //   count costs more latency hiding than the doubled fragment reuse
   //   gains).  BK stays a template param: 16 everywhere.
-  constexpr int BM = 64, BN = 64;
+  // BM/BN default 64x64; 128x64 is dispatched for the shallow-K
+  // adjoint panels (cfg5 rmatvec K=64: 4x the work per block at the
+  // same 4-panel pipeline depth — r02 A/B)
   constexpr int NBUFC = (BK >= 64) ? 1 : 2;
   constexpr int MI = (BM / 2) / TM, NJ = (BN / 2) / TM;
   using acc_t = typename CFG::acc_t;
@@ -515,9 +519,32 @@ static int cgemm_launch(void* stream, const void* A, const void* B, void* C,
                         int acc) {
   if (batch <= 0 || M <= 0 || N <= 0 || K < 0 || !A || !B || !C)
     return PAM_EARG;
+  hipStream_t s = (hipStream_t)stream;
+  // 128x64 tile for shallow-K panels (PAM_CGEMM_TILE: 0 auto, 1 force
+  // 64x64, 2 force 128x64): at the cfg5 rmatvec shape (M=N=256, K=64)
+  // the 64x64 tile runs a 4-panel pipeline with 4x the blocks of the
+  // matvec shape — doubling BM doubles per-block work at the same
+  // pipeline depth (r02 A/B).
+  static int tileov = [] {
+    const char* e = getenv("PAM_CGEMM_TILE");
+    return e ? atoi(e) : 0;
+  }();
+  const bool wide = !acc && M >= 128 && tileov == 2;  // auto condition pending the r02 A/B
+  if (wide) {
+    dim3 gridw((uint32_t)((N + 63) / 64), (uint32_t)((M + 127) / 128),
+               (uint32_t)batch);
+    if (opa)
+      hipLaunchKernelGGL((cgemm_batched_kernel<T, true, false, 16, 128, 64>),
+                         gridw, dim3(GBLK), 0, s, (const T*)A, (const T*)B,
+                         (T*)C, M, N, K, sA, sB, sC);
+    else
+      hipLaunchKernelGGL((cgemm_batched_kernel<T, false, false, 16, 128, 64>),
+                         gridw, dim3(GBLK), 0, s, (const T*)A, (const T*)B,
+                         (T*)C, M, N, K, sA, sB, sC);
+    return gcheck(hipGetLastError());
+  }
   dim3 grid((uint32_t)((N + 63) / 64), (uint32_t)((M + 63) / 64),
             (uint32_t)batch);
-  hipStream_t s = (hipStream_t)stream;
   // A BK=64 single-panel variant for K <= 64 (one barrier, no pipeline)
   // was A/B'd NEGATIVE at the cfg5 rmatvec shape (0.432 -> 0.518 ms):
   // the 133 KB LDS footprint drops occupancy to 2 WG/CU, which costs

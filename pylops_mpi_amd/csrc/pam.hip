@@ -1173,25 +1173,41 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
   if (gx64 > cap) gx64 = cap;
   dim3 grid((uint32_t)gx64, (uint32_t)gy);
   hipStream_t s = (hipStream_t)stream;
-  // rolling-window path (fd_roll_kernel) — A/B'd NEGATIVE as default:
-  // bench shape 331 -> 306 pairs/s, and the long-row (512,4096,256)
-  // shape it was built for is unchanged (4.9 TB/s both ways: that
-  // shape's wall is not neighbour-row re-read traffic).  Kept behind
-  // PAM_FD_ROLL=1 with PAM_FD_ROLL_TGT for round-2 investigation.
+  // rolling-window path: DEFAULT for long rows (r02).  r01's negative
+  // A/B predates the constexpr-folded window; the r02 PMC runs show the
+  // row-parallel kernel's neighbour-row re-reads really go to HBM once
+  // a row outgrows L2 absorption (FETCH 8.53 GB vs 4.29 algorithmic at
+  // 512x4096x256 => 4.85 TB/s algorithmic), while the rolling kernel
+  // moves exactly 1R+1W (PMC 1.035x) and measures 5.35 TB/s there with
+  // V=4/CV=8/TGT=2048 (the swept optimum; NT stores -15%, roll2's
+  // static rotation spills at CV8V4).  The bench shape (2 MiB rows)
+  // keeps the row-parallel kernel (5.78 vs 5.22).  PAM_FD_ROLL: unset =
+  // auto (rows >= PAM_FD_LONGROW bytes, default 4 MiB), -1 = force
+  // row-parallel, 1/2 = force rolling (2 = statically-rotated variant).
   static int rollov = [] {
     const char* e = getenv("PAM_FD_ROLL");
     return e ? atoi(e) : 0;
   }();
+  static int64_t longrow = [] {
+    const char* e = getenv("PAM_FD_LONGROW");
+    return (int64_t)(e ? atoll(e) : (4 << 20));
+  }();
+  const bool roll_auto =
+      rollov == 0 && V > 1 && m * (int64_t)sizeof(T) >= longrow;
   static int rolltgt = [] {
     const char* e = getenv("PAM_FD_ROLL_TGT");
-    return e ? atoi(e) : 4096;
+    return e ? atoi(e) : 2048;
   }();
-  if (rollov && V > 1) {
+  if ((rollov > 0 && V > 1) || roll_auto) {
+    // upgrade to 32-B lanes for the rolling form (its swept optimum;
+    // the row-parallel kernel's optimum stays 16 B)
+    if (roll_auto && fd_vec_override() == 0 && m % 4 == 0 && align16)
+      V = 4;
     static int cvov = [] {
       const char* e = getenv("PAM_FD_ROLL_CV");
-      return e ? atoi(e) : 4;
+      return e ? atoi(e) : 8;
     }();
-    const int CV = (cvov == 2 || cvov == 8) ? cvov : 4;
+    const int CV = (cvov == 2 || cvov == 4) ? cvov : 8;
     const int64_t mv2 = m / V;
     int64_t gx = (mv2 + (int64_t)BLK * CV - 1) / ((int64_t)BLK * CV);
     int64_t gyr = rolltgt / (gx ? gx : 1);
