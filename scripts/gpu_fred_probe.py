@@ -30,7 +30,7 @@ def main():
          + 1j * (torch.rand((nf, ns, nr), generator=g, device="cuda") - 0.5)
          ).to(torch.complex64)
     for saveGt in (False, True):
-        op = pm.MPIFredholm1(G, nv, saveGt=saveGt)
+        op = pm.MPIFredholm1(G, nv, saveGt=saveGt, dtype="complex64")
         x = pm.DistributedArray((op.shape[1],),
                                 partition=pm.Partition.BROADCAST,
                                 dtype=np.complex64)
