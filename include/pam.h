@@ -236,6 +236,15 @@ int pam_cgemm_batched(void* stream, const void* A, const void* B, void* C,
                       int64_t strideA, int64_t strideB, int64_t strideC,
                       int opa, int accumulate, int dtype);
 
+/* Batched REAL GEMM: the single-plane analogue of pam_cgemm_batched
+ * (op = N or transpose; accumulate != 0 adds into C).  The Fredholm1
+ * path for float32/float64 kernels (ref Fredholm1.py:123 on a real G)
+ * — one z-batched launch instead of a per-slice host loop. */
+int pam_gemm_batched(void* stream, const void* A, const void* B, void* C,
+                     int64_t batch, int64_t M, int64_t N, int64_t K,
+                     int64_t strideA, int64_t strideB, int64_t strideC,
+                     int opa, int accumulate, int dtype);
+
 /* Complex (conj-)transpose on interleaved (re,im) pairs: At = A^T
  * (conj=0) or A^H (conj=1) — the complex MatrixMult adjoint panels
  * (ref MatrixMult.py:416,737 `A.T.conj()`). */

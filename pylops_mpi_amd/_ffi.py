@@ -75,6 +75,11 @@ _SIGS = {
     "pam_ctranspose": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                         ctypes.c_int64, ctypes.c_int64, ctypes.c_int,
                         ctypes.c_int], ctypes.c_int),
+    "pam_gemm_batched": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                          ctypes.c_void_p, ctypes.c_int64, ctypes.c_int64,
+                          ctypes.c_int64, ctypes.c_int64, ctypes.c_int64,
+                          ctypes.c_int64, ctypes.c_int64, ctypes.c_int,
+                          ctypes.c_int, ctypes.c_int], ctypes.c_int),
     "pam_norm_local": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64,
                         ctypes.c_int, ctypes.c_double, ctypes.c_void_p,
                         ctypes.c_void_p, ctypes.c_int], ctypes.c_int),

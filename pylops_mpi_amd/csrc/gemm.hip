@@ -583,6 +583,204 @@ extern "C" int pam_cgemm_batched(void* stream, const void* A, const void* B,
 }
 
 // ---------------------------------------------------------------------------
+// batched REAL GEMM: C_b (+)= op(A_b) @ B_b over blockIdx.z, the
+// single-plane analogue of cgemm_batched_kernel.  The Fredholm path for
+// float32/float64 kernels (ref Fredholm1.py:123 `ncp.matmul(G, x)` on a
+// 3-D G) previously looped pam_gemm per slice from Python — 513 launches
+// of 4 workgroups each at the cfg5 shape (chip empty + ctypes overhead);
+// the z-batched grid fills the chip like the complex path does.
+// ---------------------------------------------------------------------------
+template <typename T, bool CT, bool ACC, int BK = 16, int BM = 64,
+          int BN = 64>
+__global__ void __launch_bounds__(GBLK) gemm_batched_kernel(
+    const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
+    int64_t M, int64_t N, int64_t K, int64_t strideA, int64_t strideB,
+    int64_t strideC) {
+  using CFG = GemmCfg<T>;
+  constexpr int TM = CFG::TM, TK = CFG::TK;
+  constexpr int NBUFC = (BK >= 64) ? 1 : 2;
+  constexpr int MI = (BM / 2) / TM, NJ = (BN / 2) / TM;
+  using acc_t = typename CFG::acc_t;
+
+  __shared__ T As[NBUFC][BK][BM + 1];  // k-major
+  __shared__ T Bs[NBUFC][BK][BN + 1];
+  const int64_t b = blockIdx.z;
+  const T* __restrict__ Ab = A + b * strideA;
+  const T* __restrict__ Bb = B + b * strideB;
+  T* __restrict__ Cb = C + b * strideC;
+  const int64_t m0 = (int64_t)blockIdx.y * BM;
+  const int64_t n0 = (int64_t)blockIdx.x * BN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int li = lane & (TM - 1);
+  const int lk = lane / TM;
+
+  acc_t s[MI][NJ];
+#pragma unroll
+  for (int i = 0; i < MI; ++i)
+#pragma unroll
+    for (int j = 0; j < NJ; ++j) s[i][j] = {};
+
+  constexpr int AE = (BM * BK) / GBLK;
+  constexpr int BE = (BK * BN) / GBLK;
+  T arg[AE], brg[BE];
+
+  auto load_panel = [&](int64_t k0) {
+#pragma unroll
+    for (int e = 0; e < AE; ++e) {
+      const int vi2 = tid + GBLK * e;
+      int kk, mm;
+      if constexpr (CT) {  // op(A)[m][k] = A[k][m] (real: no conj)
+        mm = vi2 & (BM - 1);
+        kk = vi2 / BM;
+      } else {
+        kk = vi2 & (BK - 1);
+        mm = vi2 / BK;
+      }
+      const int64_t gm = m0 + mm;
+      const int64_t gk = k0 + kk;
+      T v = 0;
+      if (gm < M && gk < K)
+        v = CT ? Ab[gk * M + gm] : Ab[gm * K + gk];
+      arg[e] = v;
+    }
+#pragma unroll
+    for (int e = 0; e < BE; ++e) {
+      const int vi2 = tid + GBLK * e;
+      const int nn = vi2 & (BN - 1);
+      const int kk = vi2 / BN;
+      const int64_t gk = k0 + kk;
+      const int64_t gn = n0 + nn;
+      brg[e] = (gk < K && gn < N) ? Bb[gk * N + gn] : (T)0;
+    }
+  };
+
+  auto store_panel = [&](int buf) {
+#pragma unroll
+    for (int e = 0; e < AE; ++e) {
+      const int vi2 = tid + GBLK * e;
+      int kk, mm;
+      if constexpr (CT) {
+        mm = vi2 & (BM - 1);
+        kk = vi2 / BM;
+      } else {
+        kk = vi2 & (BK - 1);
+        mm = vi2 / BK;
+      }
+      As[buf][kk][mm] = arg[e];
+    }
+#pragma unroll
+    for (int e = 0; e < BE; ++e) {
+      const int vi2 = tid + GBLK * e;
+      const int nn = vi2 & (BN - 1);
+      const int kk = vi2 / BN;
+      Bs[buf][kk][nn] = brg[e];
+    }
+  };
+
+  const int64_t NP = (K + BK - 1) / BK;
+  if (NP > 0) {
+    load_panel(0);
+    store_panel(0);
+    __syncthreads();
+    for (int64_t p = 0; p < NP; ++p) {
+      if (NBUFC > 1 && p + 1 < NP) load_panel((p + 1) * BK);
+      const int cur = (int)(p % NBUFC);
+      T a[2][MI], bb[2][NJ];
+#pragma unroll
+      for (int mi = 0; mi < MI; ++mi)
+        a[0][mi] = As[cur][lk][wr * (BM / 2) + mi * TM + li];
+#pragma unroll
+      for (int nj = 0; nj < NJ; ++nj)
+        bb[0][nj] = Bs[cur][lk][wc * (BN / 2) + nj * TM + li];
+#pragma unroll
+      for (int kk = 0; kk < BK / TK; ++kk) {
+        const int cf = kk & 1, nf = (kk + 1) & 1;
+        if (kk + 1 < BK / TK) {
+          const int krow = (kk + 1) * TK + lk;
+#pragma unroll
+          for (int mi = 0; mi < MI; ++mi)
+            a[nf][mi] = As[cur][krow][wr * (BM / 2) + mi * TM + li];
+#pragma unroll
+          for (int nj = 0; nj < NJ; ++nj)
+            bb[nf][nj] = Bs[cur][krow][wc * (BN / 2) + nj * TM + li];
+        }
+#pragma unroll
+        for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+          for (int nj = 0; nj < NJ; ++nj)
+            s[mi][nj] = CFG::mfma(a[cf][mi], bb[cf][nj], s[mi][nj]);
+      }
+      if (p + 1 < NP) {
+        if (NBUFC == 1) {
+          __syncthreads();
+          load_panel((p + 1) * BK);
+        }
+        store_panel((int)((p + 1) % NBUFC));
+        __syncthreads();
+      }
+    }
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int nj = 0; nj < NJ; ++nj) {
+      const int64_t r0 = m0 + wr * (BM / 2) + mi * TM;
+      const int64_t cc = n0 + wc * (BN / 2) + nj * TM + li;
+      if (cc >= N) continue;
+#pragma unroll
+      for (int reg = 0; reg < CFG::NREG; ++reg) {
+        const int64_t rr = r0 + CFG::crow(lane, reg);
+        if (rr < M) {
+          if constexpr (ACC)
+            Cb[rr * N + cc] += s[mi][nj][reg];
+          else
+            Cb[rr * N + cc] = s[mi][nj][reg];
+        }
+      }
+    }
+}
+
+template <typename T>
+static int rgemm_launch(void* stream, const void* A, const void* B, void* C,
+                        int64_t batch, int64_t M, int64_t N, int64_t K,
+                        int64_t sA, int64_t sB, int64_t sC, int opa,
+                        int acc) {
+  if (batch <= 0 || M <= 0 || N <= 0 || K < 0 || !A || !B || !C)
+    return PAM_EARG;
+  dim3 grid((uint32_t)((N + 63) / 64), (uint32_t)((M + 63) / 64),
+            (uint32_t)batch);
+  hipStream_t s = (hipStream_t)stream;
+#define RG_LAUNCH(CTV, ACCV)                                                  \
+  hipLaunchKernelGGL((gemm_batched_kernel<T, CTV, ACCV>), grid, dim3(GBLK),   \
+                     0, s, (const T*)A, (const T*)B, (T*)C, M, N, K, sA, sB, \
+                     sC)
+  if (opa && acc) RG_LAUNCH(true, true);
+  else if (opa) RG_LAUNCH(true, false);
+  else if (acc) RG_LAUNCH(false, true);
+  else RG_LAUNCH(false, false);
+#undef RG_LAUNCH
+  return gcheck(hipGetLastError());
+}
+
+extern "C" int pam_gemm_batched(void* stream, const void* A, const void* B,
+                                void* C, int64_t batch, int64_t M, int64_t N,
+                                int64_t K, int64_t strideA, int64_t strideB,
+                                int64_t strideC, int opa, int accumulate,
+                                int dtype) {
+  if (dtype == PAM_F64)
+    return rgemm_launch<double>(stream, A, B, C, batch, M, N, K, strideA,
+                                strideB, strideC, opa, accumulate);
+  if (dtype == PAM_F32)
+    return rgemm_launch<float>(stream, A, B, C, batch, M, N, K, strideA,
+                               strideB, strideC, opa, accumulate);
+  return PAM_EDTYPE;
+}
+
+// ---------------------------------------------------------------------------
 // complex (conj-)transpose: At[c][r] = (conj?)(A[r][c]) on interleaved
 // (re,im) pairs — materializes A^H for the complex MatrixMult adjoint
 // panels (ref MatrixMult.py:416,737 "A.T.conj()").  Same 32x32 LDS tile

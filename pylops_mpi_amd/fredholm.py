@@ -70,13 +70,14 @@ class MPIFredholm1(MPILinearOperator):
                 M, N, K, A.shape[1] * A.shape[2], K * N, M * N, opa, 0,
                 _ffi.dtype_code(A.dtype)), "cgemm_batched")
         else:
-            # real dtypes: per-slice MFMA panels (small batch loop)
-            for b in range(batch):
-                Ab = A[b].t().conj().contiguous() if opa else A[b]
-                _ffi.checked(_ffi.lib().pam_gemm(
-                    _stream(A), Ab.contiguous().data_ptr(),
-                    X[b].data_ptr(), Y[b].data_ptr(), M, N, K, K, N, N, 0,
-                    _ffi.dtype_code(A.dtype)), "gemm")
+            # real dtypes: one z-batched MFMA launch (opa=1 loads A
+            # transposed in-kernel; conj is a no-op on reals).  The r01
+            # form looped pam_gemm per slice from Python — 4-workgroup
+            # launches, chip empty.
+            _ffi.checked(_ffi.lib().pam_gemm_batched(
+                _stream(A), A.data_ptr(), X.data_ptr(), Y.data_ptr(),
+                batch, M, N, K, A.shape[1] * A.shape[2], K * N, M * N,
+                opa, 0, _ffi.dtype_code(A.dtype)), "gemm_batched")
         return Y
 
     # ------------------------------------------------------------- applies

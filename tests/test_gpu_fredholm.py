@@ -231,3 +231,58 @@ def test_cgls_complex_operator():
     # 30 complex CGLS iterations amplify backend fp-order differences;
     # measured divergence ~4e-8 relative
     assert_allclose(host(xinv.asarray()), xref, rtol=1e-6, atol=1e-9)
+
+
+@pytest.mark.parametrize("dtype,tol", [(np.float64, 1e-12),
+                                       (np.float32, 2e-4)])
+def test_gemm_batched_real(dtype, tol):
+    """pam_gemm_batched: z-batched real MFMA panels (the r02 replacement
+    for the per-slice host loop on real Fredholm kernels)."""
+    rng = np.random.default_rng(11)
+    B_, M, K, N = 5, 48, 33, 40
+    A = rng.standard_normal((B_, M, K)).astype(dtype)
+    X = rng.standard_normal((B_, K, N)).astype(dtype)
+    s = torch.cuda.current_stream().cuda_stream
+    Ad, Xd = dev(A).contiguous(), dev(X).contiguous()
+    Y = torch.empty((B_, M, N), dtype=Ad.dtype, device="cuda:0")
+    _ffi.checked(_ffi.lib().pam_gemm_batched(
+        s, Ad.data_ptr(), Xd.data_ptr(), Y.data_ptr(), B_, M, N, K,
+        M * K, K * N, M * N, 0, 0, _ffi.dtype_code(Ad.dtype)), "rgb")
+    assert_allclose(host(Y), A @ X, rtol=tol, atol=tol)
+    # transpose op + accumulate
+    X2 = rng.standard_normal((B_, M, N)).astype(dtype)
+    C0 = rng.standard_normal((B_, K, N)).astype(dtype)
+    X2d, Cd = dev(X2).contiguous(), dev(C0).contiguous()
+    _ffi.checked(_ffi.lib().pam_gemm_batched(
+        s, Ad.data_ptr(), X2d.data_ptr(), Cd.data_ptr(), B_, K, N, M,
+        M * K, M * N, K * N, 1, 1, _ffi.dtype_code(Ad.dtype)), "rgb")
+    assert_allclose(host(Cd), C0 + A.transpose(0, 2, 1) @ X2,
+                    rtol=tol, atol=tol * 10)
+
+
+def test_fredholm_real_dtypes_vs_oracle():
+    """Real-G Fredholm (ref test_fredholm.py float32 params) through the
+    z-batched path."""
+    import oracle
+    from oracle.ranksim import Partition as SP, SimArray
+    rng = np.random.default_rng(12)
+    nsl, nx, ny, nz = 21, 4, 6, 5
+    for dt, tol in ((np.float64, 1e-12), (np.float32, 2e-4)):
+        G = rng.standard_normal((nsl, nx, ny)).astype(dt)
+        for sg in (False, True):
+            op = pm.MPIFredholm1(dev(G), nz, saveGt=sg, dtype=dt)
+            sop = oracle.SimFredholm1([G.copy()], nz=nz, saveGt=sg)
+            x = rng.standard_normal(op.shape[1]).astype(dt)
+            y = rng.standard_normal(op.shape[0]).astype(dt)
+            xd = pm.DistributedArray.to_dist(
+                dev(x), partition=pm.Partition.BROADCAST)
+            yd = pm.DistributedArray.to_dist(
+                dev(y), partition=pm.Partition.BROADCAST)
+            got = host(op.matvec(xd).local_array)
+            want = sop.matvec(
+                SimArray([x], x.shape, partition=SP.BROADCAST)).locals[0]
+            assert_allclose(got, want, rtol=tol, atol=tol)
+            gotr = host(op.rmatvec(yd).local_array)
+            wantr = sop.rmatvec(
+                SimArray([y], y.shape, partition=SP.BROADCAST)).locals[0]
+            assert_allclose(gotr, wantr, rtol=tol, atol=tol)
