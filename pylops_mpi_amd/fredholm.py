@@ -91,8 +91,6 @@ class MPIFredholm1(MPILinearOperator):
                 f"Got  {x.partition} instead...")
         comm = x.base_comm
         gshape = self.shape[0] if forward else self.shape[1]
-        y = DistributedArray(int(gshape), comm, x.partition,
-                             dtype=self.dtype)
         dims = self.dimsd if not forward else self.dims
         xl = x.local_array.reshape(dims)
         r = comm.rank
@@ -108,8 +106,14 @@ class MPIFredholm1(MPILinearOperator):
         tiles = comm.allgather_tensors(
             y1.reshape(-1),
             [(int(n) * nmid * self.nz,) for n in self.nsls])  # ref :129,167
-        y[:] = torch.cat(tiles)
-        return y
+        # wrap the gathered tensor as the output's storage directly — it
+        # is freshly allocated here, so no defensive copy is needed (the
+        # r01 form cat'ed into a separate DistributedArray: two extra
+        # full passes per apply)
+        full = tiles[0] if len(tiles) == 1 else torch.cat(tiles)
+        return DistributedArray(int(gshape), comm, x.partition,
+                                local_array=full.reshape(-1),
+                                dtype=self.dtype)
 
     def _matvec(self, x: DistributedArray) -> DistributedArray:
         return self._apply(x, forward=True)
