@@ -306,7 +306,7 @@ extern "C" int pam_gemm(void* stream, const void* A, const void* B, void* C,
 // way, but only MFMA reaches the matrix-core rate.
 // ---------------------------------------------------------------------------
 template <typename T, bool CT, bool ACC = false, int BK = 16,
-          int BM = 64, int BN = 64>
+          int BM = 64, int BN = 64, int NB = (BK >= 64 ? 1 : 2)>
 __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
     const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
     int64_t M, int64_t N, int64_t K, int64_t strideA, int64_t strideB,
@@ -321,7 +321,7 @@ __global__ void __launch_bounds__(GBLK) cgemm_batched_kernel(
   // BM/BN default 64x64; 128x64 is dispatched for the shallow-K
   // adjoint panels (cfg5 rmatvec K=64: 4x the work per block at the
   // same 4-panel pipeline depth — r02 A/B)
-  constexpr int NBUFC = (BK >= 64) ? 1 : 2;
+  constexpr int NBUFC = NB;
   constexpr int MI = (BM / 2) / TM, NJ = (BN / 2) / TM;
   using acc_t = typename CFG::acc_t;
 
@@ -549,6 +549,39 @@ static int cgemm_launch(void* stream, const void* A, const void* B, void* C,
   // was A/B'd NEGATIVE at the cfg5 rmatvec shape (0.432 -> 0.518 ms):
   // the 133 KB LDS footprint drops occupancy to 2 WG/CU, which costs
   // more than the three extra barriers it saves.  BK=16 for all K.
+  // pipeline A/B knobs for the shallow-K adjoint (cfg5 rmatvec):
+  // PAM_CGEMM_NBUF=1 single-buffers the LDS panels (halves LDS,
+  // doubles resident WGs); PAM_CGEMM_BK=8 halves the panel depth.
+  static int nbov = [] {
+    const char* e = getenv("PAM_CGEMM_NBUF");
+    return e ? atoi(e) : 0;
+  }();
+  static int bkov = [] {
+    const char* e = getenv("PAM_CGEMM_BK");
+    return e ? atoi(e) : 0;
+  }();
+  if (!acc && nbov == 1) {
+    if (opa)
+      hipLaunchKernelGGL((cgemm_batched_kernel<T, true, false, 16, 64, 64,
+                          1>), grid, dim3(GBLK), 0, s, (const T*)A,
+                         (const T*)B, (T*)C, M, N, K, sA, sB, sC);
+    else
+      hipLaunchKernelGGL((cgemm_batched_kernel<T, false, false, 16, 64, 64,
+                          1>), grid, dim3(GBLK), 0, s, (const T*)A,
+                         (const T*)B, (T*)C, M, N, K, sA, sB, sC);
+    return gcheck(hipGetLastError());
+  }
+  if (!acc && bkov == 8) {
+    if (opa)
+      hipLaunchKernelGGL((cgemm_batched_kernel<T, true, false, 8>), grid,
+                         dim3(GBLK), 0, s, (const T*)A, (const T*)B, (T*)C,
+                         M, N, K, sA, sB, sC);
+    else
+      hipLaunchKernelGGL((cgemm_batched_kernel<T, false, false, 8>), grid,
+                         dim3(GBLK), 0, s, (const T*)A, (const T*)B, (T*)C,
+                         M, N, K, sA, sB, sC);
+    return gcheck(hipGetLastError());
+  }
   if (opa && acc)
     hipLaunchKernelGGL((cgemm_batched_kernel<T, true, true>), grid,
                        dim3(GBLK), 0, s, (const T*)A, (const T*)B, (T*)C, M,
