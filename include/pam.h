@@ -245,6 +245,15 @@ int pam_gemm_batched(void* stream, const void* A, const void* B, void* C,
                      int64_t strideA, int64_t strideB, int64_t strideC,
                      int opa, int accumulate, int dtype);
 
+/* Complex <-> real (de)interleave (the MDC chain's real extraction and
+ * complex output carrier, ref waveeqprocessing/MDC.py:55-69): unzip
+ * writes the n real parts of an interleaved complex array; zip writes n
+ * (re, 0) pairs from a real array.  dtype names the COMPLEX type. */
+int pam_unzip(void* stream, void* dst_real, const void* src_cplx, int64_t n,
+              int dtype);
+int pam_zip(void* stream, void* dst_cplx, const void* src_real, int64_t n,
+            int dtype);
+
 /* Complex (conj-)transpose on interleaved (re,im) pairs: At = A^T
  * (conj=0) or A^H (conj=1) — the complex MatrixMult adjoint panels
  * (ref MatrixMult.py:416,737 `A.T.conj()`). */

@@ -75,6 +75,10 @@ _SIGS = {
     "pam_ctranspose": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                         ctypes.c_int64, ctypes.c_int64, ctypes.c_int,
                         ctypes.c_int], ctypes.c_int),
+    "pam_unzip": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                   ctypes.c_int64, ctypes.c_int], ctypes.c_int),
+    "pam_zip": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                 ctypes.c_int64, ctypes.c_int], ctypes.c_int),
     "pam_gemm_batched": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                           ctypes.c_void_p, ctypes.c_int64, ctypes.c_int64,
                           ctypes.c_int64, ctypes.c_int64, ctypes.c_int64,

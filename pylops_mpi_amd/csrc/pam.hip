@@ -334,6 +334,103 @@ EWC_ENTRY(pam_conj(void* stream, void* y, const void* a, int64_t n,
           2, nullptr, 0.0, 0.0)
 
 // ---------------------------------------------------------------------------
+// complex <-> real (de)interleave: the MDC chain's real extraction
+// feeding rfft (ref MDC.py:55-69 — pylops' real FFT discards imag) and
+// the real->complex(0 imag) output carrier.  torch's strided
+// elementwise copies run these at ~3.7 TB/s (r02 MDC kernel trace); the
+// vectorized forms below stream at the plain copy rate.  unzip reads V
+// complex pairs per lane and writes V reals; zip the reverse.
+// ---------------------------------------------------------------------------
+template <typename T, int V>
+__global__ void __launch_bounds__(BLK) unzip_kernel(
+    T* __restrict__ dst, const T* __restrict__ src, int64_t n) {
+  const int64_t nv = n / V;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t iv = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; iv < nv;
+       iv += stride) {
+    T pair[2 * V], out[V];
+    loadv<T, V>(src + 2 * iv * V, pair);
+    loadv<T, V>(src + 2 * iv * V + V, pair + V);
+#pragma unroll
+    for (int k = 0; k < V; ++k) out[k] = pair[2 * k];
+    storev<T, V>(dst + iv * V, out);
+  }
+  const int64_t gid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (int64_t i = nv * V + gid; i < n; i += stride)
+    dst[i] = src[2 * i];
+}
+
+template <typename T, int V>
+__global__ void __launch_bounds__(BLK) zip_kernel(
+    T* __restrict__ dst, const T* __restrict__ src, int64_t n) {
+  const int64_t nv = n / V;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t iv = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; iv < nv;
+       iv += stride) {
+    T in[V], pair[2 * V];
+    loadv<T, V>(src + iv * V, in);
+#pragma unroll
+    for (int k = 0; k < V; ++k) {
+      pair[2 * k] = in[k];
+      pair[2 * k + 1] = (T)0;
+    }
+    storev<T, V>(dst + 2 * iv * V, pair);
+    storev<T, V>(dst + 2 * iv * V + V, pair + V);
+  }
+  const int64_t gid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (int64_t i = nv * V + gid; i < n; i += stride) {
+    dst[2 * i] = src[i];
+    dst[2 * i + 1] = (T)0;
+  }
+}
+
+template <typename T>
+static int zip_launch(void* stream, void* dst, const void* src, int64_t n,
+                      bool unzip) {
+  if (n < 0 || !dst || !src) return PAM_EARG;
+  if (n == 0) return 0;
+  constexpr int V = VecW<T>::value;
+  const bool aligned = ((uintptr_t)dst % 16 == 0) &&
+                       ((uintptr_t)src % 16 == 0);
+  hipStream_t s = (hipStream_t)stream;
+  const int64_t grid = grid_1d(n / (aligned ? V : 1) + 1);
+  if (unzip) {
+    if (aligned)
+      hipLaunchKernelGGL((unzip_kernel<T, V>), dim3(grid), dim3(BLK), 0, s,
+                         (T*)dst, (const T*)src, n);
+    else
+      hipLaunchKernelGGL((unzip_kernel<T, 1>), dim3(grid), dim3(BLK), 0, s,
+                         (T*)dst, (const T*)src, n);
+  } else {
+    if (aligned)
+      hipLaunchKernelGGL((zip_kernel<T, V>), dim3(grid), dim3(BLK), 0, s,
+                         (T*)dst, (const T*)src, n);
+    else
+      hipLaunchKernelGGL((zip_kernel<T, 1>), dim3(grid), dim3(BLK), 0, s,
+                         (T*)dst, (const T*)src, n);
+  }
+  return check(hipGetLastError());
+}
+
+extern "C" int pam_unzip(void* stream, void* dst_real, const void* src_cplx,
+                         int64_t n, int dtype) {
+  if (dtype == PAM_C128)
+    return zip_launch<double>(stream, dst_real, src_cplx, n, true);
+  if (dtype == PAM_C64)
+    return zip_launch<float>(stream, dst_real, src_cplx, n, true);
+  return PAM_EDTYPE;
+}
+
+extern "C" int pam_zip(void* stream, void* dst_cplx, const void* src_real,
+                       int64_t n, int dtype) {
+  if (dtype == PAM_C128)
+    return zip_launch<double>(stream, dst_cplx, src_real, n, false);
+  if (dtype == PAM_C64)
+    return zip_launch<float>(stream, dst_cplx, src_real, n, false);
+  return PAM_EDTYPE;
+}
+
+// ---------------------------------------------------------------------------
 // thresholding for ISTA/FISTA (the pylops _softthreshold/_hardthreshold
 // formulas the reference imports, ref optimization/cls_sparsity.py:10):
 //   soft real   : sign(x) * max(|x|-t, 0)

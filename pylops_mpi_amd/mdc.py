@@ -98,10 +98,19 @@ class _FusedMDC(MPILinearOperator):
                  shift: bool) -> torch.Tensor:
         t = x.local_array.reshape(self.nt, nmid * self.nv)
         if t.is_complex():
-            # contiguous real extraction: rfft on the stride-2 .real
-            # view forces a strided rocFFT plan (r01 DESIGN's named
-            # remaining cost); one coalesced copy is cheaper
-            t = t.real.contiguous()
+            # real extraction on the pam_unzip kernel: torch's strided
+            # .real copy measures ~3.7 TB/s (r02 MDC kernel trace,
+            # profiles/r02_mdc_kernel_stats.csv); the vectorized
+            # deinterleave streams at the copy rate
+            from . import _ffi
+            r = torch.empty(t.shape, device=t.device,
+                            dtype=torch.float32 if t.dtype == torch.complex64
+                            else torch.float64)
+            _ffi.checked(_ffi.lib().pam_unzip(
+                torch.cuda.current_stream(t.device).cuda_stream,
+                r.data_ptr(), t.contiguous().data_ptr(), t.numel(),
+                _ffi.dtype_code(t.dtype)), "unzip")
+            t = r
         if shift:
             t = torch.fft.ifftshift(t, dim=0)
         # NOTE the sqrt(2) conjugate-twin scaling of the pylops real-FFT
@@ -132,18 +141,30 @@ class _FusedMDC(MPILinearOperator):
     _CT = {np.dtype(np.complex64): torch.complex64,
            np.dtype(np.complex128): torch.complex128}
 
+    def _to_cplx(self, out: torch.Tensor) -> torch.Tensor:
+        """real -> complex-with-zero-imag carrier on the pam_zip kernel
+        (the composite chain's stage wrappers carry complex storage end
+        to end — mirror that; torch's .to(complex) cast is the strided
+        half-rate path)."""
+        from . import _ffi
+        z = torch.empty(out.shape, device=out.device,
+                        dtype=self._CT[self.cdtype])
+        _ffi.checked(_ffi.lib().pam_zip(
+            torch.cuda.current_stream(out.device).cuda_stream,
+            z.data_ptr(), out.contiguous().data_ptr(), out.numel(),
+            _ffi.dtype_code(z.dtype)), "zip")
+        return z
+
     def _matvec(self, x: DistributedArray) -> DistributedArray:
         f = self._fwd_fft(x, self.nr, self.twosided)       # Fop
         f = f[: self.nfreq]                                # Iop
         fr = self.Frop.matvec(self._wrap(f.contiguous()))  # Fredholm
         out = self._inv_fft(fr.local_array, self.ns, False)  # I1^H F1^H
-        # the composite chain's stage wrappers carry complex storage
-        # (real values, imag 0) end to end — mirror that
-        return self._wrap(out.to(self._CT[self.cdtype]))
+        return self._wrap(self._to_cplx(out))
 
     def _rmatvec(self, y: DistributedArray) -> DistributedArray:
         f = self._fwd_fft(y, self.ns, False)               # F1op, I1op
         f = f[: self.nfreq]
         fr = self.Frop.rmatvec(self._wrap(f.contiguous()))
         out = self._inv_fft(fr.local_array, self.nr, self.twosided)
-        return self._wrap(out.to(self._CT[self.cdtype]))
+        return self._wrap(self._to_cplx(out))

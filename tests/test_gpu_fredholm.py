@@ -286,3 +286,25 @@ def test_fredholm_real_dtypes_vs_oracle():
             wantr = sop.rmatvec(
                 SimArray([y], y.shape, partition=SP.BROADCAST)).locals[0]
             assert_allclose(gotr, wantr, rtol=tol, atol=tol)
+
+
+@pytest.mark.parametrize("dtype,tol", [(np.complex128, 0),
+                                       (np.complex64, 0)])
+def test_zip_unzip(dtype, tol):
+    """pam_unzip/pam_zip: complex<->real (de)interleave (MDC chain)."""
+    rng = np.random.default_rng(13)
+    for n in (1024, 999, 7, 65536):
+        a = crand(rng, n, dtype)
+        ad = dev(a).contiguous()
+        rt = torch.float32 if dtype == np.complex64 else torch.float64
+        r = torch.empty(n, dtype=rt, device="cuda:0")
+        s = torch.cuda.current_stream().cuda_stream
+        _ffi.checked(_ffi.lib().pam_unzip(
+            s, r.data_ptr(), ad.data_ptr(), n, _ffi.dtype_code(ad.dtype)),
+            "unzip")
+        assert np.array_equal(host(r), a.real)
+        z = torch.empty(n, dtype=ad.dtype, device="cuda:0")
+        _ffi.checked(_ffi.lib().pam_zip(
+            s, z.data_ptr(), r.data_ptr(), n, _ffi.dtype_code(ad.dtype)),
+            "zip")
+        assert np.array_equal(host(z), a.real.astype(dtype))
