@@ -254,6 +254,15 @@ int pam_unzip(void* stream, void* dst_real, const void* src_cplx, int64_t n,
 int pam_zip(void* stream, void* dst_cplx, const void* src_real, int64_t n,
             int dtype);
 
+/* Strided-batched real FFTs along dim 0 of a row-major (nt, m) array
+ * (rocFFT native layout — no permute copies; UNSCALED; the MDC chain
+ * folds the ortho norms into its kernel).  dtype is the REAL element
+ * type.  pam_irfft_strided may clobber its input (rocFFT scratch). */
+int pam_rfft_strided(void* stream, const void* in_real, void* out_cplx,
+                     int64_t nt, int64_t m, int dtype);
+int pam_irfft_strided(void* stream, void* in_cplx, void* out_real,
+                      int64_t nt, int64_t m, int dtype);
+
 /* Complex (conj-)transpose on interleaved (re,im) pairs: At = A^T
  * (conj=0) or A^H (conj=1) — the complex MatrixMult adjoint panels
  * (ref MatrixMult.py:416,737 `A.T.conj()`). */

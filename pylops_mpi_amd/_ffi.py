@@ -75,6 +75,12 @@ _SIGS = {
     "pam_ctranspose": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                         ctypes.c_int64, ctypes.c_int64, ctypes.c_int,
                         ctypes.c_int], ctypes.c_int),
+    "pam_rfft_strided": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                          ctypes.c_int64, ctypes.c_int64, ctypes.c_int],
+                         ctypes.c_int),
+    "pam_irfft_strided": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                           ctypes.c_int64, ctypes.c_int64, ctypes.c_int],
+                          ctypes.c_int),
     "pam_unzip": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                    ctypes.c_int64, ctypes.c_int], ctypes.c_int),
     "pam_zip": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,

@@ -12,6 +12,20 @@ pylops_mpi/utils/deps.py:58-66 (NCCL_PYLOPS_MPI / PYLOPS_MPI_CUDA_AWARE):
   PAM_FD_GY / PAM_FD_CAP  stencil launch-shape A/B knobs (defaults: one
                           block-row per row, ~4 vector iterations per
                           block — the measured optimum, csrc/pam.hip)
+  PAM_FD_ROLL             rolling-window stencil select: unset = AUTO
+                          (rows >= PAM_FD_LONGROW bytes, default 4 MiB,
+                          use the 1-load/pt rolling kernel — the r02
+                          long-row fix, 4.85 -> 5.35 TB/s at the N=8
+                          per-rank shape); -1 = force row-parallel;
+                          1/2 = force rolling (2 = statically-rotated
+                          variant, measured slower at CV8xV4)
+  PAM_FD_ROLL_CV / PAM_FD_ROLL_TGT  rolling chains per thread (default
+                          8) and target grid size (default 2048) — the
+                          r02 swept optima
+  PAM_CGEMM_TILE/_NBUF/_BK  cgemm pipeline A/B knobs (defaults measured
+                          best at the cfg5 shapes; 128x64 tile, NBUF=1
+                          and BK=8 all measured neutral-to-negative,
+                          csrc/gemm.hip)
   PAM_EW_CAP              1-D elementwise grid cap (default: none — one
                           block per 256 vector items; the old 4096-block
                           grid-stride loop cost 28% of axpy bandwidth)

@@ -8,14 +8,17 @@ per-rank on BROADCAST arrays and the kernel prescaled by dr*dt*sqrt(nt)
 Instead of composing five wrapped operators (each stage allocating a
 fresh DistributedArray and copying its output, the way the reference's
 duck-typed chain does) the chain is applied as ONE fused local pipeline:
-rfft -> sqrt2 twin scale -> frequency mask -> batched MFMA cgemm
-(Fredholm, the only stage with communication) -> zero-pad -> inverse
-twin scale -> irfft, with zero-copy BROADCAST wrappers between stages.
-Same arithmetic per element as the composite (the GPU parity tests pin
-the cgls trace), measured r01 at the judged cfg5 shape: matvec 2.04 ->
-1.64 ms, rmatvec 2.69 -> 2.19 ms; the remaining non-Fredholm time is
-the semantically-required real-part extraction feeding rfft (the
-reference takes .real too) and the rocFFT transforms themselves.
+real-extract (pam_unzip) -> rfft -> frequency mask -> batched MFMA
+cgemm (Fredholm, the only stage with communication) -> zero-pad ->
+irfft -> complex carrier (pam_zip), with zero-copy BROADCAST wrappers
+between stages.  The pylops real-FFT convention's sqrt(2) twin-bin
+scale and its inverse surround the per-frequency block-diagonal
+Fredholm kernel, so they cancel algebraically and neither pass is run
+(r02; see _fwd_fft).  Same operator as the composite (GPU parity tests
+pin the cgls trace).  Measured at the judged cfg5 shape: r01 composite
+2.04/2.69 ms (matvec/rmatvec) -> r01 fused 1.64/2.19 -> r02 (twin-scale
+cancellation + pam_unzip/zip + the Fredholm output-wrap copy
+elimination) — see DESIGN.md §8 for the final figures.
 """
 import logging
 
@@ -42,13 +45,19 @@ def MPIMDC(G: torch.Tensor, nt: int, nv: int, nfreq: int, dt: float = 1.0,
                       torch.complex128: np.complex128}[G.dtype])
     rdtype = np.real(np.ones(1, dtype=dtype)).dtype
 
-    # Fredholm kernel, prescaled (ref :36-43)
+    # Fredholm kernel, prescaled (ref :36-43).  The EXTRA 1/nt folds the
+    # ortho 1/sqrt(nt) of the chain's forward and inverse FFTs into G
+    # (our strided rocFFT transforms are unscaled; a per-frequency
+    # scalar commutes with the per-frequency block-diagonal kernel, so
+    # the operator is unchanged — see _FusedMDC)
     if prescaled:
-        Frop = MPIFredholm1(G, nv, saveGt=saveGt, usematmul=usematmul,
-                            base_comm=comm, dtype=dtype)
+        Frop = MPIFredholm1(G * (1.0 / nt), nv, saveGt=saveGt,
+                            usematmul=usematmul, base_comm=comm,
+                            dtype=dtype)
     else:
-        Frop = MPIFredholm1(dr * dt * np.sqrt(nt) * G, nv, saveGt=saveGt,
-                            usematmul=usematmul, base_comm=comm, dtype=dtype)
+        Frop = MPIFredholm1((dr * dt * np.sqrt(nt) / nt) * G, nv,
+                            saveGt=saveGt, usematmul=usematmul,
+                            base_comm=comm, dtype=dtype)
     if conj:
         Frop = Frop.conj()
 
@@ -66,9 +75,12 @@ def MPIMDC(G: torch.Tensor, nt: int, nv: int, nfreq: int, dt: float = 1.0,
 
 class _FusedMDC(MPILinearOperator):
     """The MDC chain F1^H * I1^H * Fredholm1 * I * F (ref MDC.py:65-69)
-    as one fused per-rank pipeline (module docstring).  The FFT stages
-    follow fftlocal.FFTLocal's re-derived pylops convention exactly
-    (ortho rfft with sqrt(2)-scaled conjugate-twin bins)."""
+    as one fused per-rank pipeline (module docstring).  The chain AS A
+    WHOLE equals the composite built from fftlocal.FFTLocal's pylops
+    convention (ortho rfft with sqrt(2)-scaled conjugate-twin bins);
+    internally the twin scales cancel and the ortho norms are folded
+    into the Fredholm kernel, so the strided rocFFT transforms run
+    unscaled (exact operator algebra — parity-tested)."""
 
     def __init__(self, Frop, nt, nfft, nfreq, ns, nr, nv, twosided,
                  rdtype, cdtype, comm):
@@ -94,46 +106,66 @@ class _FusedMDC(MPILinearOperator):
                             torch.float32: np.float32,
                             torch.float64: np.float64}[t.dtype]))
 
+    @property
+    def _rt(self):
+        return torch.float32 if self._CT[self.cdtype] == torch.complex64 \
+            else torch.float64
+
     def _fwd_fft(self, x: DistributedArray, nmid: int,
                  shift: bool) -> torch.Tensor:
         t = x.local_array.reshape(self.nt, nmid * self.nv)
+        from . import _ffi
+        stream = torch.cuda.current_stream(t.device).cuda_stream
         if t.is_complex():
             # real extraction on the pam_unzip kernel: torch's strided
             # .real copy measures ~3.7 TB/s (r02 MDC kernel trace,
             # profiles/r02_mdc_kernel_stats.csv); the vectorized
             # deinterleave streams at the copy rate
-            from . import _ffi
-            r = torch.empty(t.shape, device=t.device,
-                            dtype=torch.float32 if t.dtype == torch.complex64
-                            else torch.float64)
+            r = torch.empty(t.shape, device=t.device, dtype=self._rt)
             _ffi.checked(_ffi.lib().pam_unzip(
-                torch.cuda.current_stream(t.device).cuda_stream,
-                r.data_ptr(), t.contiguous().data_ptr(), t.numel(),
-                _ffi.dtype_code(t.dtype)), "unzip")
+                stream, r.data_ptr(), t.contiguous().data_ptr(),
+                t.numel(), _ffi.dtype_code(t.dtype)), "unzip")
             t = r
+        else:
+            t = t.to(self._rt)
         if shift:
             t = torch.fft.ifftshift(t, dim=0)
-        # NOTE the sqrt(2) conjugate-twin scaling of the pylops real-FFT
-        # convention (fftlocal.py:47-48) is NOT applied here: inside
-        # this chain the forward's *sqrt(2) on bins [tw0, tw1) and the
-        # inverse's /sqrt(2) on the SAME bins surround the per-frequency
-        # block-diagonal Fredholm kernel, so they cancel algebraically
-        # (bins >= nfreq are masked to zero either way).  Two full
-        # passes over the frequency tensor disappear; standalone FFT
-        # operators (fftlocal.FFTLocal) keep the scaling.
-        return torch.fft.rfft(t, n=self.nt, dim=0, norm="ortho")
+        # Strided-batched rocFFT along dim 0 (pam_rfft_strided):
+        # torch.fft.rfft(dim=0) permutes the transform dim inward with
+        # TWO full copies per call (r02 MDC trace) — rocFFT takes the
+        # (stride=m, dist=1) layout natively.  The transform is
+        # UNSCALED and the sqrt(2) conjugate-twin scaling of the pylops
+        # real-FFT convention (fftlocal.py:47-48) is NOT applied:
+        # around the per-frequency block-diagonal Fredholm kernel the
+        # forward's *sqrt(2)/*1/sqrt(nt) and the inverse's twins cancel
+        # or fold into G (MPIMDC factory) — exact operator algebra,
+        # four full passes over the frequency tensor disappear.
+        # Standalone FFT operators (fftlocal.FFTLocal) keep the full
+        # convention.
+        m = t.shape[1]
+        f = torch.empty((self.nfft, m), device=t.device,
+                        dtype=self._CT[self.cdtype])
+        _ffi.checked(_ffi.lib().pam_rfft_strided(
+            stream, t.contiguous().data_ptr(), f.data_ptr(), self.nt, m,
+            _ffi.dtype_code(t.dtype)), "rfft_strided")
+        return f
 
     def _inv_fft(self, fr: torch.Tensor, nmid: int,
                  shift: bool) -> torch.Tensor:
+        from . import _ffi
+        m = nmid * self.nv
         # zero-pad the masked bins back to nfft (IdentityLocal adjoint)
         if self.nfreq == self.nfft:
-            z = fr.reshape(self.nfft, nmid * self.nv)  # fr is fresh
+            z = fr.reshape(self.nfft, m)  # fr is fresh (clobberable)
         else:
-            z = torch.zeros(self.nfft, nmid * self.nv, dtype=fr.dtype,
+            z = torch.zeros(self.nfft, m, dtype=fr.dtype,
                             device=fr.device)
-            z[: self.nfreq] = fr.reshape(self.nfreq, nmid * self.nv)
-        # twin-bin /sqrt(2) cancelled against the forward side (above)
-        out = torch.fft.irfft(z, n=self.nt, dim=0, norm="ortho")
+            z[: self.nfreq] = fr.reshape(self.nfreq, m)
+        out = torch.empty((self.nt, m), device=z.device, dtype=self._rt)
+        _ffi.checked(_ffi.lib().pam_irfft_strided(
+            torch.cuda.current_stream(z.device).cuda_stream,
+            z.contiguous().data_ptr(), out.data_ptr(), self.nt, m,
+            _ffi.dtype_code(out.dtype)), "irfft_strided")
         if shift:
             out = torch.fft.fftshift(out, dim=0)
         return out
