@@ -47,6 +47,7 @@ NORM_ORDS = [("2", None), ("0", 0), ("inf", np.inf), ("ninf", -np.inf),
 CGLS_DIMS = (17, 5)
 CGLS_NITER = 10
 CGLS_DAMPS = [("d0", 0.0), ("d05", 0.5)]
+CG_NITER = 12
 MM_SHAPES = (7, 5, 9)       # N, K, M (uneven vs any P in PS)
 MM_DTYPES = ["float64", "complex128"]
 FRED_SHAPE = (21, 4, 6, 5)  # nsl, nx, ny, nz (the reference test's)
@@ -71,6 +72,25 @@ def blockdiag_mats(P):
     """Per-rank dense serial blocks (examples/plot_cgls.py:30-33 style)."""
     rng = np.random.default_rng(31)
     shapes = [[(4, 6)], [(3, 2), (5, 5)], [(2, 3)], [(4, 4), (1, 2)]]
+    return [[rng.standard_normal(s) for s in shapes[r]] for r in range(P)]
+
+
+def spd_mats(P):
+    """Per-rank SPD blocks for CG (B = M M^T + n I)."""
+    rng = np.random.default_rng(41)
+    sizes = [4, 6, 3, 5]
+    out = []
+    for r in range(P):
+        n = sizes[r]
+        M = rng.standard_normal((n, n))
+        out.append([M @ M.T + n * np.eye(n)])
+    return out
+
+
+def vstack_mats(P):
+    """Per-rank blocks sharing 7 columns (MPIVStack requirement)."""
+    rng = np.random.default_rng(51)
+    shapes = [[(4, 7)], [(3, 7), (5, 7)], [(2, 7)], [(4, 7), (1, 7)]]
     return [[rng.standard_normal(s) for s in shapes[r]] for r in range(P)]
 
 
@@ -104,10 +124,13 @@ def plane_counts(dims, P):
 
 
 def _oracle_dist_planes(xg, dims, P):
+    return _oracle_dist_counts(xg, plane_counts(dims, P))
+
+
+def _oracle_dist_counts(xg, counts):
     from oracle.ranksim import Partition, SimArray
-    counts = plane_counts(dims, P)
-    offs = np.cumsum([0] + counts)
-    return SimArray([xg[offs[r]: offs[r + 1]] for r in range(P)],
+    offs = np.cumsum([0] + list(counts))
+    return SimArray([xg[offs[r]: offs[r + 1]] for r in range(len(counts))],
                     (int(xg.size),), 0, Partition.SCATTER)
 
 
@@ -176,6 +199,29 @@ def compute_oracle():
         yg = make_global_x(nr, P, seed_shift=1)
         out[f"bd_P{P}_mv"] = bop.matvec(oracle.to_dist(xg, P)).asarray()
         out[f"bd_P{P}_rmv"] = bop.rmatvec(oracle.to_dist(yg, P)).asarray()
+        # CG on an SPD BlockDiag (ref optimization/basic.py:13 cg)
+        smats = spd_mats(P)
+        sop2 = oracle.SimBlockDiag(smats)
+        nspd = sop2.shape[0]
+        yg2 = make_global_x(nspd, P)
+        spd_counts = [int(sum(A.shape[0] for A in ms)) for ms in smats]
+        xs2, cost2 = oracle.sim_cg(
+            sop2, _oracle_dist_counts(yg2, spd_counts),
+            _oracle_dist_counts(np.zeros(nspd), spd_counts),
+            CG_NITER, tol=0.0)
+        out[f"cg_P{P}_x"] = xs2.asarray()
+        out[f"cg_P{P}_cost"] = np.asarray(cost2)
+        # VStack vs dense (ref basicoperators/VStack.py:121-150; the
+        # oracle side IS the dense algebra — reference == dense pins it)
+        vmats = vstack_mats(P)
+        flat = [A for ms in vmats for A in ms]
+        nv_rows = int(sum(A.shape[0] for A in flat))
+        xv = make_global_x(7, P)
+        yv = make_global_x(nv_rows, P, seed_shift=1)
+        out[f"vs_P{P}_mv"] = np.concatenate([A @ xv for A in flat])
+        offs = np.cumsum([0] + [A.shape[0] for A in flat])
+        out[f"vs_P{P}_rmv"] = sum(
+            A.T @ yv[offs[i]: offs[i + 1]] for i, A in enumerate(flat))
         # Fredholm1
         for dt in FRED_DTYPES:
             G = fred_G(dt)
@@ -266,6 +312,13 @@ def _ref_rank_fn(P):
     nc_bd = int(sum(sum(A.shape[1] for A in ms) for ms in bd_mats))
     bd_x = make_global_x(nc_bd, P)
     bd_y = make_global_x(nr_bd, P, seed_shift=1)
+    spd = spd_mats(P)
+    n_spd = int(sum(sum(A.shape[0] for A in ms) for ms in spd))
+    cg_y = make_global_x(n_spd, P)
+    vmats = vstack_mats(P)
+    nv_rows = int(sum(A.shape[0] for ms in vmats for A in ms))
+    vs_x = make_global_x(7, P)
+    vs_y = make_global_x(nv_rows, P, seed_shift=1)
 
     def fn(rank):
         from pylops_mpi import (DistributedArray, MPIBlockDiag,
@@ -343,6 +396,38 @@ def _ref_rank_fn(P):
         xg, yg = bd_x, bd_y
         res[f"bd_P{P}_mv"] = bop.matvec(dist_from_global(xg)).asarray()
         res[f"bd_P{P}_rmv"] = bop.rmatvec(dist_from_global(yg)).asarray()
+        # CG on an SPD BlockDiag
+        from pylops_mpi import cg
+        sops = [pylops.MatrixMult(A) for A in spd[rank]]
+        sbop = MPIBlockDiag(ops=sops)
+        spd_counts = [int(sum(A.shape[0] for A in ms)) for ms in spd]
+
+        def dist_from_counts(vec, counts):
+            d = DistributedArray(
+                global_shape=int(np.sum(counts)),
+                local_shapes=[(int(c),) for c in counts],
+                dtype=np.float64)
+            off = int(np.sum(counts[:rank], initial=0))
+            d[:] = vec[off: off + counts[rank]]
+            return d
+
+        ycg = dist_from_counts(cg_y, spd_counts)
+        x0cg = dist_from_counts(np.zeros(n_spd), spd_counts)
+        xs2, iit2, cost2 = cg(sbop, ycg, x0cg, niter=CG_NITER, tol=0.0,
+                              show=False)
+        res[f"cg_P{P}_x"] = xs2.asarray()
+        res[f"cg_P{P}_cost"] = np.asarray(cost2)
+        # VStack (matvec: BROADCAST in, SCATTER out; rmatvec: allreduce)
+        from pylops_mpi import MPIVStack
+        vops = [pylops.MatrixMult(A) for A in vmats[rank]]
+        vop = MPIVStack(ops=vops)
+        xvd = DistributedArray(global_shape=7,
+                               partition=Partition.BROADCAST,
+                               dtype=np.float64)
+        xvd[:] = vs_x
+        res[f"vs_P{P}_mv"] = vop.matvec(xvd).asarray()
+        yvd = dist_from_global(vs_y)
+        res[f"vs_P{P}_rmv"] = vop.rmatvec(yvd).asarray()
         # Fredholm1
         for dt in FRED_DTYPES:
             G = fred_G(dt)
