@@ -263,6 +263,23 @@ int pam_rfft_strided(void* stream, const void* in_real, void* out_cplx,
 int pam_irfft_strided(void* stream, void* in_cplx, void* out_real,
                       int64_t nt, int64_t m, int dtype);
 
+/* Contiguous-batch variants ((m, nt) layout, transform along the last
+ * dim) — the MDC chain transposes itself (pam_unzip_t / pam_zip_t /
+ * pam_ctranspose) because rocFFT's strided real plans insert full
+ * pack/unpack copies. */
+int pam_rfft_contig(void* stream, const void* in_real, void* out_cplx,
+                    int64_t nt, int64_t m, int dtype);
+int pam_irfft_contig(void* stream, void* in_cplx, void* out_real,
+                     int64_t nt, int64_t m, int dtype);
+
+/* Transpose-fused (de)interleave: unzip_t = complex (nt, m) -> real
+ * (m, nt) (real parts); zip_t = real (m, nt) -> complex (nt, m) with
+ * zero imag.  dtype names the COMPLEX type. */
+int pam_unzip_t(void* stream, void* dst_real, const void* src_cplx,
+                int64_t nt, int64_t m, int dtype);
+int pam_zip_t(void* stream, void* dst_cplx, const void* src_real,
+              int64_t nt, int64_t m, int dtype);
+
 /* Complex (conj-)transpose on interleaved (re,im) pairs: At = A^T
  * (conj=0) or A^H (conj=1) — the complex MatrixMult adjoint panels
  * (ref MatrixMult.py:416,737 `A.T.conj()`). */

@@ -81,6 +81,18 @@ _SIGS = {
     "pam_irfft_strided": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                            ctypes.c_int64, ctypes.c_int64, ctypes.c_int],
                           ctypes.c_int),
+    "pam_rfft_contig": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                         ctypes.c_int64, ctypes.c_int64, ctypes.c_int],
+                        ctypes.c_int),
+    "pam_irfft_contig": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                          ctypes.c_int64, ctypes.c_int64, ctypes.c_int],
+                         ctypes.c_int),
+    "pam_unzip_t": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                     ctypes.c_int64, ctypes.c_int64, ctypes.c_int],
+                    ctypes.c_int),
+    "pam_zip_t": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                   ctypes.c_int64, ctypes.c_int64, ctypes.c_int],
+                  ctypes.c_int),
     "pam_unzip": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                    ctypes.c_int64, ctypes.c_int], ctypes.c_int),
     "pam_zip": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,

@@ -431,6 +431,102 @@ extern "C" int pam_zip(void* stream, void* dst_cplx, const void* src_real,
 }
 
 // ---------------------------------------------------------------------------
+// transpose-fused (de)interleave for the MDC chain's FFT layout (r02):
+// rocFFT's strided real plans insert their own full pack/unpack copy
+// kernels (~500 us/direction at the cfg5 shape, r02 trace), so the
+// chain transposes to the time-contiguous layout itself — fused with
+// the complex<->real conversion it already needs, at the LDS-tiled
+// transpose rate.  unzipT: complex (nt, m) -> real (m, nt) (real parts
+// only).  zipT: real (m, nt) -> complex (nt, m) with zero imag.
+// 32x32 LDS tiles (+1 pad), 256 threads as 8x32.
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void __launch_bounds__(BLK) unzipT_kernel(
+    T* __restrict__ dst, const T* __restrict__ src, int64_t nt, int64_t m) {
+  __shared__ T tile[32][33];
+  const int64_t k0 = (int64_t)blockIdx.y * 32;  // time rows
+  const int64_t j0 = (int64_t)blockIdx.x * 32;  // columns
+  const int tr = threadIdx.x / 32;
+  const int tc = threadIdx.x % 32;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int64_t k = k0 + tr + 8 * i;
+    if (k < nt && j0 + tc < m)
+      tile[tr + 8 * i][tc] = src[2 * (k * m + j0 + tc)];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int64_t j = j0 + tr + 8 * i;
+    if (j < m && k0 + tc < nt)
+      dst[j * nt + k0 + tc] = tile[tc][tr + 8 * i];
+  }
+}
+
+template <typename T>
+__global__ void __launch_bounds__(BLK) zipT_kernel(
+    T* __restrict__ dst, const T* __restrict__ src, int64_t nt, int64_t m) {
+  __shared__ T tile[32][33];
+  const int64_t j0 = (int64_t)blockIdx.y * 32;  // columns (src rows)
+  const int64_t k0 = (int64_t)blockIdx.x * 32;  // time
+  const int tr = threadIdx.x / 32;
+  const int tc = threadIdx.x % 32;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int64_t j = j0 + tr + 8 * i;
+    if (j < m && k0 + tc < nt)
+      tile[tr + 8 * i][tc] = src[j * nt + k0 + tc];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int64_t k = k0 + tr + 8 * i;
+    if (k < nt && j0 + tc < m) {
+      const int64_t off = 2 * (k * m + j0 + tc);
+      dst[off] = tile[tc][tr + 8 * i];
+      dst[off + 1] = (T)0;
+    }
+  }
+}
+
+extern "C" int pam_unzip_t(void* stream, void* dst_real,
+                           const void* src_cplx, int64_t nt, int64_t m,
+                           int dtype) {
+  if (nt <= 0 || m <= 0 || !dst_real || !src_cplx) return PAM_EARG;
+  dim3 grid((uint32_t)((m + 31) / 32), (uint32_t)((nt + 31) / 32));
+  hipStream_t s = (hipStream_t)stream;
+  if (dtype == PAM_C128) {
+    hipLaunchKernelGGL((unzipT_kernel<double>), grid, dim3(BLK), 0, s,
+                       (double*)dst_real, (const double*)src_cplx, nt, m);
+    return check(hipGetLastError());
+  }
+  if (dtype == PAM_C64) {
+    hipLaunchKernelGGL((unzipT_kernel<float>), grid, dim3(BLK), 0, s,
+                       (float*)dst_real, (const float*)src_cplx, nt, m);
+    return check(hipGetLastError());
+  }
+  return PAM_EDTYPE;
+}
+
+extern "C" int pam_zip_t(void* stream, void* dst_cplx, const void* src_real,
+                         int64_t nt, int64_t m, int dtype) {
+  if (nt <= 0 || m <= 0 || !dst_cplx || !src_real) return PAM_EARG;
+  dim3 grid((uint32_t)((nt + 31) / 32), (uint32_t)((m + 31) / 32));
+  hipStream_t s = (hipStream_t)stream;
+  if (dtype == PAM_C128) {
+    hipLaunchKernelGGL((zipT_kernel<double>), grid, dim3(BLK), 0, s,
+                       (double*)dst_cplx, (const double*)src_real, nt, m);
+    return check(hipGetLastError());
+  }
+  if (dtype == PAM_C64) {
+    hipLaunchKernelGGL((zipT_kernel<float>), grid, dim3(BLK), 0, s,
+                       (float*)dst_cplx, (const float*)src_real, nt, m);
+    return check(hipGetLastError());
+  }
+  return PAM_EDTYPE;
+}
+
+// ---------------------------------------------------------------------------
 // thresholding for ISTA/FISTA (the pylops _softthreshold/_hardthreshold
 // formulas the reference imports, ref optimization/cls_sparsity.py:10):
 //   soft real   : sign(x) * max(|x|-t, 0)

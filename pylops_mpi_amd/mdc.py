@@ -111,16 +111,54 @@ class _FusedMDC(MPILinearOperator):
         return torch.float32 if self._CT[self.cdtype] == torch.complex64 \
             else torch.float64
 
+    # PAM_MDC_FFT=strided falls back to the strided rocFFT plans (the
+    # r02 A/B loser at the cfg5 shape: rocFFT's strided real plans run
+    # their own full pack/unpack copy kernels; the contig pipeline
+    # replaces them with the transpose-FUSED unzip/zip kernels + one
+    # LDS-tiled ctranspose per direction)
+    _CONTIG = __import__("os").environ.get("PAM_MDC_FFT", "contig") \
+        != "strided"
+
     def _fwd_fft(self, x: DistributedArray, nmid: int,
                  shift: bool) -> torch.Tensor:
-        t = x.local_array.reshape(self.nt, nmid * self.nv)
+        """real-extract + rfft along the time axis -> (nfft, m) complex.
+
+        The transform is UNSCALED and the pylops real-FFT convention's
+        sqrt(2) twin-bin scale is NOT applied: around the per-frequency
+        block-diagonal Fredholm kernel the twins cancel and the ortho
+        1/sqrt(nt) of both directions folds into G (MPIMDC factory) —
+        exact operator algebra, parity-tested.  Standalone FFT
+        operators (fftlocal.FFTLocal) keep the full convention."""
         from . import _ffi
+        t = x.local_array.reshape(self.nt, nmid * self.nv)
+        m = t.shape[1]
         stream = torch.cuda.current_stream(t.device).cuda_stream
+        ct = self._CT[self.cdtype]
+        rcode = _ffi.dtype_code(self._rt)
+        if self._CONTIG:
+            # complex (nt, m) -> real (m, nt), transpose fused with the
+            # real extraction
+            r = torch.empty((m, self.nt), device=t.device, dtype=self._rt)
+            if t.is_complex():
+                _ffi.checked(_ffi.lib().pam_unzip_t(
+                    stream, r.data_ptr(), t.contiguous().data_ptr(),
+                    self.nt, m, _ffi.dtype_code(t.dtype)), "unzip_t")
+            else:
+                _ffi.checked(_ffi.lib().pam_transpose(
+                    stream, t.to(self._rt).contiguous().data_ptr(),
+                    r.data_ptr(), self.nt, m, rcode), "transpose")
+            if shift:
+                r = torch.fft.ifftshift(r, dim=1)
+            F = torch.empty((m, self.nfft), device=t.device, dtype=ct)
+            _ffi.checked(_ffi.lib().pam_rfft_contig(
+                stream, r.contiguous().data_ptr(), F.data_ptr(), self.nt,
+                m, rcode), "rfft_contig")
+            f = torch.empty((self.nfft, m), device=t.device, dtype=ct)
+            _ffi.checked(_ffi.lib().pam_ctranspose(
+                stream, F.data_ptr(), f.data_ptr(), m, self.nfft, 0,
+                _ffi.dtype_code(ct)), "ctranspose")
+            return f
         if t.is_complex():
-            # real extraction on the pam_unzip kernel: torch's strided
-            # .real copy measures ~3.7 TB/s (r02 MDC kernel trace,
-            # profiles/r02_mdc_kernel_stats.csv); the vectorized
-            # deinterleave streams at the copy rate
             r = torch.empty(t.shape, device=t.device, dtype=self._rt)
             _ffi.checked(_ffi.lib().pam_unzip(
                 stream, r.data_ptr(), t.contiguous().data_ptr(),
@@ -130,30 +168,22 @@ class _FusedMDC(MPILinearOperator):
             t = t.to(self._rt)
         if shift:
             t = torch.fft.ifftshift(t, dim=0)
-        # Strided-batched rocFFT along dim 0 (pam_rfft_strided):
-        # torch.fft.rfft(dim=0) permutes the transform dim inward with
-        # TWO full copies per call (r02 MDC trace) — rocFFT takes the
-        # (stride=m, dist=1) layout natively.  The transform is
-        # UNSCALED and the sqrt(2) conjugate-twin scaling of the pylops
-        # real-FFT convention (fftlocal.py:47-48) is NOT applied:
-        # around the per-frequency block-diagonal Fredholm kernel the
-        # forward's *sqrt(2)/*1/sqrt(nt) and the inverse's twins cancel
-        # or fold into G (MPIMDC factory) — exact operator algebra,
-        # four full passes over the frequency tensor disappear.
-        # Standalone FFT operators (fftlocal.FFTLocal) keep the full
-        # convention.
-        m = t.shape[1]
-        f = torch.empty((self.nfft, m), device=t.device,
-                        dtype=self._CT[self.cdtype])
+        f = torch.empty((self.nfft, m), device=t.device, dtype=ct)
         _ffi.checked(_ffi.lib().pam_rfft_strided(
             stream, t.contiguous().data_ptr(), f.data_ptr(), self.nt, m,
-            _ffi.dtype_code(t.dtype)), "rfft_strided")
+            rcode), "rfft_strided")
         return f
 
     def _inv_fft(self, fr: torch.Tensor, nmid: int,
                  shift: bool) -> torch.Tensor:
+        """zero-pad + irfft + complex carrier -> (nt, m) complex (the
+        composite chain's stage wrappers carry complex storage end to
+        end — mirrored)."""
         from . import _ffi
         m = nmid * self.nv
+        ct = self._CT[self.cdtype]
+        rcode = _ffi.dtype_code(self._rt)
+        stream = torch.cuda.current_stream(fr.device).cuda_stream
         # zero-pad the masked bins back to nfft (IdentityLocal adjoint)
         if self.nfreq == self.nfft:
             z = fr.reshape(self.nfft, m)  # fr is fresh (clobberable)
@@ -161,42 +191,47 @@ class _FusedMDC(MPILinearOperator):
             z = torch.zeros(self.nfft, m, dtype=fr.dtype,
                             device=fr.device)
             z[: self.nfreq] = fr.reshape(self.nfreq, m)
-        out = torch.empty((self.nt, m), device=z.device, dtype=self._rt)
+        if self._CONTIG:
+            Zt = torch.empty((m, self.nfft), device=z.device, dtype=ct)
+            _ffi.checked(_ffi.lib().pam_ctranspose(
+                stream, z.contiguous().data_ptr(), Zt.data_ptr(),
+                self.nfft, m, 0, _ffi.dtype_code(ct)), "ctranspose")
+            r = torch.empty((m, self.nt), device=z.device, dtype=self._rt)
+            _ffi.checked(_ffi.lib().pam_irfft_contig(
+                stream, Zt.data_ptr(), r.data_ptr(), self.nt, m, rcode),
+                "irfft_contig")
+            if shift:
+                r = torch.fft.fftshift(r, dim=1)
+            out = torch.empty((self.nt, m), device=z.device, dtype=ct)
+            _ffi.checked(_ffi.lib().pam_zip_t(
+                stream, out.data_ptr(), r.contiguous().data_ptr(),
+                self.nt, m, _ffi.dtype_code(ct)), "zip_t")
+            return out
+        r = torch.empty((self.nt, m), device=z.device, dtype=self._rt)
         _ffi.checked(_ffi.lib().pam_irfft_strided(
-            torch.cuda.current_stream(z.device).cuda_stream,
-            z.contiguous().data_ptr(), out.data_ptr(), self.nt, m,
-            _ffi.dtype_code(out.dtype)), "irfft_strided")
+            stream, z.contiguous().data_ptr(), r.data_ptr(), self.nt, m,
+            rcode), "irfft_strided")
         if shift:
-            out = torch.fft.fftshift(out, dim=0)
+            r = torch.fft.fftshift(r, dim=0)
+        out = torch.empty((self.nt, m), device=z.device, dtype=ct)
+        _ffi.checked(_ffi.lib().pam_zip(
+            stream, out.data_ptr(), r.contiguous().data_ptr(),
+            r.numel(), _ffi.dtype_code(ct)), "zip")
         return out
 
     _CT = {np.dtype(np.complex64): torch.complex64,
            np.dtype(np.complex128): torch.complex128}
-
-    def _to_cplx(self, out: torch.Tensor) -> torch.Tensor:
-        """real -> complex-with-zero-imag carrier on the pam_zip kernel
-        (the composite chain's stage wrappers carry complex storage end
-        to end — mirror that; torch's .to(complex) cast is the strided
-        half-rate path)."""
-        from . import _ffi
-        z = torch.empty(out.shape, device=out.device,
-                        dtype=self._CT[self.cdtype])
-        _ffi.checked(_ffi.lib().pam_zip(
-            torch.cuda.current_stream(out.device).cuda_stream,
-            z.data_ptr(), out.contiguous().data_ptr(), out.numel(),
-            _ffi.dtype_code(z.dtype)), "zip")
-        return z
 
     def _matvec(self, x: DistributedArray) -> DistributedArray:
         f = self._fwd_fft(x, self.nr, self.twosided)       # Fop
         f = f[: self.nfreq]                                # Iop
         fr = self.Frop.matvec(self._wrap(f.contiguous()))  # Fredholm
         out = self._inv_fft(fr.local_array, self.ns, False)  # I1^H F1^H
-        return self._wrap(self._to_cplx(out))
+        return self._wrap(out)
 
     def _rmatvec(self, y: DistributedArray) -> DistributedArray:
         f = self._fwd_fft(y, self.ns, False)               # F1op, I1op
         f = f[: self.nfreq]
         fr = self.Frop.rmatvec(self._wrap(f.contiguous()))
         out = self._inv_fft(fr.local_array, self.nr, self.twosided)
-        return self._wrap(self._to_cplx(out))
+        return self._wrap(out)
