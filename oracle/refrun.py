@@ -83,11 +83,17 @@ def run_reference(P, fn):
         t.start()
     for t in threads:
         t.join(timeout=300.0)
+    first_err = next((e for e in errs if e is not None), None)
     for t in threads:
         if t.is_alive():
+            # a raised rank leaves the others at a barrier — surface the
+            # root cause, not the hang
+            if first_err is not None:
+                raise RuntimeError(
+                    f"reference run hung after a rank raised: "
+                    f"{first_err!r}") from first_err
             raise RuntimeError("reference run deadlocked (thread alive "
                                "after 300 s)")
-    for e in errs:
-        if e is not None:
-            raise e
+    if first_err is not None:
+        raise first_err
     return outs

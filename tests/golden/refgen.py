@@ -33,8 +33,12 @@ from oracle import matmult as om  # noqa: E402
 
 GOLDEN_PATH = os.path.join(_HERE, "golden_ref.npz")
 
-PS = (1, 2, 3, 4)
+PS = (1, 2, 3, 4, 8)
 DIMS = [(32,), (17, 5), (16, 4, 3)]
+# P=8 splits the small DIMS to 2 rows/rank, below the 5-point stencils'
+# ghost width — the REFERENCE itself raises there — so P=8 uses taller
+# row counts (>= 4 rows per rank)
+DIMS_P8 = [(40,), (33, 5), (32, 4, 3)]
 SAMPLING = 1.5
 FD1_CASES = [("forward", 3, False), ("backward", 3, False),
              ("centered", 3, False), ("centered", 3, True),
@@ -42,6 +46,7 @@ FD1_CASES = [("forward", 3, False), ("backward", 3, False),
 FD2_CASES = [("forward", False), ("backward", False),
              ("centered", False), ("centered", True)]
 MATH_N = (13, 4)
+MATH_N8 = (33, 4)   # P=8: >= 4 rows/rank so width-2 ghosts exist
 NORM_ORDS = [("2", None), ("0", 0), ("inf", np.inf), ("ninf", -np.inf),
              ("1p5", 1.5)]
 CGLS_DIMS = (17, 5)
@@ -72,7 +77,8 @@ def blockdiag_mats(P):
     """Per-rank dense serial blocks (examples/plot_cgls.py:30-33 style)."""
     rng = np.random.default_rng(31)
     shapes = [[(4, 6)], [(3, 2), (5, 5)], [(2, 3)], [(4, 4), (1, 2)]]
-    return [[rng.standard_normal(s) for s in shapes[r]] for r in range(P)]
+    return [[rng.standard_normal(s) for s in shapes[r % 4]]
+            for r in range(P)]
 
 
 def spd_mats(P):
@@ -81,7 +87,7 @@ def spd_mats(P):
     sizes = [4, 6, 3, 5]
     out = []
     for r in range(P):
-        n = sizes[r]
+        n = sizes[r % 4]
         M = rng.standard_normal((n, n))
         out.append([M @ M.T + n * np.eye(n)])
     return out
@@ -91,7 +97,8 @@ def vstack_mats(P):
     """Per-rank blocks sharing 7 columns (MPIVStack requirement)."""
     rng = np.random.default_rng(51)
     shapes = [[(4, 7)], [(3, 7), (5, 7)], [(2, 7)], [(4, 7), (1, 7)]]
-    return [[rng.standard_normal(s) for s in shapes[r]] for r in range(P)]
+    return [[rng.standard_normal(s) for s in shapes[r % 4]]
+            for r in range(P)]
 
 
 def fred_G(dtype):
@@ -139,7 +146,7 @@ def compute_oracle():
     out = {}
     for P in PS:
         # FD operators
-        for dims in DIMS:
+        for dims in (DIMS_P8 if P >= 8 else DIMS):
             n = int(np.prod(dims))
             xg = make_global_x(n, P)
             yg = make_global_x(n, P, seed_shift=1)
@@ -156,7 +163,8 @@ def compute_oracle():
                 out[key + "_mv"] = op.matvec(oracle.to_dist(xg, P)).asarray()
                 out[key + "_rmv"] = op.rmatvec(oracle.to_dist(yg, P)).asarray()
         # array math
-        n = int(np.prod(MATH_N))
+        mn = MATH_N8 if P >= 8 else MATH_N
+        n = int(np.prod(mn))
         xg = make_global_x(n, P)
         yg = make_global_x(n, P, seed_shift=1)
         dx = oracle.to_dist(xg, P)
@@ -169,7 +177,7 @@ def compute_oracle():
             d = dxa if name == "1p5" else dx
             out[f"math_P{P}_norm{name}"] = np.asarray(d.norm(o))
         out[f"math_P{P}_addmul"] = ((dx + dy) * dx).asarray()
-        d2 = oracle.to_dist(xg.reshape(MATH_N), P)
+        d2 = oracle.to_dist(xg.reshape(mn), P)
         for w in (1, 2):
             gh = d2.add_ghost_cells(cells_front=w, cells_back=w)
             out[f"math_P{P}_ghost{w}"] = np.concatenate(
@@ -220,8 +228,19 @@ def compute_oracle():
         yv = make_global_x(nv_rows, P, seed_shift=1)
         out[f"vs_P{P}_mv"] = np.concatenate([A @ xv for A in flat])
         offs = np.cumsum([0] + [A.shape[0] for A in flat])
-        out[f"vs_P{P}_rmv"] = sum(
-            A.T @ yv[offs[i]: offs[i + 1]] for i, A in enumerate(flat))
+        # fold in the reference's exact order: per-rank vstack-sum, then
+        # rank-ordered allreduce (bitwise at any P)
+        per_rank, idx = [], 0
+        for ms in vmats:
+            segs = []
+            for A in ms:
+                segs.append(A.T @ yv[offs[idx]: offs[idx + 1]])
+                idx += 1
+            per_rank.append(np.sum(np.vstack(segs), axis=0))
+        acc = per_rank[0]
+        for v in per_rank[1:]:
+            acc = acc + v
+        out[f"vs_P{P}_rmv"] = acc
         # Fredholm1
         for dt in FRED_DTYPES:
             G = fred_G(dt)
@@ -297,11 +316,12 @@ def _ref_rank_fn(P):
     # the global numpy RNG (the reference test recipe's np.random.seed),
     # which is not thread-safe across the P rank threads
     fd_inputs = {}
-    for dims in DIMS:
+    for dims in (DIMS_P8 if P >= 8 else DIMS):
         n = int(np.prod(dims))
         fd_inputs[dims] = (make_global_x(n, P),
                            make_global_x(n, P, seed_shift=1))
-    nmath = int(np.prod(MATH_N))
+    mn = MATH_N8 if P >= 8 else MATH_N
+    nmath = int(np.prod(mn))
     math_x = make_global_x(nmath, P)
     math_y = make_global_x(nmath, P, seed_shift=1)
     ncgls = int(np.prod(CGLS_DIMS))
@@ -336,7 +356,7 @@ def _ref_rank_fn(P):
             return d
 
         res = {}
-        for dims in DIMS:
+        for dims in (DIMS_P8 if P >= 8 else DIMS):
             n = int(np.prod(dims))
             xg, yg = fd_inputs[dims]
             for kind, order, edge in FD1_CASES:
@@ -364,10 +384,10 @@ def _ref_rank_fn(P):
             res[f"math_P{P}_norm{name}"] = np.asarray(
                 d.norm() if o is None else d.norm(o))
         res[f"math_P{P}_addmul"] = ((dx + dy) * dx).asarray()
-        d2 = DistributedArray(global_shape=MATH_N, dtype=np.float64)
-        counts = [oracle.local_split(MATH_N, P, r)[0] for r in range(P)]
+        d2 = DistributedArray(global_shape=mn, dtype=np.float64)
+        counts = [oracle.local_split(mn, P, r)[0] for r in range(P)]
         off = int(np.sum(counts[:rank], initial=0))
-        d2[:] = xg.reshape(MATH_N)[off: off + d2.local_shape[0]]
+        d2[:] = xg.reshape(mn)[off: off + d2.local_shape[0]]
         for w in (1, 2):
             res[f"__perrank__math_P{P}_ghost{w}"] = d2.add_ghost_cells(
                 cells_front=w, cells_back=w)
