@@ -226,11 +226,12 @@ def test_gemv_vs_numpy():
         A = rng.standard_normal((nr, nc))
         x = rng.standard_normal(nc)
         y = rng.standard_normal(nr)
-        op = pm.DenseLocal(dev(A))
-        assert_allclose(host(op.matvec(dev(x))), A @ x,
-                        rtol=1e-13, atol=1e-13)
-        assert_allclose(host(op.rmatvec(dev(y))), A.T @ y,
-                        rtol=1e-13, atol=1e-13)
+        for saveAt in (True, False):
+            op = pm.DenseLocal(dev(A), saveAt=saveAt)
+            assert_allclose(host(op.matvec(dev(x))), A @ x,
+                            rtol=1e-13, atol=1e-13)
+            assert_allclose(host(op.rmatvec(dev(y))), A.T @ y,
+                            rtol=1e-13, atol=1e-13)
 
 
 def test_gemv_deterministic():
