@@ -182,6 +182,13 @@ def compute_oracle():
             gh = d2.add_ghost_cells(cells_front=w, cells_back=w)
             out[f"math_P{P}_ghost{w}"] = np.concatenate(
                 [g.ravel() for g in gh])
+        # redistribute axis 0 -> axis 1 (ref :493-552): the result is the
+        # BALANCED column split of the global array — dense expectation
+        G2 = xg.reshape(mn)
+        cols = [oracle.local_split((mn[1],), P, r)[0] for r in range(P)]
+        co = np.cumsum([0] + cols)
+        out[f"math_P{P}_redist"] = np.concatenate(
+            [G2[:, co[r]: co[r + 1]].ravel() for r in range(P)])
         # CGLS on FD1 centered3.  x0 must be plane-aligned: the
         # reference's reshaped wrapper leaves operator OUTPUTS on the
         # plane split (ref decorators.py:79-82), and CGLS subtracts
@@ -391,6 +398,8 @@ def _ref_rank_fn(P):
         for w in (1, 2):
             res[f"__perrank__math_P{P}_ghost{w}"] = d2.add_ghost_cells(
                 cells_front=w, cells_back=w)
+        rd = d2.redistribute(axis=1)
+        res[f"__perrank__math_P{P}_redist"] = np.asarray(rd.local_array)
         # CGLS
         nc = ncgls
         xg = cgls_x
