@@ -44,3 +44,41 @@ def main():
     print(f"bd pair saveAt=True {timeit(pair2)*1e6:8.1f} us")
 
 main()
+
+def more():
+    init_default_comm(torch.device("cuda:0"))
+    g = torch.Generator(device="cuda").manual_seed(42)
+    n = 4096
+    Ad = torch.rand((n, n), generator=g, dtype=torch.float64,
+                    device="cuda") * 2 - 1
+    xv = torch.rand(n, generator=g, dtype=torch.float64, device="cuda")
+    op = pm.DenseLocal(Ad, saveAt=False)
+    bd = pm.MPIBlockDiag([op])
+    x = pm.DistributedArray((n,))
+    x[:] = xv
+    y0 = bd.matvec(x)
+    def pair_sync():
+        yy = bd.matvec(x)
+        torch.cuda.synchronize()
+        bd.rmatvec(yy)
+    print(f"pair w/ mid-sync {timeit(pair_sync)*1e6:8.1f} us")
+    def pair_fixed_input():
+        bd.matvec(x)
+        bd.rmatvec(y0)     # rmv input NOT the fresh mv output
+    print(f"pair fixed-input {timeit(pair_fixed_input)*1e6:8.1f} us")
+    def rr():
+        bd.rmatvec(y0)
+        bd.rmatvec(y0)
+    print(f"rmv+rmv          {timeit(rr)*1e6:8.1f} us")
+    def mm():
+        bd.matvec(x)
+        bd.matvec(x)
+    print(f"mv+mv            {timeit(mm)*1e6:8.1f} us")
+    op_raw = pm.DenseLocal(Ad, saveAt=False)
+    def raw_pair():
+        yy = op_raw.matvec(xv)
+        op_raw.rmatvec(yy)
+    print(f"raw mv->rmv      {timeit(raw_pair)*1e6:8.1f} us")
+
+if __name__ == "__main__" or True:
+    more()
