@@ -280,6 +280,15 @@ int pam_unzip_t(void* stream, void* dst_real, const void* src_cplx,
 int pam_zip_t(void* stream, void* dst_cplx, const void* src_real,
               int64_t nt, int64_t m, int dtype);
 
+/* 256^2-tile f32 GEMM with async global->LDS staging (glds deep
+ * pipeline): C (+)= At^T @ B for a K-MAJOR A (At = A^T, [K, M]
+ * row-major) and row-major B [K, N], C [M, N].  Fast path only:
+ * returns PAM_EARG unless M,N % 256 == 0, K % 32 == 0 and pointers are
+ * 16-B aligned — callers fall back to pam_gemm. */
+int pam_gemm_kt(void* stream, const void* At, const void* B, void* C,
+                int64_t M, int64_t N, int64_t K, int accumulate,
+                int dtype);
+
 /* Complex (conj-)transpose on interleaved (re,im) pairs: At = A^T
  * (conj=0) or A^H (conj=1) — the complex MatrixMult adjoint panels
  * (ref MatrixMult.py:416,737 `A.T.conj()`). */

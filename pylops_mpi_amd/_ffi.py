@@ -114,6 +114,10 @@ _SIGS = {
                   ctypes.c_void_p, ctypes.c_int64, ctypes.c_int64,
                   ctypes.c_int64, ctypes.c_int64, ctypes.c_int64,
                   ctypes.c_int64, ctypes.c_int, ctypes.c_int], ctypes.c_int),
+    "pam_gemm_kt": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                     ctypes.c_void_p, ctypes.c_int64, ctypes.c_int64,
+                     ctypes.c_int64, ctypes.c_int, ctypes.c_int],
+                    ctypes.c_int),
     "pam_transpose": ([ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                        ctypes.c_int64, ctypes.c_int64, ctypes.c_int],
                       ctypes.c_int),
