@@ -70,33 +70,55 @@ __global__ void __launch_bounds__(G256_BLK, 1) gemm256_f32_kernel(
     }
   };
 
+  // Schedule (guide §pipelining-across-barriers): the p+1 DMA spans the
+  // whole MFMA(p) loop.  Per iteration: issue glds(p+1) [16 instrs per
+  // wave], then wait vmcnt(16) — drains glds(p), LEAVES p+1 in flight —
+  // raw s_barrier (NOT __syncthreads: its fence would emit vmcnt(0) and
+  // drain the span), MFMA(p), raw s_barrier before the next overwrite.
   const int64_t NP = K / BK;
   glds_panel(0, 0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
   for (int64_t p = 0; p < NP; ++p) {
     const int cur = (int)(p & 1);
-    if (p + 1 < NP) glds_panel(cur ^ 1, (p + 1) * BK);
-    // MFMA on panel p (reads LDS only) overlaps the p+1 DMA
+    if (p + 1 < NP) {
+      glds_panel(cur ^ 1, (p + 1) * BK);
+      asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    // MFMA on panel p (reads LDS only) overlaps the p+1 DMA.  Fragments
+    // are register-double-buffered one k-pair ahead (same scheme as
+    // gemm_kernel): at 1 wave/SIMD there is no cross-wave latency
+    // hiding, so the next ds_reads must issue under the current MFMAs.
+    float a[2][4], b[2][4];
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+      a[0][mi] = As[cur][lk][wr * 128 + mi * 32 + li];
+#pragma unroll
+    for (int nj = 0; nj < 4; ++nj)
+      b[0][nj] = Bs[cur][lk][wc * 128 + nj * 32 + li];
 #pragma unroll
     for (int kp = 0; kp < BK / 2; ++kp) {
-      const int krow = kp * 2 + lk;
-      float a[4], b[4];
+      const int cf = kp & 1, nf = (kp + 1) & 1;
+      if (kp + 1 < BK / 2) {
+        const int krow = (kp + 1) * 2 + lk;
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
-        a[mi] = As[cur][krow][wr * 128 + mi * 32 + li];
+        for (int mi = 0; mi < 4; ++mi)
+          a[nf][mi] = As[cur][krow][wr * 128 + mi * 32 + li];
 #pragma unroll
-      for (int nj = 0; nj < 4; ++nj)
-        b[nj] = Bs[cur][krow][wc * 128 + nj * 32 + li];
+        for (int nj = 0; nj < 4; ++nj)
+          b[nf][nj] = Bs[cur][krow][wc * 128 + nj * 32 + li];
+      }
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
         for (int nj = 0; nj < 4; ++nj)
           acc[mi][nj] = __builtin_amdgcn_mfma_f32_32x32x2f32(
-              a[mi], b[nj], acc[mi][nj], 0, 0, 0);
+              a[cf][mi], b[cf][nj], acc[mi][nj], 0, 0, 0);
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
   }
 
   // epilogue: v_mfma_f32_32x32x2 C mapping — col = lane&31,
