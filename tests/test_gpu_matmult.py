@@ -187,3 +187,36 @@ def test_cgemm_accumulate(dtype):
         M * K, K * N, M * N, 0, 1, _ffi.dtype_code(Ad.dtype)), "cgemm")
     tol = 1e-12 if dtype == "complex128" else 2e-4
     assert_allclose(Cd.cpu().numpy(), C0 + A @ X, rtol=tol, atol=tol * 10)
+
+
+def test_gemm_kt_glds():
+    """pam_gemm_kt (256^2-tile glds pipeline, k-major A): correctness vs
+    pam_gemm.  Kept as a measured NEGATIVE for throughput (r02: 125-127
+    TF vs 136 at 8192^3 — 1 block/CU loses the 4-WG occupancy the 128^2
+    register-pipeline kernel enjoys; DESIGN.md negative-results list)."""
+    n = 512
+    rng = np.random.default_rng(14)
+    A = rng.standard_normal((n, n)).astype(np.float32)
+    B = rng.standard_normal((n, n)).astype(np.float32)
+    Ad, Bd = dev(A).contiguous(), dev(B).contiguous()
+    At = torch.empty((n, n), dtype=torch.float32, device="cuda:0")
+    C = torch.empty((n, n), dtype=torch.float32, device="cuda:0")
+    s = torch.cuda.current_stream().cuda_stream
+    _ffi.checked(_ffi.lib().pam_transpose(
+        s, Ad.data_ptr(), At.data_ptr(), n, n, 1), "t")
+    _ffi.checked(_ffi.lib().pam_gemm_kt(
+        s, At.data_ptr(), Bd.data_ptr(), C.data_ptr(), n, n, n, 0, 1),
+        "gemm_kt")
+    ref = A.astype(np.float64) @ B.astype(np.float64)
+    assert_allclose(C.cpu().numpy(), ref, rtol=2e-4, atol=1e-2)
+    # accumulate variant
+    C0 = rng.standard_normal((n, n)).astype(np.float32)
+    Cd = dev(C0).contiguous()
+    _ffi.checked(_ffi.lib().pam_gemm_kt(
+        s, At.data_ptr(), Bd.data_ptr(), Cd.data_ptr(), n, n, n, 1, 1),
+        "gemm_kt_acc")
+    assert_allclose(Cd.cpu().numpy(), C0 + ref, rtol=2e-4, atol=1e-2)
+    # fast-path precondition: unaligned sizes are refused, not wrong
+    bad = _ffi.lib().pam_gemm_kt(
+        s, At.data_ptr(), Bd.data_ptr(), C.data_ptr(), n - 8, n, n, 0, 1)
+    assert bad != 0
