@@ -39,18 +39,52 @@ class Identity(LinearOperator):
         return x.copy()
 
 
-class _ImportOnly(LinearOperator):
-    _name = "stub"
+class _SerialDerivative(LinearOperator):
+    """Serial pylops FirstDerivative/SecondDerivative stand-in for the
+    reference's Gradient/Laplacian local blocks (ref Gradient.py:108-116,
+    Laplacian.py:97-126).  pylops itself is absent, so the
+    per-axis stencil is supplied by the repo oracle's rank-1 simulation
+    of the SAME published formulas (oracle/stencils.py — already pinned
+    against the reference's distributed axis-0 operators at P=1..8);
+    what the ref-parity suite then pins is the DISTRIBUTED composition:
+    StackedVStack / BlockDiag mechanics and the scaled/summed composite
+    algebra."""
 
-    def __init__(self, *a, **k):
-        raise NotImplementedError(
-            f"pylops stub: serial {self._name} is not implemented — the "
-            "ref-parity suite does not construct it")
+    def __init__(self, dims, axis=0, sampling=1.0, kind="centered",
+                 edge=False, dtype="float64"):
+        dims = (dims,) if isinstance(dims, (int, np.integer)) \
+            else tuple(int(d) for d in dims)
+        n = int(np.prod(dims))
+        super().__init__(dtype=np.dtype(dtype), shape=(n, n))
+        self.dims = dims        # after super(): the base derives a flat
+        self.axis = int(axis) % len(dims)
+        moved = (dims[self.axis],) + tuple(
+            d for i, d in enumerate(dims) if i != self.axis)
+        self._sim = self._make_sim(moved, sampling, kind, edge)
+
+    def _roll(self, x, fwd):
+        from oracle.ranksim import to_dist
+        arr = np.moveaxis(np.asarray(x).reshape(self.dims), self.axis, 0)
+        shp = arr.shape
+        d = to_dist(arr.ravel(), 1)
+        res = (self._sim.matvec(d) if fwd else self._sim.rmatvec(d))
+        out = np.moveaxis(res.asarray().reshape(shp), 0, self.axis)
+        return np.ascontiguousarray(out).ravel()
+
+    def _matvec(self, x):
+        return self._roll(x, True)
+
+    def _rmatvec(self, x):
+        return self._roll(x, False)
 
 
-class FirstDerivative(_ImportOnly):
-    _name = "FirstDerivative"
+class FirstDerivative(_SerialDerivative):
+    def _make_sim(self, moved, sampling, kind, edge):
+        from oracle.stencils import SimFirstDerivative
+        return SimFirstDerivative(moved, sampling, kind, edge, order=3)
 
 
-class SecondDerivative(_ImportOnly):
-    _name = "SecondDerivative"
+class SecondDerivative(_SerialDerivative):
+    def _make_sim(self, moved, sampling, kind, edge):
+        from oracle.stencils import SimSecondDerivative
+        return SimSecondDerivative(moved, sampling, kind, edge)

@@ -53,6 +53,11 @@ CGLS_DIMS = (17, 5)
 CGLS_NITER = 10
 CGLS_DAMPS = [("d0", 0.0), ("d05", 0.5)]
 CG_NITER = 12
+GRAD_DIMS = (10, 6, 4)
+GRAD_SAMP = (1.5, 2.0, 0.5)
+LAP_AXES = (0, 1, 2)
+LAP_W = (1.0, 2.0, 0.5)
+LAP_SAMP = (1.0, 1.5, 2.0)
 MM_SHAPES = (7, 5, 9)       # N, K, M (uneven vs any P in PS)
 MM_DTYPES = ["float64", "complex128"]
 FRED_SHAPE = (21, 4, 6, 5)  # nsl, nx, ny, nz (the reference test's)
@@ -123,6 +128,22 @@ def fred_x(dtype, forward):
 def fred_split(P):
     nsl = FRED_SHAPE[0]
     return [oracle.local_split((nsl,), P, r)[0] for r in range(P)]
+
+
+def _serial_axis_fd(xg, dims, axis, sampling, edge, fwd, second=False):
+    """Serial derivative along ``axis`` of the global array (the exact
+    map the reference composes for Gradient/Laplacian — distribution-
+    independent dense expectation)."""
+    from oracle.ranksim import to_dist
+    from oracle.stencils import SimFirstDerivative, SimSecondDerivative
+    arr = np.moveaxis(xg.reshape(dims), axis, 0)
+    moved = arr.shape
+    sim = (SimSecondDerivative(moved, sampling, "centered", edge) if second
+           else SimFirstDerivative(moved, sampling, "centered", edge, 3))
+    d = to_dist(arr.ravel(), 1)
+    res = (sim.matvec(d) if fwd else sim.rmatvec(d)).asarray()
+    return np.ascontiguousarray(
+        np.moveaxis(res.reshape(moved), 0, axis)).ravel()
 
 
 def plane_counts(dims, P):
@@ -248,6 +269,13 @@ def compute_oracle():
         for v in per_rank[1:]:
             acc = acc + v
         out[f"vs_P{P}_rmv"] = acc
+        # Gradient / Laplacian (composed operators; dense expectations
+        # are distribution-independent).  P <= 4 only: at P=8 the
+        # 10-row dims leave the reference's reshaped rebalance with
+        # ghost moves larger than a rank's block — the REFERENCE itself
+        # raises there.
+        if P <= 4:
+            out.update(_grad_lap_oracle(P))
         # Fredholm1
         for dt in FRED_DTYPES:
             G = fred_G(dt)
@@ -346,6 +374,10 @@ def _ref_rank_fn(P):
     nv_rows = int(sum(A.shape[0] for ms in vmats for A in ms))
     vs_x = make_global_x(7, P)
     vs_y = make_global_x(nv_rows, P, seed_shift=1)
+    ng = int(np.prod(GRAD_DIMS))
+    grad_x = make_global_x(ng, P)
+    grad_ys = [make_global_x(ng, P, seed_shift=2 + i) for i in range(3)]
+    lap_y = make_global_x(ng, P, seed_shift=5)
 
     def fn(rank):
         from pylops_mpi import (DistributedArray, MPIBlockDiag,
@@ -457,6 +489,31 @@ def _ref_rank_fn(P):
         res[f"vs_P{P}_mv"] = vop.matvec(xvd).asarray()
         yvd = dist_from_global(vs_y)
         res[f"vs_P{P}_rmv"] = vop.rmatvec(yvd).asarray()
+        # Gradient (StackedVStack composition) + Laplacian (scaled-sum
+        # composite algebra), serial axis>=1 blocks via the pylops stub.
+        # P <= 4 only (the reference's own rebalance cannot ghost the
+        # 10-row dims at P=8 — see the oracle-side note).
+        if P <= 4:
+            from pylops_mpi import MPIGradient, MPILaplacian
+            gop = MPIGradient(dims=GRAD_DIMS, sampling=GRAD_SAMP, edge=False,
+                              kind="centered")
+            gx = dist_from_global(grad_x)
+            ym = gop.matvec(gx)
+            for i in range(3):
+                res[f"grad_P{P}_mv{i}"] = ym.distarrays[i].asarray()
+            # fill the stacked input with seeded globals (component splits
+            # come from the matvec output structure)
+            for i, di in enumerate(ym.distarrays):
+                counts = [int(np.prod(sh)) for sh in di.local_shapes]
+                off = int(np.sum(counts[:rank], initial=0))
+                di[:] = grad_ys[i][off: off + counts[rank]].reshape(
+                    di.local_shape)
+            res[f"grad_P{P}_rmv"] = gop.rmatvec(ym).asarray()
+            lop = MPILaplacian(dims=GRAD_DIMS, axes=LAP_AXES, weights=LAP_W,
+                               sampling=LAP_SAMP, edge=False, kind="centered")
+            res[f"lap_P{P}_mv"] = lop.matvec(gx).asarray()
+            res[f"lap_P{P}_rmv"] = lop.rmatvec(
+                dist_from_global(lap_y)).asarray()
         # Fredholm1
         for dt in FRED_DTYPES:
             G = fred_G(dt)
@@ -525,3 +582,26 @@ def _ref_mm_fn(P):
                 res[key + "_rmv"] = op.rmatvec(yd).asarray()
         return res
     return fn
+
+
+def _grad_lap_oracle(P):
+    out = {}
+    ng = int(np.prod(GRAD_DIMS))
+    xg = make_global_x(ng, P)
+    for i in range(3):
+        out[f"grad_P{P}_mv{i}"] = _serial_axis_fd(
+            xg, GRAD_DIMS, i, GRAD_SAMP[i], False, True)
+    ys = [make_global_x(ng, P, seed_shift=2 + i) for i in range(3)]
+    out[f"grad_P{P}_rmv"] = sum(
+        _serial_axis_fd(ys[i], GRAD_DIMS, i, GRAD_SAMP[i], False,
+                        False) for i in range(3))
+    out[f"lap_P{P}_mv"] = sum(
+        LAP_W[i] * _serial_axis_fd(xg, GRAD_DIMS, LAP_AXES[i],
+                                   LAP_SAMP[i], False, True,
+                                   second=True) for i in range(3))
+    yl = make_global_x(ng, P, seed_shift=5)
+    out[f"lap_P{P}_rmv"] = sum(
+        LAP_W[i] * _serial_axis_fd(yl, GRAD_DIMS, LAP_AXES[i],
+                                   LAP_SAMP[i], False, False,
+                                   second=True) for i in range(3))
+    return out
