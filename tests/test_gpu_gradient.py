@@ -169,3 +169,37 @@ def test_mpinonstatconv_world1():
     u = pm.DistributedArray.to_dist(dev(rng.standard_normal(32)))
     v = pm.DistributedArray.to_dist(dev(rng.standard_normal(32)))
     assert pm.dottest(op, u, v, rtol=1e-10)
+
+
+def test_gradient_laplacian_vs_reference_fixtures():
+    """Direct product <-> REFERENCE pin for the composed operators: the
+    P=1 grad_/lap_ fixtures in golden_ref.npz were generated by
+    executing /root/reference/pylops_mpi (tests/golden/refgen.py)."""
+    import os
+    import sys
+    sys.path.insert(0, os.path.join(os.path.dirname(__file__), "golden"))
+    import refgen
+    g = np.load(refgen.GOLDEN_PATH)
+    ng = int(np.prod(refgen.GRAD_DIMS))
+    xg = refgen.make_global_x(ng, 1)
+    gop = pm.MPIGradient(dims=refgen.GRAD_DIMS, sampling=refgen.GRAD_SAMP,
+                         edge=False, kind="centered")
+    xd = pm.DistributedArray.to_dist(dev(xg))
+    ym = gop.matvec(xd)
+    for i in range(3):
+        assert_allclose(host(ym.distarrays[i].asarray()),
+                        g[f"grad_P1_mv{i}"], rtol=1e-12, atol=1e-12)
+    ys = [refgen.make_global_x(ng, 1, seed_shift=2 + i) for i in range(3)]
+    for i, di in enumerate(ym.distarrays):
+        di[:] = dev(ys[i]).reshape(di.local_array.shape)
+    assert_allclose(host(gop.rmatvec(ym).asarray()), g["grad_P1_rmv"],
+                    rtol=1e-12, atol=1e-12)
+    lop = pm.MPILaplacian(dims=refgen.GRAD_DIMS, axes=refgen.LAP_AXES,
+                          weights=refgen.LAP_W, sampling=refgen.LAP_SAMP,
+                          edge=False, kind="centered")
+    assert_allclose(host(lop.matvec(xd).asarray()), g["lap_P1_mv"],
+                    rtol=1e-12, atol=1e-12)
+    yl = refgen.make_global_x(ng, 1, seed_shift=5)
+    assert_allclose(host(lop.rmatvec(
+        pm.DistributedArray.to_dist(dev(yl))).asarray()),
+        g["lap_P1_rmv"], rtol=1e-12, atol=1e-12)
