@@ -88,6 +88,10 @@ class Group:
     def Get_size(self):
         return len(self._ranks)
 
+    @staticmethod
+    def Translate_ranks(g1, ranks, g2):
+        return [g2._ranks.index(g1._ranks[r]) for r in ranks]
+
 
 # --------------------------------------------------------------- world
 class _World:
@@ -261,31 +265,42 @@ class Comm:
             return q
 
     def Send(self, sendspec, dest=0, tag=0):
+        if dest == PROC_NULL:
+            return  # MPI: communication with PROC_NULL is a no-op
         send = _buf_of(sendspec)
         self._q(self._r(), dest, tag).put(_mem_flat(send).copy())
 
     send = Send  # object send at the reference's sites is also an array
 
     def Recv(self, recvspec, source=0, tag=0):
+        if source == PROC_NULL:
+            return  # no-op; recv buffer untouched
         data = self._q(source, self._r(), tag).get(timeout=_RECV_TIMEOUT)
         recv = _buf_of(recvspec)
         _mem_flat(recv)[: data.size] = data.reshape(-1)
 
     def recv(self, source=0, tag=0):
+        if source == PROC_NULL:
+            return None
         return self._q(source, self._r(), tag).get(timeout=_RECV_TIMEOUT)
 
     def Sendrecv(self, sendbuf=None, dest=0, sendtag=0, recvbuf=None,
                  source=0, recvtag=0):
-        send = _buf_of(sendbuf)
-        self._q(self._r(), dest, sendtag).put(_mem_flat(send).copy())
-        data = self._q(source, self._r(), recvtag).get(
-            timeout=_RECV_TIMEOUT)
-        recv = _buf_of(recvbuf)
-        _mem_flat(recv)[: data.size] = data.reshape(-1)
+        if dest != PROC_NULL:
+            send = _buf_of(sendbuf)
+            self._q(self._r(), dest, sendtag).put(_mem_flat(send).copy())
+        if source != PROC_NULL:
+            data = self._q(source, self._r(), recvtag).get(
+                timeout=_RECV_TIMEOUT)
+            recv = _buf_of(recvbuf)
+            _mem_flat(recv)[: data.size] = data.reshape(-1)
 
     def sendrecv(self, sendobj=None, dest=0, sendtag=0, source=0,
                  recvtag=0):
-        self._q(self._r(), dest, sendtag).put(copy.deepcopy(sendobj))
+        if dest != PROC_NULL:
+            self._q(self._r(), dest, sendtag).put(copy.deepcopy(sendobj))
+        if source == PROC_NULL:
+            return None  # MPI: PROC_NULL recv completes with no data
         return self._q(source, self._r(), recvtag).get(
             timeout=_RECV_TIMEOUT)
 
@@ -324,11 +339,56 @@ class Comm:
     def Split_type(self, split_type, key=0):  # COMM_TYPE_SHARED: one node
         return self.Split(color=0, key=key)
 
+    def Create_cart(self, dims, periods=None, reorder=False):
+        """Cartesian topology (ref basicoperators/Halo.py:53,231-241):
+        identity rank order (a valid 'reorder' outcome), row-major
+        coords, PROC_NULL beyond non-periodic edges."""
+        r = self._r()
+        newc = CartComm(self, dims, periods) if r == 0 else None
+        slots = self._xchg(newc)
+        return slots[0]
+
     def Free(self):
         pass
 
     def Dup(self):
         return self
+
+
+class CartComm(Comm):
+    def __init__(self, parent, dims, periods):
+        super().__init__(parent._world, list(parent._members))
+        self._dims = [int(d) for d in dims]
+        self._periods = list(periods) if periods is not None \
+            else [False] * len(self._dims)
+
+    def Get_coords(self, rank):
+        coords = []
+        for d in reversed(self._dims):
+            coords.append(rank % d)
+            rank //= d
+        return list(reversed(coords))
+
+    def _ravel(self, coords):
+        r = 0
+        for d, c in zip(self._dims, coords):
+            r = r * d + c
+        return r
+
+    def Shift(self, direction, disp=1):
+        me = self._r()
+        coords = self.Get_coords(me)
+
+        def nbr(sign):
+            c = list(coords)
+            c[direction] += sign * disp
+            if 0 <= c[direction] < self._dims[direction]:
+                return self._ravel(c)
+            if self._periods[direction]:
+                c[direction] %= self._dims[direction]
+                return self._ravel(c)
+            return PROC_NULL
+        return nbr(-1), nbr(+1)
 
 
 # ----------------------------------------------------------- COMM_WORLD

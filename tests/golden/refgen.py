@@ -276,6 +276,16 @@ def compute_oracle():
         # raises there.
         if P <= 4:
             out.update(_grad_lap_oracle(P))
+        # MPIHalo (Cartesian ghost exchange, ref Halo.py) vs the
+        # restated window semantics
+        for ci, (hdims, grid, hspec) in enumerate(_halo_cases(P)):
+            rngh = np.random.default_rng(90 + ci)
+            G = rngh.standard_normal(hdims)
+            out[f"halo_P{P}_c{ci}_mv"] = np.concatenate(
+                [_halo_window(G, grid, q, hspec).ravel()
+                 for q in range(P)])
+            out[f"halo_P{P}_c{ci}_rmv"] = np.concatenate(
+                [b.ravel() for b in _halo_blocks(G, grid)])
         # Fredholm1
         for dt in FRED_DTYPES:
             G = fred_G(dt)
@@ -514,6 +524,23 @@ def _ref_rank_fn(P):
             res[f"lap_P{P}_mv"] = lop.matvec(gx).asarray()
             res[f"lap_P{P}_rmv"] = lop.rmatvec(
                 dist_from_global(lap_y)).asarray()
+        # MPIHalo
+        if P <= 4:
+            from pylops_mpi import MPIHalo
+            for ci, (hdims, grid, hspec) in enumerate(_halo_cases(P)):
+                rngh = np.random.default_rng(90 + ci)
+                G = rngh.standard_normal(hdims)
+                hop = MPIHalo(hdims, hspec, proc_grid_shape=grid)
+                blocks = _halo_blocks(G, grid)
+                counts = [b.size for b in blocks]
+                hx = DistributedArray(
+                    global_shape=int(np.prod(hdims)),
+                    local_shapes=[(int(v),) for v in counts],
+                    dtype=np.float64)
+                hx[:] = blocks[rank].ravel()
+                hy = hop.matvec(hx)
+                res[f"halo_P{P}_c{ci}_mv"] = hy.asarray()
+                res[f"halo_P{P}_c{ci}_rmv"] = hop.rmatvec(hy).asarray()
         # Fredholm1
         for dt in FRED_DTYPES:
             G = fred_G(dt)
@@ -582,6 +609,67 @@ def _ref_mm_fn(P):
                 res[key + "_rmv"] = op.rmatvec(yd).asarray()
         return res
     return fn
+
+
+def _halo_cases(P):
+    """(dims, grid, halo_spec) per P — Cartesian grids tiling P ranks."""
+    if P == 1:
+        return [((9, 8), (1, 1), 1), ((9, 8), (1, 1), (2, 2, 1, 1))]
+    if P == 2:
+        return [((9, 8), (1, 2), 1), ((9, 8), (2, 1), (2, 2, 1, 1))]
+    if P == 4:
+        return [((9, 8), (1, 4), 1), ((9, 8), (2, 2), (2, 2, 1, 1))]
+    return []
+
+
+def _halo_blocks(G, grid):
+    import math as _m
+    blocks = []
+    for q in range(int(np.prod(grid))):
+        coords = np.unravel_index(q, grid)
+        sl = []
+        for gdim, cc, pp in zip(G.shape, coords, grid):
+            blk = _m.ceil(gdim / pp)
+            sl.append(slice(cc * blk, min(cc * blk + blk, gdim)))
+        blocks.append(G[tuple(sl)].copy())
+    return blocks
+
+
+def _halo_window(G, grid, rank, halo_spec):
+    """The reference's matvec semantics (ref Halo.py:197-227,362-398):
+    scalar halo widths trim to 0 at global borders; tuple widths keep
+    their size with zero padding beyond the domain."""
+    import math as _m
+    nd = G.ndim
+    coords = np.unravel_index(rank, grid)
+    if isinstance(halo_spec, (int, np.integer)):
+        halo = [halo_spec] * (2 * nd)
+        for a in range(nd):
+            if coords[a] == 0:
+                halo[2 * a] = 0
+            if coords[a] == grid[a] - 1:
+                halo[2 * a + 1] = 0
+    else:
+        h = tuple(halo_spec)
+        halo = list(h if len(h) == 2 * nd else
+                    sum(((d, d) for d in h), ()))
+    starts, ends = [], []
+    for gdim, c, pp in zip(G.shape, coords, grid):
+        blk = _m.ceil(gdim / pp)
+        starts.append(c * blk)
+        ends.append(min(c * blk + blk, gdim))
+    ext = tuple((ends[a] - starts[a]) + halo[2 * a] + halo[2 * a + 1]
+                for a in range(nd))
+    out = np.zeros(ext, dtype=G.dtype)
+    src, dst = [], []
+    for a in range(nd):
+        lo = starts[a] - halo[2 * a]
+        hi = ends[a] + halo[2 * a + 1]
+        s0, s1 = max(lo, 0), min(hi, G.shape[a])
+        src.append(slice(s0, s1))
+        dst.append(slice(s0 - lo, (s0 - lo) + (s1 - s0)))
+    out[tuple(dst)] = G[tuple(src)]
+    return out
 
 
 def _grad_lap_oracle(P):
