@@ -286,6 +286,22 @@ def compute_oracle():
                  for q in range(P)])
             out[f"halo_P{P}_c{ci}_rmv"] = np.concatenate(
                 [b.ravel() for b in _halo_blocks(G, grid)])
+        # NonStationaryConvolve1D: the distributed halo/blockdiag
+        # composition equals the GLOBAL serial convolution at P <= 2
+        # (both ranks take the edge branches).  At P >= 3 the middle-
+        # rank branch's sliced anchor set (ihidx+-1, ref
+        # NonStatConvolve1d.py:178-185) clamps filter interpolation in
+        # the outer halo margin where the global operator still
+        # interpolates — a boundary-localized difference inherent to
+        # the reference's slicing, so those P are not fixtured.
+        if P <= 2:
+            from oracle.nsconv import serial_nsconv_mv, serial_nsconv_rmv
+            ndims, hs, ih = _nsc_setup()
+            nn = int(np.prod(ndims))
+            xn = make_global_x(nn, P)
+            yn = make_global_x(nn, P, seed_shift=1)
+            out[f"nsc_P{P}_mv"] = serial_nsconv_mv(xn, ndims, hs, ih, 0)
+            out[f"nsc_P{P}_rmv"] = serial_nsconv_rmv(yn, ndims, hs, ih, 0)
         # Fredholm1
         for dt in FRED_DTYPES:
             G = fred_G(dt)
@@ -541,6 +557,18 @@ def _ref_rank_fn(P):
                 hy = hop.matvec(hx)
                 res[f"halo_P{P}_c{ci}_mv"] = hy.asarray()
                 res[f"halo_P{P}_c{ci}_rmv"] = hop.rmatvec(hy).asarray()
+        # NonStationaryConvolve1D composition (P <= 2 — see the
+        # oracle-side note)
+        if P <= 2:
+            from pylops_mpi.signalprocessing import \
+                MPINonStationaryConvolve1D
+            ndims, hs, ih = _nsc_setup()
+            nn = int(np.prod(ndims))
+            nop = MPINonStationaryConvolve1D(ndims, hs, ih)
+            xn = dist_from_global(make_global_x(nn, P))
+            yn = dist_from_global(make_global_x(nn, P, seed_shift=1))
+            res[f"nsc_P{P}_mv"] = nop.matvec(xn).asarray()
+            res[f"nsc_P{P}_rmv"] = nop.rmatvec(yn).asarray()
         # Fredholm1
         for dt in FRED_DTYPES:
             G = fred_G(dt)
@@ -670,6 +698,16 @@ def _halo_window(G, grid, rank, halo_spec):
         dst.append(slice(s0 - lo, (s0 - lo) + (s1 - s0)))
     out[tuple(dst)] = G[tuple(src)]
     return out
+
+
+def _nsc_setup():
+    # 48-sample domain, anchors every 6 (halo = spacing + hsize//2 + 1
+    # must fit the 12-sample P=4 blocks, ref NonStatConvolve1d.py:120-136)
+    nf, hsize = 8, 5
+    rng = np.random.default_rng(77)
+    hs = rng.standard_normal((nf, hsize))
+    ih = np.arange(3, 48, 6)
+    return (48,), hs, ih
 
 
 def _grad_lap_oracle(P):

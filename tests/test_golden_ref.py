@@ -41,7 +41,7 @@ def test_same_key_sets(golden, oracle_out):
 
 @pytest.mark.parametrize("prefix", [
     "fd1_", "fd2_", "math_", "cgls_", "cg_", "bd_", "fred_",
-    "mm_", "vs_", "grad_", "lap_", "halo_"])
+    "mm_", "vs_", "grad_", "lap_", "halo_", "nsc_"])
 def test_oracle_matches_reference(golden, oracle_out, prefix):
     keys = [k for k in golden.files if k.startswith(prefix)]
     assert keys, f"no golden keys with prefix {prefix}"
