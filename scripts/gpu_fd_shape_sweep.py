@@ -13,15 +13,17 @@ def main():
     g = torch.Generator(device="cuda").manual_seed(1)
     x = pm.DistributedArray((n,))
     x[:] = torch.randn(n, generator=g, dtype=torch.float64, device="cuda")
+    rmv = os.environ.get("RMATVEC") == "1"
+    apply = op.rmatvec if rmv else op.matvec
     for _ in range(3):
-        y = op.matvec(x)
+        y = apply(x)
     torch.cuda.synchronize()
     t = time.perf_counter()
     for _ in range(20):
-        y = op.matvec(x)
+        y = apply(x)
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t) / 20
-    print(f"dims={dims} gy={os.environ.get('PAM_FD_GY','-')} "
+    print(f"dims={dims} {'rmv' if rmv else 'mv'} gy={os.environ.get('PAM_FD_GY','-')} "
           f"cap={os.environ.get('PAM_FD_CAP','-')}: {dt*1e3:7.3f} ms "
           f"{16*n/dt/1e12:5.2f} TB/s")
 
