@@ -41,6 +41,24 @@ __device__ __forceinline__ void loadv(const T* __restrict__ p, T* v) {
   }
 }
 
+// nontemporal-load variant (TH_NT): the rolling stencil reads every
+// input element exactly once, so L2 allocation on its loads is pure
+// overhead — A/B'd via PAM_FD_NTL (r02).
+template <typename T, int V>
+__device__ __forceinline__ void loadv_nt(const T* __restrict__ p, T* v) {
+  if constexpr (V == 1) {
+    v[0] = __builtin_nontemporal_load(p);
+  } else if constexpr (sizeof(T) == 8 && V == 4) {
+    v[0] = __builtin_nontemporal_load(p);
+    v[1] = __builtin_nontemporal_load(p + 1);
+    v[2] = __builtin_nontemporal_load(p + 2);
+    v[3] = __builtin_nontemporal_load(p + 3);
+  } else {
+#pragma unroll
+    for (int k = 0; k < V; ++k) v[k] = __builtin_nontemporal_load(p + k);
+  }
+}
+
 template <typename T, int V>
 __device__ __forceinline__ void storev(T* __restrict__ p, const T* v) {
   if constexpr (V == 1) {

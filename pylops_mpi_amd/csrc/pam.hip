@@ -1123,7 +1123,8 @@ __device__ __forceinline__ void roll_terms(const T (&win)[2 * W + 1][V],
 // mix) is the round-2 question.
 // Guards: load indices are clamped to existing rows (halo planes
 // included); term/edge masks ignore the clamped garbage.
-template <typename T, int OP, int V, int CV, bool NTS = false>
+template <typename T, int OP, int V, int CV, bool NTS = false,
+          bool NTL = false>
 __global__ void __launch_bounds__(BLK) fd_roll_kernel(
     Rows<T> R, T* __restrict__ y, int64_t row0, int64_t N, T c, int edge,
     int64_t rbegin, int64_t rend) {
@@ -1160,7 +1161,8 @@ __global__ void __launch_bounds__(BLK) fd_roll_kernel(
   for (int q = 0; q < CV; ++q)
 #pragma unroll
     for (int k = 0; k < NROLL; ++k)
-      loadv<T, V>(R.row(clamp_row(c0 - W + k)) + js[q], buf[q][k]);
+      if constexpr (NTL) loadv_nt<T, V>(R.row(clamp_row(c0 - W + k)) + js[q], buf[q][k]);
+      else loadv<T, V>(R.row(clamp_row(c0 - W + k)) + js[q], buf[q][k]);
   for (int64_t i = c0; i < c1; ++i) {
     const int64_t g = row0 + i;
     // issue ALL next-row loads first (CV independent chains)
@@ -1168,7 +1170,8 @@ __global__ void __launch_bounds__(BLK) fd_roll_kernel(
     const int64_t rn = clamp_row(i + 1 + W);
 #pragma unroll
     for (int q = 0; q < CV; ++q)
-      loadv<T, V>(R.row(rn) + js[q], nxt[q]);
+      if constexpr (NTL) loadv_nt<T, V>(R.row(rn) + js[q], nxt[q]);
+      else loadv<T, V>(R.row(rn) + js[q], nxt[q]);
 #pragma unroll
     for (int q = 0; q < CV; ++q) {
       T acc[V];
@@ -1411,6 +1414,11 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
     if (gyr > 65535) gyr = 65535;
     dim3 gridr((uint32_t)gx, (uint32_t)gyr);
     const bool rnt = fd_nt_override() != 0;
+    static int ntlov = [] {
+      const char* e = getenv("PAM_FD_NTL");
+      return e ? atoi(e) : 0;
+    }();
+    const bool rntl = ntlov != 0;
 #define ROLL_LAUNCH(KER, VV, CC)                                              \
   do {                                                                        \
     if (rnt)                                                                  \
@@ -1422,6 +1430,15 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
                          R, (T*)y, row0, nglob, (T)coeff, edge, rbegin,       \
                          rend);                                               \
   } while (0)
+#define ROLL_LAUNCH1(VV, CC)                                                  \
+  do {                                                                        \
+    if (rntl)                                                                 \
+      hipLaunchKernelGGL((fd_roll_kernel<T, OP, VV, CC, false, true>),        \
+                         gridr, dim3(BLK), 0, s, R, (T*)y, row0, nglob,       \
+                         (T)coeff, edge, rbegin, rend);                       \
+    else                                                                      \
+      ROLL_LAUNCH(fd_roll_kernel, VV, CC);                                    \
+  } while (0)
     if (rollov == 2) {        // statically-rotated variant
       if (V == 4 && CV == 2) ROLL_LAUNCH(fd_roll2_kernel, 4, 2);
       else if (V == 4 && CV == 8) ROLL_LAUNCH(fd_roll2_kernel, 4, 8);
@@ -1430,13 +1447,14 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
       else if (CV == 8) ROLL_LAUNCH(fd_roll2_kernel, 2, 8);
       else ROLL_LAUNCH(fd_roll2_kernel, 2, 4);
     } else {
-      if (V == 4 && CV == 2) ROLL_LAUNCH(fd_roll_kernel, 4, 2);
-      else if (V == 4 && CV == 8) ROLL_LAUNCH(fd_roll_kernel, 4, 8);
-      else if (V == 4) ROLL_LAUNCH(fd_roll_kernel, 4, 4);
-      else if (CV == 2) ROLL_LAUNCH(fd_roll_kernel, 2, 2);
-      else if (CV == 8) ROLL_LAUNCH(fd_roll_kernel, 2, 8);
-      else ROLL_LAUNCH(fd_roll_kernel, 2, 4);
+      if (V == 4 && CV == 2) ROLL_LAUNCH1(4, 2);
+      else if (V == 4 && CV == 8) ROLL_LAUNCH1(4, 8);
+      else if (V == 4) ROLL_LAUNCH1(4, 4);
+      else if (CV == 2) ROLL_LAUNCH1(2, 2);
+      else if (CV == 8) ROLL_LAUNCH1(2, 8);
+      else ROLL_LAUNCH1(2, 4);
     }
+#undef ROLL_LAUNCH1
 #undef ROLL_LAUNCH
     return check(hipGetLastError());
   }
