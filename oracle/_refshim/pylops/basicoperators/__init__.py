@@ -25,18 +25,22 @@ class MatrixMult(LinearOperator):
 
 
 class Identity(LinearOperator):
-    """pylops.Identity (N == M case; MDC's freq mask uses the
-    rectangular form — not exercised through this stub)."""
+    """pylops.Identity including the rectangular form (N < M truncates,
+    the adjoint zero-pads) — MDC's frequency mask (ref MDC.py:61-64)."""
 
-    def __init__(self, N, M=None, dtype="float64"):
+    def __init__(self, N, M=None, inplace=True, dtype="float64"):
         M = N if M is None else M
         super().__init__(dtype=np.dtype(dtype), shape=(N, M))
 
     def _matvec(self, x):
-        return x.copy()
+        N, M = self.shape
+        return x[:N].copy() if N <= M else np.concatenate(
+            [x, np.zeros(N - M, dtype=x.dtype)])
 
     def _rmatvec(self, x):
-        return x.copy()
+        N, M = self.shape
+        return x[:M].copy() if M <= N else np.concatenate(
+            [x, np.zeros(M - N, dtype=x.dtype)])
 
 
 class _SerialDerivative(LinearOperator):

@@ -302,6 +302,26 @@ def compute_oracle():
             yn = make_global_x(nn, P, seed_shift=1)
             out[f"nsc_P{P}_mv"] = serial_nsconv_mv(xn, ndims, hs, ih, 0)
             out[f"nsc_P{P}_rmv"] = serial_nsconv_rmv(yn, ndims, hs, ih, 0)
+        # MDC chain (composite F1^H I1^H Fr I F with the serial FFT
+        # convention held common — pins the reference's chain
+        # construction: prescale, masks, product/adjoint composites)
+        if P <= 2:
+            nt_, ns_, nr_, nv_, nfreq_, Gm = _mdc_setup()
+            mop = oracle.SimMDC(
+                [Gm[off: off + c] for off, c in _mdc_slices(P)],
+                nt_, nv_, nfreq_, dt=0.4, dr=2.0, twosided=False)
+            xm = make_global_x(nt_ * nr_ * nv_, P)
+            ymv = make_global_x(nt_ * ns_ * nv_, P, seed_shift=1)
+            from oracle.ranksim import Partition as _SP, SimArray as _SA
+            out[f"mdc_P{P}_mv"] = mop.matvec(
+                _SA([xm.copy() for _ in range(P)], xm.shape,
+                    partition=_SP.BROADCAST)).locals[0]
+            # the reference records max|imag| of its (complex-typed)
+            # output; the chain is exactly real, so pin it to 0
+            out[f"mdc_P{P}_mv_imagmax"] = np.array([0.0])
+            out[f"mdc_P{P}_rmv"] = mop.rmatvec(
+                _SA([ymv.copy() for _ in range(P)], ymv.shape,
+                    partition=_SP.BROADCAST)).locals[0]
         # Fredholm1
         for dt in FRED_DTYPES:
             G = fred_G(dt)
@@ -569,6 +589,29 @@ def _ref_rank_fn(P):
             yn = dist_from_global(make_global_x(nn, P, seed_shift=1))
             res[f"nsc_P{P}_mv"] = nop.matvec(xn).asarray()
             res[f"nsc_P{P}_rmv"] = nop.rmatvec(yn).asarray()
+        # MDC chain (serial FFT/Identity via the pylops stubs)
+        if P <= 2:
+            from pylops_mpi.waveeqprocessing import MPIMDC
+            nt_, ns_, nr_, nv_, nfreq_, Gm = _mdc_setup()
+            goff, gcnt = _mdc_slices(P)[rank]
+            mop = MPIMDC(Gm[goff: goff + gcnt], nt=nt_, nv=nv_,
+                         nfreq=nfreq_, dt=0.4, dr=2.0, twosided=False)
+            xm = make_global_x(nt_ * nr_ * nv_, P)
+            ymv = make_global_x(nt_ * ns_ * nv_, P, seed_shift=1)
+            xd = DistributedArray(global_shape=xm.size,
+                                  partition=Partition.BROADCAST,
+                                  dtype=np.float64)
+            xd[:] = xm
+            got = mop.matvec(xd).asarray()
+            res[f"mdc_P{P}_mv"] = np.real(got)
+            res[f"mdc_P{P}_mv_imagmax"] = np.array(
+                [float(np.max(np.abs(np.imag(got))))])
+            yd = DistributedArray(global_shape=ymv.size,
+                                  partition=Partition.BROADCAST,
+                                  dtype=np.float64)
+            yd[:] = ymv
+            gotr = mop.rmatvec(yd).asarray()
+            res[f"mdc_P{P}_rmv"] = np.real(gotr)
         # Fredholm1
         for dt in FRED_DTYPES:
             G = fred_G(dt)
@@ -698,6 +741,22 @@ def _halo_window(G, grid, rank, halo_spec):
         dst.append(slice(s0 - lo, (s0 - lo) + (s1 - s0)))
     out[tuple(dst)] = G[tuple(src)]
     return out
+
+
+def _mdc_slices(P):
+    nfreq = 8  # the Fredholm kernel carries the MASKED frequencies only
+    counts = [oracle.local_split((nfreq,), P, r)[0] for r in range(P)]
+    offs = np.cumsum([0] + counts)
+    return [(int(offs[r]), int(counts[r])) for r in range(P)]
+
+
+def _mdc_setup():
+    nt, ns, nr, nv, nfreq = 20, 3, 4, 2, 8
+    nfft = (nt + 1 + 1) // 2  # ceil((nt+1)/2)
+    rng = np.random.default_rng(66)
+    G = (rng.standard_normal((nfft, ns, nr))
+         + 1j * rng.standard_normal((nfft, ns, nr)))[:nfreq]
+    return nt, ns, nr, nv, nfreq, G
 
 
 def _nsc_setup():

@@ -10,8 +10,10 @@ threads over the mpi4py/pylops shims).  This test runs EVERYWHERE
 whose first link is tests/test_gpu_parity.py.  Covers every FD kind/
 order/edge at P in {1..4} on 1/2/3-D dims, DistributedArray math and
 ghost cells, CGLS traces (damped and undamped), BlockDiag, Fredholm1
-(both dtypes, saveGt both ways) and MatrixMult block+SUMMA (real and
-complex) — 348 pinned arrays.
+(both dtypes, saveGt both ways), MatrixMult block+SUMMA (real and
+complex), Gradient/Laplacian, MPIHalo, NonStationaryConvolve1D and the
+MDC chain (prescale + masks + composite products/adjoints, serial FFT
+convention held common) — 506 pinned arrays.
 """
 import os
 import sys
@@ -41,7 +43,7 @@ def test_same_key_sets(golden, oracle_out):
 
 @pytest.mark.parametrize("prefix", [
     "fd1_", "fd2_", "math_", "cgls_", "cg_", "bd_", "fred_",
-    "mm_", "vs_", "grad_", "lap_", "halo_", "nsc_"])
+    "mm_", "vs_", "grad_", "lap_", "halo_", "nsc_", "mdc_"])
 def test_oracle_matches_reference(golden, oracle_out, prefix):
     keys = [k for k in golden.files if k.startswith(prefix)]
     assert keys, f"no golden keys with prefix {prefix}"
