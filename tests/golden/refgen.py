@@ -304,8 +304,11 @@ def compute_oracle():
             out[f"nsc_P{P}_rmv"] = serial_nsconv_rmv(yn, ndims, hs, ih, 0)
         # MDC chain (composite F1^H I1^H Fr I F with the serial FFT
         # convention held common — pins the reference's chain
-        # construction: prescale, masks, product/adjoint composites)
-        if P <= 2:
+        # construction: prescale, masks, product/adjoint composites);
+        # P<=4: at P=8 the nfreq=8 split leaves 1 freq/rank and the
+        # REFERENCE itself raises ("at least 2 or more elements in the
+        # first dimension")
+        if P <= 4:
             nt_, ns_, nr_, nv_, nfreq_, Gm = _mdc_setup()
             mop = oracle.SimMDC(
                 [Gm[off: off + c] for off, c in _mdc_slices(P)],
@@ -424,6 +427,13 @@ def _ref_rank_fn(P):
     grad_x = make_global_x(ng, P)
     grad_ys = [make_global_x(ng, P, seed_shift=2 + i) for i in range(3)]
     lap_y = make_global_x(ng, P, seed_shift=5)
+    nsc_ndims, nsc_hs, nsc_ih = _nsc_setup()
+    nn_nsc = int(np.prod(nsc_ndims))
+    nsc_x = make_global_x(nn_nsc, P)
+    nsc_y = make_global_x(nn_nsc, P, seed_shift=1)
+    mdc_nt, mdc_ns, mdc_nr, mdc_nv, mdc_nfreq, mdc_G = _mdc_setup()
+    mdc_x = make_global_x(mdc_nt * mdc_nr * mdc_nv, P)
+    mdc_y = make_global_x(mdc_nt * mdc_ns * mdc_nv, P, seed_shift=1)
 
     def fn(rank):
         from pylops_mpi import (DistributedArray, MPIBlockDiag,
@@ -582,22 +592,20 @@ def _ref_rank_fn(P):
         if P <= 2:
             from pylops_mpi.signalprocessing import \
                 MPINonStationaryConvolve1D
-            ndims, hs, ih = _nsc_setup()
-            nn = int(np.prod(ndims))
-            nop = MPINonStationaryConvolve1D(ndims, hs, ih)
-            xn = dist_from_global(make_global_x(nn, P))
-            yn = dist_from_global(make_global_x(nn, P, seed_shift=1))
+            nop = MPINonStationaryConvolve1D(nsc_ndims, nsc_hs, nsc_ih)
+            xn = dist_from_global(nsc_x)
+            yn = dist_from_global(nsc_y)
             res[f"nsc_P{P}_mv"] = nop.matvec(xn).asarray()
             res[f"nsc_P{P}_rmv"] = nop.rmatvec(yn).asarray()
         # MDC chain (serial FFT/Identity via the pylops stubs)
-        if P <= 2:
+        # (P<=4 — the reference needs >=2 freqs/rank, see oracle side)
+        if P <= 4:
             from pylops_mpi.waveeqprocessing import MPIMDC
-            nt_, ns_, nr_, nv_, nfreq_, Gm = _mdc_setup()
             goff, gcnt = _mdc_slices(P)[rank]
-            mop = MPIMDC(Gm[goff: goff + gcnt], nt=nt_, nv=nv_,
-                         nfreq=nfreq_, dt=0.4, dr=2.0, twosided=False)
-            xm = make_global_x(nt_ * nr_ * nv_, P)
-            ymv = make_global_x(nt_ * ns_ * nv_, P, seed_shift=1)
+            mop = MPIMDC(mdc_G[goff: goff + gcnt], nt=mdc_nt, nv=mdc_nv,
+                         nfreq=mdc_nfreq, dt=0.4, dr=2.0, twosided=False)
+            xm = mdc_x
+            ymv = mdc_y
             xd = DistributedArray(global_shape=xm.size,
                                   partition=Partition.BROADCAST,
                                   dtype=np.float64)

@@ -13,7 +13,7 @@ ghost cells, CGLS traces (damped and undamped), BlockDiag, Fredholm1
 (both dtypes, saveGt both ways), MatrixMult block+SUMMA (real and
 complex), Gradient/Laplacian, MPIHalo, NonStationaryConvolve1D and the
 MDC chain (prescale + masks + composite products/adjoints, serial FFT
-convention held common) — 506 pinned arrays.
+convention held common, P=1..4) — 508 pinned arrays.
 """
 import os
 import sys
