@@ -552,6 +552,10 @@ static int cgemm_launch(void* stream, const void* A, const void* B, void* C,
   // pipeline A/B knobs for the shallow-K adjoint (cfg5 rmatvec):
   // PAM_CGEMM_NBUF=1 single-buffers the LDS panels (halves LDS,
   // doubles resident WGs); PAM_CGEMM_BK=8 halves the panel depth.
+  // r02 sweep at the cfg5 shapes (profiles/r02s5/sweep.log): BK=8 wins
+  // the shallow-K adjoint (K=64: 0.262->0.240 ms saveGt=off,
+  // 0.206->0.200 on) and loses the K=256 matvec (0.207->0.218), so
+  // K<=64 auto-selects BK=8; PAM_CGEMM_BK=16 forces the deep panel.
   static int nbov = [] {
     const char* e = getenv("PAM_CGEMM_NBUF");
     return e ? atoi(e) : 0;
@@ -571,7 +575,7 @@ static int cgemm_launch(void* stream, const void* A, const void* B, void* C,
                          (const T*)B, (T*)C, M, N, K, sA, sB, sC);
     return gcheck(hipGetLastError());
   }
-  if (!acc && bkov == 8) {
+  if (!acc && (bkov == 8 || (bkov == 0 && K <= 64))) {
     if (opa)
       hipLaunchKernelGGL((cgemm_batched_kernel<T, true, false, 8>), grid,
                          dim3(GBLK), 0, s, (const T*)A, (const T*)B, (T*)C,
