@@ -45,6 +45,8 @@ from .stencils import (  # noqa: F401
     SimSecondDerivative,
 )
 from .cgls import sim_cgls, sim_cg  # noqa: F401
+from .sparsity import (sim_ista, sim_fista,  # noqa: F401
+                       sim_power_iteration, powerit_rand)
 from .blockdiag import SimBlockDiag  # noqa: F401
 from .fredholm import (SimFredholm1, SimMDC,  # noqa: F401
                        serial_rfft_adj, serial_rfft_op)

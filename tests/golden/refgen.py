@@ -53,6 +53,10 @@ CGLS_DIMS = (17, 5)
 CGLS_NITER = 10
 CGLS_DAMPS = [("d0", 0.0), ("d05", 0.5)]
 CG_NITER = 12
+ISTA_NITER = 10
+ISTA_EPS = 0.2
+ISTA_ALPHA = 0.002   # < 1/lambda_max(B^T B) for the SPD BlockDiag
+POWIT_NITER = 12
 GRAD_DIMS = (10, 6, 4)
 GRAD_SAMP = (1.5, 2.0, 0.5)
 LAP_AXES = (0, 1, 2)
@@ -247,6 +251,28 @@ def compute_oracle():
             CG_NITER, tol=0.0)
         out[f"cg_P{P}_x"] = xs2.asarray()
         out[f"cg_P{P}_cost"] = np.asarray(cost2)
+        # ISTA/FISTA on the same SPD BlockDiag (ref optimization/
+        # sparsity.py:11,136 -> cls_sparsity.py ISTA/FISTA)
+        ysp = _oracle_dist_counts(yg2, spd_counts)
+        x0sp = _oracle_dist_counts(np.zeros(nspd), spd_counts)
+        for tk in ("soft", "hard"):
+            xi, _, ci = oracle.sim_ista(sop2, ysp, x0sp, ISTA_NITER,
+                                        ISTA_EPS, ISTA_ALPHA,
+                                        threshkind=tk, tol=0.0)
+            out[f"ista_P{P}_{tk}_x"] = xi.asarray()
+            out[f"ista_P{P}_{tk}_cost"] = np.asarray(ci)
+        xf, _, cf = oracle.sim_fista(sop2, ysp, x0sp, ISTA_NITER,
+                                     ISTA_EPS, ISTA_ALPHA, tol=0.0)
+        out[f"fista_P{P}_x"] = xf.asarray()
+        out[f"fista_P{P}_cost"] = np.asarray(cf)
+        # power_iteration (ref optimization/eigs.py:10-102; the rand
+        # init is the shared rank-deterministic powerit_rand draw)
+        me, bk, nit = oracle.sim_power_iteration(sop2, spd_counts,
+                                                 niter=POWIT_NITER,
+                                                 tol=1e-5)
+        out[f"powit_P{P}_eig"] = np.array([me])
+        out[f"powit_P{P}_vec"] = bk.asarray()
+        out[f"powit_P{P}_n"] = np.array([float(nit)])
         # VStack vs dense (ref basicoperators/VStack.py:121-150; the
         # oracle side IS the dense algebra — reference == dense pins it)
         vmats = vstack_mats(P)
@@ -378,8 +404,23 @@ def compute_reference():
     from oracle.refrun import run_reference
 
     out = {}
+    from oracle.sparsity import powerit_rand
+
+    def _rank_rand(*shape):
+        # rank-deterministic stand-in for the eigs.py:72-75 init draw
+        # (threads share the global RNG, so the real np.random.rand
+        # would race AND be order-dependent)
+        from mpi4py import MPI
+        assert len(shape) == 1
+        return powerit_rand(MPI.COMM_WORLD.Get_rank(), int(shape[0]))
+
     for P in PS:
-        outs = run_reference(P, _ref_rank_fn(P))
+        saved_rand = np.random.rand
+        np.random.rand = _rank_rand
+        try:
+            outs = run_reference(P, _ref_rank_fn(P))
+        finally:
+            np.random.rand = saved_rand
         for key, val in outs[0].items():
             if key.startswith("__perrank__"):
                 # rank-resolved pieces (ghost cells) come back per rank
@@ -533,6 +574,31 @@ def _ref_rank_fn(P):
         xs2, iit2, cost2 = cg(sbop, ycg, x0cg, niter=CG_NITER, tol=0.0,
                               show=False)
         res[f"cg_P{P}_x"] = xs2.asarray()
+        # ISTA/FISTA on the same SPD BlockDiag (ref optimization/
+        # sparsity.py:11,136)
+        from pylops_mpi.optimization.sparsity import ista, fista
+        ysp = dist_from_counts(cg_y, spd_counts)
+        x0sp = dist_from_counts(np.zeros(n_spd), spd_counts)
+        for tk in ("soft", "hard"):
+            xi, _, ci = ista(sbop, ysp, x0sp, niter=ISTA_NITER,
+                             eps=ISTA_EPS, alpha=ISTA_ALPHA,
+                             threshkind=tk, tol=0.0, show=False)
+            res[f"ista_P{P}_{tk}_x"] = xi.asarray()
+            res[f"ista_P{P}_{tk}_cost"] = np.asarray(ci)
+        xf, _, cf = fista(sbop, ysp, x0sp, niter=ISTA_NITER,
+                          eps=ISTA_EPS, alpha=ISTA_ALPHA, tol=0.0,
+                          show=False)
+        res[f"fista_P{P}_x"] = xf.asarray()
+        res[f"fista_P{P}_cost"] = np.asarray(cf)
+        # power_iteration (ref optimization/eigs.py:10-102; np.random
+        # .rand is patched rank-deterministically by compute_reference)
+        from pylops_mpi.optimization.eigs import power_iteration
+        b0 = dist_from_counts(np.zeros(n_spd), spd_counts)
+        me, bk, nit = power_iteration(sbop, b0, niter=POWIT_NITER,
+                                      tol=1e-5)
+        res[f"powit_P{P}_eig"] = np.array([float(me)])
+        res[f"powit_P{P}_vec"] = bk.asarray()
+        res[f"powit_P{P}_n"] = np.array([float(nit)])
         res[f"cg_P{P}_cost"] = np.asarray(cost2)
         # VStack (matvec: BROADCAST in, SCATTER out; rmatvec: allreduce)
         from pylops_mpi import MPIVStack
