@@ -47,7 +47,8 @@ from .stencils import (  # noqa: F401
 from .cgls import sim_cgls, sim_cg  # noqa: F401
 from .sparsity import (sim_ista, sim_fista,  # noqa: F401
                        sim_power_iteration, powerit_rand)
-from .blockdiag import SimBlockDiag  # noqa: F401
+from .blockdiag import (SimBlockDiag, SimStackedBlockDiag,  # noqa: F401
+                        SimStackedVStack)
 from .fredholm import (SimFredholm1, SimMDC,  # noqa: F401
                        serial_rfft_adj, serial_rfft_op)
 from .nsconv import (serial_nsconv_mv,  # noqa: F401

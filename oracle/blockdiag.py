@@ -9,7 +9,7 @@ from typing import List, Sequence
 
 import numpy as np
 
-from .ranksim import SimArray, reshaped_apply
+from .ranksim import SimArray, SimStackedArray, reshaped_apply
 
 
 class SimBlockDiag:
@@ -51,3 +51,41 @@ class SimBlockDiag:
             r0 += A.shape[0]
             c0 += A.shape[1]
         return out
+
+
+class SimStackedBlockDiag:
+    """ref basicoperators/BlockDiag.py:147-189 MPIStackedBlockDiag:
+    component-wise application of stacked MPI operators."""
+
+    def __init__(self, ops):
+        self.ops = list(ops)
+        self.shape = (int(sum(op.shape[0] for op in self.ops)),
+                      int(sum(op.shape[1] for op in self.ops)))
+
+    def matvec(self, x: SimStackedArray) -> SimStackedArray:
+        return SimStackedArray([op.matvec(xx) for op, xx
+                                in zip(self.ops, x.arrays)])
+
+    def rmatvec(self, x: SimStackedArray) -> SimStackedArray:
+        return SimStackedArray([op.rmatvec(xx) for op, xx
+                                in zip(self.ops, x.arrays)])
+
+
+class SimStackedVStack:
+    """ref basicoperators/VStack.py:153-203 MPIStackedVStack: matvec
+    fans one model out to every operator; rmatvec folds component
+    adjoints in operator order (:199-203)."""
+
+    def __init__(self, ops):
+        self.ops = list(ops)
+        self.shape = (int(sum(op.shape[0] for op in self.ops)),
+                      int(self.ops[0].shape[1]))
+
+    def matvec(self, x: SimArray) -> SimStackedArray:
+        return SimStackedArray([op.matvec(x) for op in self.ops])
+
+    def rmatvec(self, x: SimStackedArray) -> SimArray:
+        y = self.ops[0].rmatvec(x.arrays[0])
+        for xx, op in zip(x.arrays[1:], self.ops[1:]):
+            y = y + op.rmatvec(xx)
+        return y

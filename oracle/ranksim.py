@@ -197,6 +197,72 @@ class SimArray:
         return add_ghost_cells(self.locals, cf, cb, axis=self.axis)
 
 
+class SimStackedArray:
+    """List-of-SimArrays stand-in for a StackedDistributedArray
+    (ref DistributedArray.py:1041-1300): component-wise math, dot as
+    the rank-ordered sum of component dots (:1231-1255), norm as the
+    ord-power fold of component norms (:1257-1281)."""
+
+    def __init__(self, arrays: List["SimArray"]):
+        self.arrays = list(arrays)
+        self.narrays = len(self.arrays)
+
+    def copy(self) -> "SimStackedArray":
+        return SimStackedArray([a.copy() for a in self.arrays])
+
+    def conj(self) -> "SimStackedArray":
+        return SimStackedArray([a.conj() for a in self.arrays])
+
+    def zeros_like(self) -> "SimStackedArray":
+        return SimStackedArray([a.zeros_like() for a in self.arrays])
+
+    def __neg__(self):
+        return SimStackedArray([-a for a in self.arrays])
+
+    def __add__(self, other: "SimStackedArray"):
+        return SimStackedArray([a + b for a, b
+                                in zip(self.arrays, other.arrays)])
+
+    def __sub__(self, other: "SimStackedArray"):
+        return self.__add__(-other)
+
+    def __mul__(self, x):
+        if isinstance(x, SimStackedArray):
+            return SimStackedArray([a * b for a, b
+                                    in zip(self.arrays, x.arrays)])
+        return SimStackedArray([a * x for a in self.arrays])
+
+    __rmul__ = __mul__
+
+    def __iadd__(self, other: "SimStackedArray"):
+        for i in range(self.narrays):
+            self.arrays[i] = self.arrays[i] + other.arrays[i]
+        return self
+
+    def __isub__(self, other: "SimStackedArray"):
+        return self.__iadd__(-other)
+
+    def dot(self, other: "SimStackedArray", vdot: bool = False):
+        dotprod = 0.0
+        for a, b in zip(self.arrays, other.arrays):
+            dotprod += a.dot(b, vdot=vdot)
+        return dotprod
+
+    def norm(self, ord: Optional[int] = None):
+        norms = np.hstack([a.norm(ord) for a in self.arrays])
+        ord = 2 if ord is None else ord
+        if ord == 0:
+            return float(np.sum(norms))
+        if ord == np.inf:
+            return float(np.max(norms))
+        if ord == -np.inf:
+            return float(np.min(norms))
+        return float(np.power(np.sum(np.power(norms, ord)), 1.0 / ord))
+
+    def asarray(self) -> np.ndarray:
+        return np.concatenate([a.asarray().ravel() for a in self.arrays])
+
+
 def reshaped_apply(body, dims: Tuple[int, ...], x: SimArray,
                    target_counts: Optional[Sequence[int]] = None) -> SimArray:
     """The ``@reshaped`` wrapper, ref utils/decorators.py:44-82.

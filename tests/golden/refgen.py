@@ -102,6 +102,19 @@ def spd_mats(P):
     return out
 
 
+def spd2_mats(P):
+    """Second SPD family (same sizes, fresh seed) for the stacked
+    VStack/CGLS cases."""
+    rng = np.random.default_rng(43)
+    sizes = [4, 6, 3, 5]
+    out = []
+    for r in range(P):
+        n = sizes[r % 4]
+        M = rng.standard_normal((n, n))
+        out.append([M @ M.T + n * np.eye(n)])
+    return out
+
+
 def vstack_mats(P):
     """Per-rank blocks sharing 7 columns (MPIVStack requirement)."""
     rng = np.random.default_rng(51)
@@ -295,6 +308,52 @@ def compute_oracle():
         for v in per_rank[1:]:
             acc = acc + v
         out[f"vs_P{P}_rmv"] = acc
+        # HStack dense (ref HStack.py:90-107: VStack-of-adjoints fold —
+        # matvec is the vs_rmv-style rank-ordered reduce of A.T @ seg)
+        hsegs = [yv[offs[i]: offs[i + 1]] for i in range(len(flat))]
+        hper, idx = [], 0
+        for ms in vmats:
+            ss = []
+            for A in ms:
+                ss.append(A.T @ hsegs[idx])
+                idx += 1
+            hper.append(np.sum(np.vstack(ss), axis=0))
+        hacc = hper[0]
+        for v in hper[1:]:
+            hacc = hacc + v
+        out[f"hs_P{P}_mv"] = hacc
+        out[f"hs_P{P}_rmv"] = np.concatenate([A @ xv for A in flat])
+        # Stacked operators + stacked arrays (ref BlockDiag.py:147-189,
+        # VStack.py:153-203; SimStackedArray ranksim mirror of
+        # DistributedArray.py:1041-1300)
+        from oracle.ranksim import SimStackedArray
+        sbd_o = oracle.SimStackedBlockDiag([bop, sop2])
+        xst = SimStackedArray([
+            oracle.to_dist(make_global_x(nc2, P), P),
+            _oracle_dist_counts(make_global_x(nspd, P, seed_shift=6),
+                                spd_counts)])
+        out[f"sbd_P{P}_mv"] = sbd_o.matvec(xst).asarray()
+        yst = SimStackedArray([
+            oracle.to_dist(make_global_x(nr, P, seed_shift=1), P),
+            _oracle_dist_counts(make_global_x(nspd, P, seed_shift=7),
+                                spd_counts)])
+        out[f"sbd_P{P}_rmv"] = sbd_o.rmatvec(yst).asarray()
+        svop_o = oracle.SimStackedVStack(
+            [sop2, oracle.SimBlockDiag(spd2_mats(P))])
+        xsv = _oracle_dist_counts(yg2, spd_counts)
+        out[f"svs_P{P}_mv"] = svop_o.matvec(xsv).asarray()
+        ysv = SimStackedArray([
+            _oracle_dist_counts(make_global_x(nspd, P, seed_shift=8),
+                                spd_counts),
+            _oracle_dist_counts(make_global_x(nspd, P, seed_shift=9),
+                                spd_counts)])
+        out[f"svs_P{P}_rmv"] = svop_o.rmatvec(ysv).asarray()
+        # CGLS over the stacked VStack (damped, stacked data vector)
+        xsc, costsc = oracle.sim_cgls(
+            svop_o, ysv, _oracle_dist_counts(np.zeros(nspd), spd_counts),
+            CGLS_NITER, damp=0.4, tol=0.0)
+        out[f"scgls_P{P}_x"] = xsc.asarray()
+        out[f"scgls_P{P}_cost"] = np.asarray(costsc)
         # Gradient / Laplacian (composed operators; dense expectations
         # are distribution-independent).  P <= 4 only: at P=8 the
         # 10-row dims leave the reference's reshaped rebalance with
@@ -468,6 +527,11 @@ def _ref_rank_fn(P):
     grad_x = make_global_x(ng, P)
     grad_ys = [make_global_x(ng, P, seed_shift=2 + i) for i in range(3)]
     lap_y = make_global_x(ng, P, seed_shift=5)
+    spd2 = spd2_mats(P)
+    sbd_x2 = make_global_x(n_spd, P, seed_shift=6)
+    sbd_y2 = make_global_x(n_spd, P, seed_shift=7)
+    scgls_y1 = make_global_x(n_spd, P, seed_shift=8)
+    scgls_y2 = make_global_x(n_spd, P, seed_shift=9)
     nsc_ndims, nsc_hs, nsc_ih = _nsc_setup()
     nn_nsc = int(np.prod(nsc_ndims))
     nsc_x = make_global_x(nn_nsc, P)
@@ -611,6 +675,45 @@ def _ref_rank_fn(P):
         res[f"vs_P{P}_mv"] = vop.matvec(xvd).asarray()
         yvd = dist_from_global(vs_y)
         res[f"vs_P{P}_rmv"] = vop.rmatvec(yvd).asarray()
+        # HStack (= VStack-of-adjoints, adjointed: ref HStack.py:90-107)
+        from pylops_mpi import MPIHStack
+        hop = MPIHStack(ops=[pylops.MatrixMult(A.T) for A in vmats[rank]])
+        res[f"hs_P{P}_mv"] = hop.matvec(dist_from_global(vs_y)).asarray()
+        yhd = DistributedArray(global_shape=7,
+                               partition=Partition.BROADCAST,
+                               dtype=np.float64)
+        yhd[:] = vs_x
+        res[f"hs_P{P}_rmv"] = hop.rmatvec(yhd).asarray()
+        # Stacked operators + stacked arrays (ref BlockDiag.py:147-189,
+        # VStack.py:153-203, DistributedArray.py:1041-1300)
+        from pylops_mpi import StackedDistributedArray
+        from pylops_mpi.basicoperators import (MPIStackedBlockDiag,
+                                               MPIStackedVStack)
+        sbop_st = MPIStackedBlockDiag(ops=[bop, sbop])
+        xst = StackedDistributedArray(
+            [dist_from_global(bd_x), dist_from_counts(sbd_x2, spd_counts)])
+        res[f"sbd_P{P}_mv"] = np.concatenate(
+            [d.asarray() for d in sbop_st.matvec(xst).distarrays])
+        yst = StackedDistributedArray(
+            [dist_from_global(bd_y), dist_from_counts(sbd_y2, spd_counts)])
+        res[f"sbd_P{P}_rmv"] = np.concatenate(
+            [d.asarray() for d in sbop_st.rmatvec(yst).distarrays])
+        sops2 = [pylops.MatrixMult(A) for A in spd2[rank]]
+        sbop2 = MPIBlockDiag(ops=sops2)
+        svop = MPIStackedVStack(ops=[sbop, sbop2])
+        xsv = dist_from_counts(cg_y, spd_counts)
+        res[f"svs_P{P}_mv"] = np.concatenate(
+            [d.asarray() for d in svop.matvec(xsv).distarrays])
+        ysv = StackedDistributedArray(
+            [dist_from_counts(scgls_y1, spd_counts),
+             dist_from_counts(scgls_y2, spd_counts)])
+        res[f"svs_P{P}_rmv"] = svop.rmatvec(ysv).asarray()
+        # CGLS over the stacked VStack (damped): stacked data vector
+        x0sv = dist_from_counts(np.zeros(n_spd), spd_counts)
+        xsc, _, _, _, _, costsc = cgls(svop, ysv, x0sv, niter=CGLS_NITER,
+                                       damp=0.4, tol=0.0, show=False)
+        res[f"scgls_P{P}_x"] = xsc.asarray()
+        res[f"scgls_P{P}_cost"] = np.asarray(costsc)
         # Gradient (StackedVStack composition) + Laplacian (scaled-sum
         # composite algebra), serial axis>=1 blocks via the pylops stub.
         # P <= 4 only (the reference's own rebalance cannot ghost the

@@ -320,3 +320,41 @@ def test_stacked_operator_algebra_vs_dense():
             [pm.MPIBlockDiag([pm.DenseLocal(dev(A1))])])
     with _pt.raises(ValueError, match="non-negative integer"):
         _ = S ** -1
+
+
+def test_ista_fista_vs_reference_fixtures():
+    """Product ISTA/FISTA on the HIP path vs the REFERENCE-generated
+    P=1 ista_/fista_ fixtures in golden_ref.npz (executed
+    /root/reference/pylops_mpi via tests/golden/refgen.py): same SPD
+    BlockDiag, y, explicit alpha — full 10-iteration model and cost
+    trajectories."""
+    import os
+    import sys
+    sys.path.insert(0, os.path.join(os.path.dirname(__file__), "golden"))
+    import refgen
+    g = np.load(refgen.GOLDEN_PATH)
+    mats = refgen.spd_mats(1)[0]
+    n_spd = int(sum(A.shape[0] for A in mats))
+    yg = refgen.make_global_x(n_spd, 1)
+    op = pm.MPIBlockDiag([pm.DenseLocal(dev(A)) for A in mats])
+
+    def x0():
+        d = pm.DistributedArray((n_spd,))
+        d[:] = 0.0
+        return d
+
+    y = pm.DistributedArray.to_dist(dev(yg))
+    for tk in ("soft", "hard"):
+        xs, niters, cost = pm.ista(
+            op, y, x0(), niter=refgen.ISTA_NITER, eps=refgen.ISTA_EPS,
+            alpha=refgen.ISTA_ALPHA, threshkind=tk, tol=0.0)
+        assert_allclose(host(xs.asarray()), g[f"ista_P1_{tk}_x"],
+                        rtol=1e-10, atol=1e-12)
+        assert_allclose(np.asarray(cost), g[f"ista_P1_{tk}_cost"],
+                        rtol=1e-10)
+    xs, niters, cost = pm.fista(
+        op, y, x0(), niter=refgen.ISTA_NITER, eps=refgen.ISTA_EPS,
+        alpha=refgen.ISTA_ALPHA, tol=0.0)
+    assert_allclose(host(xs.asarray()), g["fista_P1_x"],
+                    rtol=1e-10, atol=1e-12)
+    assert_allclose(np.asarray(cost), g["fista_P1_cost"], rtol=1e-10)
