@@ -47,6 +47,8 @@ from .stencils import (  # noqa: F401
 from .cgls import sim_cgls, sim_cg  # noqa: F401
 from .sparsity import (sim_ista, sim_fista,  # noqa: F401
                        sim_power_iteration, powerit_rand)
+from .proximal import (sim_proximal_gradient_l2l1,  # noqa: F401
+                       sim_admml2_l1, SimScaledOp)
 from .blockdiag import (SimBlockDiag, SimStackedBlockDiag,  # noqa: F401
                         SimStackedVStack)
 from .fredholm import (SimFredholm1, SimMDC,  # noqa: F401

@@ -299,3 +299,85 @@ def half_threshold(x: np.ndarray, thresh: float) -> np.ndarray:
         fac = (2.0 / 3.0) * (1.0 + np.cos(2.0 * np.pi / 3.0
                                           - (2.0 / 3.0) * phi))
     return np.where(a > cut, x * fac, 0.0)
+
+
+class SimScaledOp:
+    """scalar * MPILinearOperator wrap (pylops _ScaledLinearOperator
+    semantics the reference reaches via ``sqrttau * A``)."""
+
+    def __init__(self, op, s):
+        self.op = op
+        self.s = float(s)
+        self.shape = op.shape
+
+    def matvec(self, x):
+        return self.s * self.op.matvec(x)
+
+    def rmatvec(self, x):
+        return self.s * self.op.rmatvec(x)
+
+
+def sim_proximal_gradient_l2l1(Op, b, x0, tau, sigma_l1, niter,
+                               acceleration=None, epsg=1.0):
+    """Rank-sim of the reference's ProximalGradient loop (ref proximal/
+    optimization/primal.py:135-168) with proxf = MPIL2(Op, b)
+    (grad = Op^H(Op x - b), ref proximal/proximal/L2.py:180-189) and
+    proxg = MPIProxOperator(pyproximal L1) (per-rank soft threshold,
+    ref proximal/ProxOperator.py:113-121)."""
+    epsg = np.asarray(epsg, dtype=float)
+    if epsg.size == 1:
+        epsg = epsg * np.ones(niter)
+    t = 1.0
+    x = x0.copy()
+    y = x.copy()
+    for iiter in range(niter):
+        xold = x.copy()
+        g = Op.rmatvec(Op.matvec(y) - b)
+        v = y - tau * g
+        v.locals = [soft_threshold(a, epsg[iiter] * tau * sigma_l1)
+                    for a in v.locals]
+        x = v
+        if acceleration == "vandenberghe":
+            omega = iiter / (iiter + 3)
+        elif acceleration == "fista":
+            told = t
+            t = (1.0 + np.sqrt(1.0 + 4.0 * t ** 2)) / 2.0
+            omega = (told - 1.0) / t
+        else:
+            omega = 0
+        y = x + omega * (x - xold)
+    return x
+
+
+def sim_admml2_l1(Op, b, A, x0, tau, sigma_l1, niter, solver_niter,
+                  solver_tol=0.0, gfirst=False):
+    """Rank-sim of the reference's ADMML2 (ref proximal/optimization/
+    primal.py:306-340): augmented CGLS over MPIStackedVStack
+    [Op, (1/sqrt(tau)) A], per-rank L1 prox on z, running dual u."""
+    from .blockdiag import SimStackedVStack
+    from .cgls import sim_cgls
+    from .ranksim import SimStackedArray
+
+    x = x0.copy()
+    z = A.matvec(x)
+    u = z.zeros_like()
+    sqrttau = 1.0 / sqrt(tau)
+    Opreg = SimStackedVStack([Op, SimScaledOp(A, sqrttau)])
+
+    def _prox(v):
+        v.locals = [soft_threshold(a, tau * sigma_l1) for a in v.locals]
+        return v
+
+    for _ in range(niter):
+        if gfirst:
+            Ax = A.matvec(x)
+            z = _prox(Ax + u)
+            breg = SimStackedArray([b, sqrttau * (z - u)])
+            x, _ = sim_cgls(Opreg, breg, x, solver_niter, tol=solver_tol)
+        else:
+            breg = SimStackedArray([b, sqrttau * (z - u)])
+            x, _ = sim_cgls(Opreg, breg, x, solver_niter, tol=solver_tol)
+            Ax = A.matvec(x)
+            z = _prox(Ax + u)
+        u = u + Ax - z
+    return x, z

@@ -354,6 +354,22 @@ def compute_oracle():
             CGLS_NITER, damp=0.4, tol=0.0)
         out[f"scgls_P{P}_x"] = xsc.asarray()
         out[f"scgls_P{P}_cost"] = np.asarray(costsc)
+        # proximal subpackage (rank-sims of ProximalGradient / ADMML2
+        # with MPIL2 grad + per-rank pyproximal-L1 prox)
+        ypg_o = _oracle_dist_counts(make_global_x(nspd, P, seed_shift=10),
+                                    spd_counts)
+        for an, acc in (("none", None), ("fista", "fista")):
+            xpg = oracle.sim_proximal_gradient_l2l1(
+                sop2, ypg_o, _oracle_dist_counts(np.zeros(nspd),
+                                                 spd_counts),
+                0.01, 0.3, 10, acceleration=acc)
+            out[f"pg_P{P}_{an}_x"] = xpg.asarray()
+        xa, za = oracle.sim_admml2_l1(
+            sop2, ypg_o, svop_o.ops[1],
+            _oracle_dist_counts(np.zeros(nspd), spd_counts),
+            0.05, 0.3, 6, 8, 0.0)
+        out[f"admm_P{P}_x"] = xa.asarray()
+        out[f"admm_P{P}_z"] = za.asarray()
         # Gradient / Laplacian (composed operators; dense expectations
         # are distribution-independent).  P <= 4 only: at P=8 the
         # 10-row dims leave the reference's reshaped rebalance with
@@ -532,6 +548,7 @@ def _ref_rank_fn(P):
     sbd_y2 = make_global_x(n_spd, P, seed_shift=7)
     scgls_y1 = make_global_x(n_spd, P, seed_shift=8)
     scgls_y2 = make_global_x(n_spd, P, seed_shift=9)
+    pg_y = make_global_x(n_spd, P, seed_shift=10)
     nsc_ndims, nsc_hs, nsc_ih = _nsc_setup()
     nn_nsc = int(np.prod(nsc_ndims))
     nsc_x = make_global_x(nn_nsc, P)
@@ -714,6 +731,28 @@ def _ref_rank_fn(P):
                                        damp=0.4, tol=0.0, show=False)
         res[f"scgls_P{P}_x"] = xsc.asarray()
         res[f"scgls_P{P}_cost"] = np.asarray(costsc)
+        # proximal subpackage (ref proximal/ProxOperator.py:113-121,
+        # proximal/proximal/L2.py:180-189, proximal/optimization/
+        # primal.py:135-168 and :306-340)
+        from pyproximal import L1
+        from pylops_mpi.proximal import MPIL2, MPIProxOperator
+        from pylops_mpi.proximal.optimization.primal import (ADMML2,
+                                                             ProximalGradient)
+        l1 = MPIProxOperator(L1(sigma=0.3))
+        ypg = dist_from_counts(pg_y, spd_counts)
+        l2 = MPIL2(Op=sbop, b=ypg,
+                   x0=dist_from_counts(np.zeros(n_spd), spd_counts))
+        for an, acc in (("none", None), ("fista", "fista")):
+            xpg = ProximalGradient(
+                l2, l1, x0=dist_from_counts(np.zeros(n_spd), spd_counts),
+                epsg=1.0, tau=0.01, niter=10, acceleration=acc)
+            res[f"pg_P{P}_{an}_x"] = xpg.asarray()
+        xa, za = ADMML2(l1, sbop, ypg, sbop2,
+                        dist_from_counts(np.zeros(n_spd), spd_counts),
+                        tau=0.05, niter=6,
+                        kwargs_solver={"niter": 8, "tol": 0.0})
+        res[f"admm_P{P}_x"] = xa.asarray()
+        res[f"admm_P{P}_z"] = za.asarray()
         # Gradient (StackedVStack composition) + Laplacian (scaled-sum
         # composite algebra), serial axis>=1 blocks via the pylops stub.
         # P <= 4 only (the reference's own rebalance cannot ghost the
