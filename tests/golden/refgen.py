@@ -227,6 +227,26 @@ def compute_oracle():
         co = np.cumsum([0] + cols)
         out[f"math_P{P}_redist"] = np.concatenate(
             [G2[:, co[r]: co[r + 1]].ravel() for r in range(P)])
+        # masked reductions (ref DistributedArray.py:74-99, :715,
+        # :745-788): each rank reduces over its mask group only, in
+        # rank order within the group
+        if P >= 4:
+            mk = [r // 2 for r in range(P)]
+            cnts = [oracle.local_split((n,), P, r)[0] for r in range(P)]
+            moff = np.cumsum([0] + cnts)
+            chx = [xg[moff[r]: moff[r + 1]] for r in range(P)]
+            chy = [yg[moff[r]: moff[r + 1]] for r in range(P)]
+            dots, n2s, nis = [], [], []
+            for r in range(P):
+                grp = [s_ for s_ in range(P) if mk[s_] == mk[r]]
+                dots.append(sum(np.dot(chx[s_], chy[s_]) for s_ in grp))
+                n2s.append(np.power(
+                    sum(np.sum(np.abs(np.float_power(chx[s_], 2)))
+                        for s_ in grp), 0.5))
+                nis.append(max(np.max(np.abs(chx[s_])) for s_ in grp))
+            out[f"mask_P{P}_dot"] = np.asarray(dots)
+            out[f"mask_P{P}_norm2"] = np.asarray(n2s)
+            out[f"mask_P{P}_norminf"] = np.asarray(nis)
         # CGLS on FD1 centered3.  x0 must be plane-aligned: the
         # reference's reshaped wrapper leaves operator OUTPUTS on the
         # plane split (ref decorators.py:79-82), and CGLS subtracts
@@ -594,6 +614,7 @@ def _ref_rank_fn(P):
         n = nmath
         xg, yg = math_x, math_y
         dx, dy = dist_from_global(xg), dist_from_global(yg)
+        counts0 = [oracle.local_split((n,), P, r)[0] for r in range(P)]
         res[f"math_P{P}_dot"] = np.asarray(dx.dot(dy))
         dxa = dist_from_global(np.abs(xg))
         for name, o in NORM_ORDS:
@@ -610,6 +631,22 @@ def _ref_rank_fn(P):
                 cells_front=w, cells_back=w)
         rd = d2.redistribute(axis=1)
         res[f"__perrank__math_P{P}_redist"] = np.asarray(rd.local_array)
+        # masked (sub-communicator) reductions: dot and norms reduce
+        # over the mask group only (ref DistributedArray.py:74-99,
+        # :194-195, :715, :745-788) — results are rank-dependent
+        if P >= 4:
+            mk = [r // 2 for r in range(P)]
+            md = DistributedArray(global_shape=n, mask=mk,
+                                  dtype=np.float64)
+            nd = DistributedArray(global_shape=n, mask=mk,
+                                  dtype=np.float64)
+            offm = int(np.sum(counts0[:rank], initial=0))
+            md[:] = xg[offm: offm + md.local_shape[0]]
+            nd[:] = yg[offm: offm + nd.local_shape[0]]
+            res[f"__perrank__mask_P{P}_dot"] = np.atleast_1d(md.dot(nd))
+            res[f"__perrank__mask_P{P}_norm2"] = np.atleast_1d(md.norm())
+            res[f"__perrank__mask_P{P}_norminf"] = np.atleast_1d(
+                md.norm(np.inf))
         # CGLS
         nc = ncgls
         xg = cgls_x

@@ -17,7 +17,8 @@ convention held common, P=1..4), ISTA/FISTA (soft+hard) and
 power_iteration (rank-deterministic init draw), MPIHStack, stacked
 operators (MPIStackedBlockDiag/VStack + StackedDistributedArray) and
 damped CGLS over a stacked VStack, and the proximal subpackage
-(ProximalGradient none/fista + ADMML2 over MPIL2/L1) — 613 pinned
+(ProximalGradient none/fista + ADMML2 over MPIL2/L1), and masked
+sub-communicator reductions (dot/norms per mask group) — 619 pinned
 arrays.
 """
 import os
@@ -49,7 +50,7 @@ def test_same_key_sets(golden, oracle_out):
 @pytest.mark.parametrize("prefix", [
     "fd1_", "fd2_", "math_", "cgls_", "cg_", "bd_", "fred_",
     "mm_", "vs_", "grad_", "lap_", "halo_", "nsc_", "mdc_",
-    "ista_", "fista_", "powit_", "hs_", "sbd_", "svs_", "scgls_", "pg_", "admm_"])
+    "ista_", "fista_", "powit_", "hs_", "sbd_", "svs_", "scgls_", "pg_", "admm_", "mask_"])
 def test_oracle_matches_reference(golden, oracle_out, prefix):
     keys = [k for k in golden.files if k.startswith(prefix)]
     assert keys, f"no golden keys with prefix {prefix}"
