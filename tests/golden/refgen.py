@@ -491,6 +491,41 @@ def compute_oracle():
                     [v.ravel() for v in mv])
                 out[key + "_rmv"] = np.concatenate(
                     [v.ravel() for v in rmv])
+    # grid helpers (ref MatrixMult.py:24-171): active-grid selection,
+    # ceil-division block slices, allgather reassembly — dense
+    # restatement of the published arithmetic, incl. inactive ranks
+    import math as _math
+    for P in PS:
+        for (N, M) in ((7, 9), (2, 9)):
+            pp = _math.isqrt(P)
+            ad = min(N, M, pp)
+            active = [r for r in range(P)
+                      if (r // pp) < ad and (r % pp) < ad]
+            ppn = _math.isqrt(len(active))
+            grid, blocks = [], []
+            for r in range(P):
+                row, col = divmod(r, pp)
+                act = row < ad and col < ad
+                if act:
+                    nr_ = active.index(r)
+                    nrow, ncol = divmod(nr_, ppn)
+                    grid.append([nr_, nrow, ncol, 1.0])
+                    new_r = -(-N // ppn) * ppn
+                    new_c = -(-M // ppn) * ppn
+                    blkr, blkc = new_r // ppn, new_c // ppn
+                    rs, cs = nrow * blkr, ncol * blkc
+                    re_, ce = min(rs + blkr, N), min(cs + blkc, M)
+                    blocks.append([rs, re_, cs, ce])
+                else:
+                    grid.append([r, row, col, 0.0])
+                    blocks.append([-1.0, -1.0, -1.0, -1.0])
+            out[f"mmu_P{P}_{N}x{M}_grid"] = np.asarray(
+                grid, dtype=float).ravel()
+            out[f"mmu_P{P}_{N}x{M}_block"] = np.asarray(
+                blocks, dtype=float).ravel()
+            # block_gather reassembles the blocks exactly into G
+            rng = np.random.default_rng(55)
+            out[f"mmu_P{P}_{N}x{M}_gather"] = rng.standard_normal((N, M))
     return out
 
 
@@ -528,6 +563,15 @@ def compute_reference():
         outs = run_reference(P, _ref_mm_fn(P))
         for key, val in outs[0].items():
             out[key] = np.asarray(val)
+    for P in PS:
+        outs = run_reference(P, _ref_mmutil_fn(P))
+        for key, val in outs[0].items():
+            if key.startswith("__perrank__"):
+                real = key[len("__perrank__"):]
+                out[real] = np.concatenate(
+                    [np.asarray(o[key]).ravel() for o in outs])
+            else:
+                out[key] = np.asarray(val)
     return out
 
 
@@ -931,6 +975,48 @@ def _ref_mm_fn(P):
                     local_shapes=[(int(v),) for v in ycounts], dtype=dt)
                 yd[:] = ylocals[rank]
                 res[key + "_rmv"] = op.rmatvec(yd).asarray()
+        return res
+    return fn
+
+
+def _ref_mmutil_fn(P):
+    """active_grid_comm / local_block_split / block_gather pins (ref
+    MatrixMult.py:24-171) — pure grid arithmetic plus the allgather
+    reassembly, incl. inactive ranks at non-square P."""
+    def fn(rank):
+        from mpi4py import MPI
+        from pylops_mpi import DistributedArray
+        from pylops_mpi.basicoperators.MatrixMult import (
+            active_grid_comm, block_gather, local_block_split)
+        res = {}
+        for (N, M) in ((7, 9), (2, 9)):
+            comm, nr, row, col, act = active_grid_comm(
+                MPI.COMM_WORLD, N, M)
+            res[f"__perrank__mmu_P{P}_{N}x{M}_grid"] = np.array(
+                [nr, row, col, int(act)], dtype=float)
+            if act:
+                rs, cs = local_block_split((N, M), nr, comm)
+                res[f"__perrank__mmu_P{P}_{N}x{M}_block"] = np.array(
+                    [rs.start, rs.stop, cs.start, cs.stop], dtype=float)
+                rng = np.random.default_rng(55)
+                G = rng.standard_normal((N, M))
+                pa = comm.Get_size()
+                blocks = []
+                for q in range(pa):
+                    qrs, qcs = local_block_split((N, M), q, comm)
+                    blocks.append(G[qrs, qcs])
+                xd = DistributedArray(
+                    global_shape=int(sum(b.size for b in blocks)),
+                    base_comm=comm,
+                    local_shapes=[(int(b.size),) for b in blocks],
+                    dtype=np.float64)
+                xd[:] = blocks[nr].ravel()
+                C = block_gather(xd, (N, M), comm)   # collective
+                if rank == 0:
+                    res[f"mmu_P{P}_{N}x{M}_gather"] = C
+            else:
+                res[f"__perrank__mmu_P{P}_{N}x{M}_block"] = np.full(
+                    4, -1.0)
         return res
     return fn
 
