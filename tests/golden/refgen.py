@@ -526,6 +526,15 @@ def compute_oracle():
             # block_gather reassembles the blocks exactly into G
             rng = np.random.default_rng(55)
             out[f"mmu_P{P}_{N}x{M}_gather"] = rng.standard_normal((N, M))
+    # the non-square-world flow at P=8: the operator runs on the 2x2
+    # ACTIVE sub-grid, so expectations are the P=4 dense folds
+    rng = np.random.default_rng(77)
+    A = rng.standard_normal(MM_SHAPES[:2])
+    X = rng.standard_normal(MM_SHAPES[1:])
+    out["mmsub_P8_block_mv"] = np.concatenate(
+        [v.ravel() for v in om.block_expected_mv(A, X, 4)])
+    out["mmsub_P8_summa_mv"] = np.concatenate(
+        [v.ravel() for v in om.summa_expected_mv(A, X, 4)])
     return out
 
 
@@ -1017,6 +1026,34 @@ def _ref_mmutil_fn(P):
             else:
                 res[f"__perrank__mmu_P{P}_{N}x{M}_block"] = np.full(
                     4, -1.0)
+        # the documented non-square-world flow (ref MatrixMult.py
+        # docstrings): build MPIMatrixMult on the ACTIVE sub-comm of a
+        # non-square world; inactive ranks stay idle
+        if P == 8:
+            from pylops_mpi.basicoperators.MatrixMult import MPIMatrixMult
+            Nm, Km, Mm = MM_SHAPES
+            rng = np.random.default_rng(77)
+            A = rng.standard_normal((Nm, Km))
+            X = rng.standard_normal((Km, Mm))
+            comm, nr, _, _, act = active_grid_comm(MPI.COMM_WORLD, Nm, Mm)
+            if act:
+                pa = comm.Get_size()
+                for kind in ("block", "summa"):
+                    if kind == "block":
+                        inputs = om.block_inputs(A, X, pa)
+                    else:
+                        inputs = om.summa_inputs(A, X, pa)
+                    op = MPIMatrixMult(inputs[nr][0], Mm, kind=kind,
+                                       base_comm=comm, dtype="float64")
+                    counts = [v.size for _, v in inputs]
+                    xd = DistributedArray(
+                        global_shape=int(sum(counts)), base_comm=comm,
+                        local_shapes=[(int(v),) for v in counts],
+                        dtype=np.float64)
+                    xd[:] = inputs[nr][1].ravel()
+                    yfull = op.matvec(xd).asarray()   # collective
+                    if rank == 0:
+                        res[f"mmsub_P{P}_{kind}_mv"] = yfull
         return res
     return fn
 

@@ -20,8 +20,9 @@ damped CGLS over a stacked VStack, and the proximal subpackage
 (ProximalGradient none/fista + ADMML2 over MPIL2/L1), and masked
 sub-communicator reductions (dot/norms per mask group), and the
 MatrixMult grid helpers (active_grid_comm / local_block_split /
-block_gather, incl. inactive ranks at non-square P) — 649 pinned
-arrays.
+block_gather, incl. inactive ranks at non-square P, plus the full
+non-square-world flow: block+SUMMA MatrixMult on the active 2x2
+sub-communicator of an 8-rank world) — 651 pinned arrays.
 """
 import os
 import sys
@@ -52,7 +53,7 @@ def test_same_key_sets(golden, oracle_out):
 @pytest.mark.parametrize("prefix", [
     "fd1_", "fd2_", "math_", "cgls_", "cg_", "bd_", "fred_",
     "mm_", "vs_", "grad_", "lap_", "halo_", "nsc_", "mdc_",
-    "ista_", "fista_", "powit_", "hs_", "sbd_", "svs_", "scgls_", "pg_", "admm_", "mask_", "mmu_"])
+    "ista_", "fista_", "powit_", "hs_", "sbd_", "svs_", "scgls_", "pg_", "admm_", "mask_", "mmu_", "mmsub_"])
 def test_oracle_matches_reference(golden, oracle_out, prefix):
     keys = [k for k in golden.files if k.startswith(prefix)]
     assert keys, f"no golden keys with prefix {prefix}"
