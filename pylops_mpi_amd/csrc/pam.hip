@@ -1378,8 +1378,18 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
   // V=4/CV=8/TGT=2048 (the swept optimum; NT stores -15%, roll2's
   // static rotation spills at CV8V4).  The bench shape (2 MiB rows)
   // keeps the row-parallel kernel (5.78 vs 5.22).  PAM_FD_ROLL: unset =
-  // auto (rows >= PAM_FD_LONGROW bytes, default 4 MiB), -1 = force
-  // row-parallel, 1/2 = force rolling (2 = statically-rotated variant).
+  // auto, -1 = force row-parallel, 1/2 = force rolling (2 = the
+  // statically-rotated variant).  The r02s10/s11 12-shape crossover
+  // sweep (profiles/) refined the auto rule: the rolling kernel is
+  // flat (~5.0-5.3 TB/s) while row-parallel is peaky (4.8-5.7), and
+  // row-parallel loses not only above 4 MiB rows but ALSO at mid-size
+  // rows whenever the local row count is not a multiple of 1024 (its
+  // gy row-grid then maps raggedly across the 8 XCDs and the L2
+  // neighbour-row grouping degrades: 1536-row shapes lose 4-7% at
+  // 2.5-3 MiB rows).  Rule (fits all 12 measured shapes): roll when
+  // rowbytes > PAM_FD_LONGROW (default 4 MiB, strict — the 4 MiB
+  // power-of-two shape prefers row-parallel), or when rowbytes >=
+  // 5/8 of it (2.5 MiB) and nrows % 1024 != 0.
   static int rollov = [] {
     const char* e = getenv("PAM_FD_ROLL");
     return e ? atoi(e) : 0;
@@ -1388,8 +1398,11 @@ static int fd_launch(void* stream, int edge, const void* x, const void* gf,
     const char* e = getenv("PAM_FD_LONGROW");
     return (int64_t)(e ? atoll(e) : (4 << 20));
   }();
+  const int64_t rowbytes = m * (int64_t)sizeof(T);
   const bool roll_auto =
-      rollov == 0 && V > 1 && m * (int64_t)sizeof(T) >= longrow;
+      rollov == 0 && V > 1 &&
+      (rowbytes > longrow ||
+       (rowbytes * 8 >= longrow * 5 && (nrows % 1024) != 0));
   static int rolltgt = [] {
     const char* e = getenv("PAM_FD_ROLL_TGT");
     return e ? atoi(e) : 2048;
