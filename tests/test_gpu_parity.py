@@ -147,6 +147,31 @@ def test_large_3d_fd1():
                     sop.rmatvec(sx).asarray(), rtol=1e-13, atol=1e-14)
 
 
+@pytest.mark.parametrize("dims", [
+    (8, 1280, 256),    # 2.5 MiB rows + ragged nrows -> NEW ragged branch
+    (6, 2048, 512),    # 8 MiB rows -> long-row branch
+])
+def test_fd1_rolling_dispatch_vs_oracle(dims):
+    """Both auto-roll branches of the r02 dispatch rule (pam.hip
+    fd_launch: rowbytes > 4 MiB, or >= 2.5 MiB with nrows % 1024 != 0)
+    route to the rolling-window kernel — pin it against the oracle at
+    oracle-checkable sizes, all FD kinds."""
+    n = int(np.prod(dims))
+    rng = np.random.default_rng(11)
+    xg = rng.standard_normal(n)
+    for kind, order in (("centered", 3), ("centered", 5),
+                        ("forward", 3), ("backward", 3)):
+        op = pm.MPIFirstDerivative(dims, sampling=1.5, kind=kind,
+                                   order=order)
+        sop = oracle.SimFirstDerivative(dims, 1.5, kind, False, order)
+        x = pm.DistributedArray.to_dist(dev(xg))
+        sx = oracle.to_dist(xg, 1)
+        assert_allclose(host(op.matvec(x).asarray()),
+                        sop.matvec(sx).asarray(), rtol=1e-13, atol=1e-14)
+        assert_allclose(host(op.rmatvec(x).asarray()),
+                        sop.rmatvec(sx).asarray(), rtol=1e-13, atol=1e-14)
+
+
 # -------------------------------------------------------------- dottest
 @pytest.mark.parametrize("make_op,make_sop", [
     (lambda: pm.MPIFirstDerivative((24, 5), 0.7, "centered", order=5),
