@@ -247,6 +247,10 @@ def compute_oracle():
             out[f"mask_P{P}_dot"] = np.asarray(dots)
             out[f"mask_P{P}_norm2"] = np.asarray(n2s)
             out[f"mask_P{P}_norminf"] = np.asarray(nis)
+            out[f"mask_P{P}_asarray"] = np.concatenate(
+                [np.concatenate([chx[s_] for s_ in range(P)
+                                 if mk[s_] == mk[r]])
+                 for r in range(P)])
         # CGLS on FD1 centered3.  x0 must be plane-aligned: the
         # reference's reshaped wrapper leaves operator OUTPUTS on the
         # plane split (ref decorators.py:79-82), and CGLS subtracts
@@ -700,6 +704,11 @@ def _ref_rank_fn(P):
             res[f"__perrank__mask_P{P}_norm2"] = np.atleast_1d(md.norm())
             res[f"__perrank__mask_P{P}_norminf"] = np.atleast_1d(
                 md.norm(np.inf))
+            # masked asarray gathers over the sub-communicator only
+            # (ref DistributedArray.py:401-436) — collective on every
+            # rank, result differs per mask group
+            res[f"__perrank__mask_P{P}_asarray"] = md.asarray(
+                masked=True)
         # CGLS
         nc = ncgls
         xg = cgls_x

@@ -22,7 +22,7 @@ sub-communicator reductions (dot/norms per mask group), and the
 MatrixMult grid helpers (active_grid_comm / local_block_split /
 block_gather, incl. inactive ranks at non-square P, plus the full
 non-square-world flow: block+SUMMA MatrixMult on the active 2x2
-sub-communicator of an 8-rank world) — 651 pinned arrays.
+sub-communicator of an 8-rank world) — 653 pinned arrays.
 """
 import os
 import sys
