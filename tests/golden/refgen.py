@@ -604,7 +604,6 @@ def _ref_rank_fn(P):
     ncgls = int(np.prod(CGLS_DIMS))
     cgls_x = make_global_x(ncgls, P)
     bd_mats = blockdiag_mats(P)
-    import oracle as _oracle_pkg
     nr_bd = int(sum(sum(A.shape[0] for A in ms) for ms in bd_mats))
     nc_bd = int(sum(sum(A.shape[1] for A in ms) for ms in bd_mats))
     bd_x = make_global_x(nc_bd, P)
