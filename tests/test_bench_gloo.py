@@ -6,10 +6,19 @@ before any 8-GPU lease exists.  --smoke-gloo swaps the GPU kernel for a
 tiny torch stand-in; it never touches the measured path."""
 import json
 import os
+import socket
 import subprocess
 import sys
 
 import pytest
+
+
+def _free_port():
+    # a fixed port flakes: TIME_WAIT from the previous world's
+    # rendezvous can hold it for ~60 s
+    with socket.socket() as sck:
+        sck.bind(("127.0.0.1", 0))
+        return sck.getsockname()[1]
 
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
@@ -22,7 +31,7 @@ def test_bench_smoke_gloo(world):
     cmd = [sys.executable, "-m", "torch.distributed.run",
            "--nnodes=1", f"--nproc-per-node={world}",
            "--master-addr", "127.0.0.1",
-           "--master-port", str(29150 + world),
+           "--master-port", str(_free_port()),
            os.path.join(ROOT, "bench.py"),
            "--gpus", str(world), "--steps", "2", "--warmup", "1",
            "--smoke-gloo"]
